@@ -1,0 +1,248 @@
+"""ctypes binding over libtezsort.so — the PRODUCT compute path.
+
+No CPU fallback exists here by design: if the HIP extension is missing or a
+GPU is unavailable, calls raise immediately.  The CPU oracle
+under /root/repo/oracle is test-only and is never imported from this package.
+"""
+import ctypes
+import os
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB = os.path.join(_DIR, "libtezsort.so")
+
+KEY_BYTES = 0
+KEY_TEXT = 1
+CMP_TEZBYTES = 0
+CMP_TEXT = 1
+
+
+class TzsConf(ctypes.Structure):
+    _fields_ = [
+        ("num_partitions", ctypes.c_int32),
+        ("key_type", ctypes.c_int32),
+        ("value_type", ctypes.c_int32),
+        ("comparator", ctypes.c_int32),
+        ("rle", ctypes.c_int32),
+        ("send_empty_partition_details", ctypes.c_int32),
+        ("io_sort_factor", ctypes.c_int32),
+        ("final_merge_enabled", ctypes.c_int32),
+        ("sort_buffer_bytes", ctypes.c_int64),
+        ("device", ctypes.c_int32),
+        ("world_size", ctypes.c_int32),
+        ("rank", ctypes.c_int32),
+        ("reserved0", ctypes.c_int32),
+    ]
+
+
+class TzsIndexRecord(ctypes.Structure):
+    _fields_ = [("start_offset", ctypes.c_int64),
+                ("raw_length", ctypes.c_int64),
+                ("part_length", ctypes.c_int64)]
+
+
+class TzsCounters(ctypes.Structure):
+    _fields_ = [(n, ctypes.c_int64) for n in
+                ("output_records", "output_bytes", "output_bytes_with_overhead",
+                 "spilled_records", "num_spills", "rle_applied")]
+
+
+class TzsTimes(ctypes.Structure):
+    _fields_ = [(n, ctypes.c_int64) for n in
+                ("absorb_ns", "composite_ns", "sort_ns", "permute_ns", "emit_ns",
+                 "crc_ns", "merge_ns", "total_ns", "sort_passes",
+                 "dominant_kernel_ns")]
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB):
+        raise RuntimeError(
+            f"tez_amd native engine missing: {_LIB} not built. "
+            "Run __graft_entry__.build() (hipcc --offload-arch=gfx950). "
+            "There is no CPU fallback.")
+    L = ctypes.CDLL(_LIB)
+    c = ctypes
+    L.tzs_last_error.restype = c.c_char_p
+    L.tzs_conf_default.argtypes = [c.POINTER(TzsConf), c.c_int32]
+    L.tzs_sorter_create.argtypes = [c.POINTER(TzsConf), c.POINTER(c.c_void_p)]
+    L.tzs_sorter_write.argtypes = [c.c_void_p, c.c_void_p, c.c_int32, c.c_void_p,
+                                   c.c_int32, c.c_int32]
+    L.tzs_sorter_write_batch_device.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p,
+                                                c.c_void_p, c.c_void_p, c.c_int64]
+    L.tzs_sorter_spill.argtypes = [c.c_void_p]
+    L.tzs_sorter_flush.argtypes = [c.c_void_p]
+    L.tzs_sorter_num_spills.argtypes = [c.c_void_p]
+    L.tzs_sorter_output.argtypes = [c.c_void_p, c.POINTER(c.c_void_p),
+                                    c.POINTER(c.c_int64), c.POINTER(TzsIndexRecord)]
+    L.tzs_sorter_spill_output.argtypes = [c.c_void_p, c.c_int32, c.POINTER(c.c_void_p),
+                                          c.POINTER(c.c_int64), c.POINTER(TzsIndexRecord)]
+    L.tzs_sorter_write_files.argtypes = [c.c_void_p, c.c_char_p, c.c_char_p]
+    L.tzs_sorter_counters.argtypes = [c.c_void_p, c.POINTER(TzsCounters)]
+    L.tzs_sorter_times.argtypes = [c.c_void_p, c.POINTER(TzsTimes)]
+    L.tzs_sorter_close.argtypes = [c.c_void_p]
+    L.tzs_generate.argtypes = [c.c_uint64, c.c_int64, c.c_int32, c.c_int32, c.c_int32,
+                               c.POINTER(TzsConf), c.POINTER(c.c_void_p),
+                               c.POINTER(c.c_void_p), c.POINTER(c.c_void_p),
+                               c.POINTER(c.c_void_p)]
+    L.tzs_free_device.argtypes = [c.c_void_p]
+    L.tzs_memcpy_d2h.argtypes = [c.c_void_p, c.c_void_p, c.c_uint64]
+    L.tzs_memcpy_h2d.argtypes = [c.c_void_p, c.c_void_p, c.c_uint64]
+    L.tzs_malloc_device.argtypes = [c.c_uint64, c.POINTER(c.c_void_p)]
+    L.tzs_device_available.restype = c.c_int
+    L.tzs_test_crc_combine.restype = c.c_uint32
+    L.tzs_test_crc_combine.argtypes = [c.c_uint32, c.c_uint32, c.c_uint64]
+    L.tzs_test_crc32.restype = c.c_uint32
+    L.tzs_test_crc32.argtypes = [c.c_uint32, c.c_void_p, c.c_uint64]
+    _lib = L
+    return L
+
+
+def _ck(rc, what):
+    if rc < 0:
+        raise RuntimeError(f"{what} failed rc={rc}: {lib().tzs_last_error().decode()}")
+    return rc
+
+
+def make_conf(num_partitions, **kw):
+    c = TzsConf()
+    lib().tzs_conf_default(ctypes.byref(c), num_partitions)
+    for k, v in kw.items():
+        setattr(c, k, v)
+    return c
+
+
+class Sorter:
+    """The ExternalSorter replacement behind OrderedPartitionedKVOutput
+    (see include/tezsort.h citations)."""
+
+    def __init__(self, conf: TzsConf):
+        self.conf = conf
+        h = ctypes.c_void_p()
+        _ck(lib().tzs_sorter_create(ctypes.byref(conf), ctypes.byref(h)), "create")
+        self.h = h
+
+    def write(self, key: bytes, val: bytes, partition: int = -1):
+        _ck(lib().tzs_sorter_write(self.h, key, len(key), val, len(val), partition),
+            "write")
+
+    def write_batch_device(self, d_data, d_off, d_klen, d_part, n):
+        _ck(lib().tzs_sorter_write_batch_device(
+            self.h, d_data, d_off, d_klen, d_part, n), "write_batch")
+
+    def spill(self):
+        return _ck(lib().tzs_sorter_spill(self.h), "spill")
+
+    def flush(self):
+        _ck(lib().tzs_sorter_flush(self.h), "flush")
+
+    def num_spills(self):
+        return lib().tzs_sorter_num_spills(self.h)
+
+    def output(self):
+        """Returns (ifile_bytes: bytes, index: list[(start,raw,part)])."""
+        p = ctypes.c_void_p()
+        n = ctypes.c_int64()
+        idx = (TzsIndexRecord * self.conf.num_partitions)()
+        _ck(lib().tzs_sorter_output(self.h, ctypes.byref(p), ctypes.byref(n), idx),
+            "output")
+        buf = bytearray(n.value)
+        if n.value:
+            ba = (ctypes.c_char * n.value).from_buffer(buf)
+            _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba), p, n.value), "d2h")
+        return bytes(buf), [(r.start_offset, r.raw_length, r.part_length) for r in idx]
+
+    def spill_output(self, spill_id):
+        p = ctypes.c_void_p()
+        n = ctypes.c_int64()
+        idx = (TzsIndexRecord * self.conf.num_partitions)()
+        _ck(lib().tzs_sorter_spill_output(self.h, spill_id, ctypes.byref(p),
+                                          ctypes.byref(n), idx), "spill_output")
+        buf = bytearray(n.value)
+        if n.value:
+            ba = (ctypes.c_char * n.value).from_buffer(buf)
+            _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba), p, n.value), "d2h")
+        return bytes(buf), [(r.start_offset, r.raw_length, r.part_length) for r in idx]
+
+    def write_files(self, local_dir: str, unique_id: str):
+        _ck(lib().tzs_sorter_write_files(self.h, local_dir.encode(), unique_id.encode()),
+            "write_files")
+
+    def counters(self):
+        c = TzsCounters()
+        _ck(lib().tzs_sorter_counters(self.h, ctypes.byref(c)), "counters")
+        return {n: getattr(c, n) for n, _ in c._fields_}
+
+    def times(self):
+        t = TzsTimes()
+        _ck(lib().tzs_sorter_times(self.h, ctypes.byref(t)), "times")
+        return {n: getattr(t, n) for n, _ in t._fields_}
+
+    def close(self):
+        if self.h:
+            lib().tzs_sorter_close(self.h)
+            self.h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def upload_records(data: bytes, off, klen, part=None):
+    """Upload host columnar records to the device; returns device pointers
+    (caller frees via free_device).  off: list/array u64 [n+1]; klen u32 [n]."""
+    import numpy as np
+    L = lib()
+    n = len(klen)
+    d = ctypes.c_void_p()
+    o = ctypes.c_void_p()
+    k = ctypes.c_void_p()
+    p = ctypes.c_void_p()
+    data_n = max(len(data), 1)
+    _ck(L.tzs_malloc_device(data_n, ctypes.byref(d)), "malloc")
+    _ck(L.tzs_malloc_device(8 * (n + 1), ctypes.byref(o)), "malloc")
+    _ck(L.tzs_malloc_device(max(4 * n, 4), ctypes.byref(k)), "malloc")
+    if data:
+        _ck(L.tzs_memcpy_h2d(d, data, len(data)), "h2d")
+    off_a = np.asarray(off, dtype=np.uint64)
+    klen_a = np.asarray(klen, dtype=np.uint32)
+    _ck(L.tzs_memcpy_h2d(o, off_a.ctypes.data, 8 * (n + 1)), "h2d")
+    if n:
+        _ck(L.tzs_memcpy_h2d(k, klen_a.ctypes.data, 4 * n), "h2d")
+    if part is not None:
+        part_a = np.asarray(part, dtype=np.int32)
+        _ck(L.tzs_malloc_device(max(4 * n, 4), ctypes.byref(p)), "malloc")
+        if n:
+            _ck(L.tzs_memcpy_h2d(p, part_a.ctypes.data, 4 * n), "h2d")
+        return d, o, k, p
+    return d, o, k, None
+
+
+def generate(seed, n, kind, klen, vlen, conf):
+    L = lib()
+    d = ctypes.c_void_p()
+    o = ctypes.c_void_p()
+    k = ctypes.c_void_p()
+    p = ctypes.c_void_p()
+    _ck(L.tzs_generate(seed, n, kind, klen, vlen, ctypes.byref(conf), ctypes.byref(d),
+                       ctypes.byref(o), ctypes.byref(k), ctypes.byref(p)), "generate")
+    return d, o, k, p
+
+
+def free_device(*ptrs):
+    for p in ptrs:
+        if p:
+            lib().tzs_free_device(p)
+
+
+def device_available():
+    try:
+        return bool(lib().tzs_device_available())
+    except RuntimeError:
+        return False

@@ -1,0 +1,55 @@
+"""TezRuntimeConfiguration key mapping (same key names drive the engine —
+SURVEY §2 'TezRuntimeConfiguration' row; defaults from
+tez-runtime-library/.../api/TezRuntimeConfiguration.java):
+  tez.runtime.io.sort.mb = 100                (:117)
+  tez.runtime.io.sort.factor = 100            (:106)
+  tez.runtime.key.class / value.class
+  tez.runtime.key.comparator.class
+  tez.runtime.empty.partitions.info-via-events.enabled = true
+  tez.runtime.enable.final-merge.in.output = true
+  tez.runtime.sorter.class = PIPELINED        (:170; both sorters map to the
+                                               one GPU engine — SURVEY §2)
+"""
+from . import _engine
+
+
+_KEY_CLASSES = {
+    "org.apache.hadoop.io.BytesWritable": (_engine.KEY_BYTES, _engine.CMP_TEZBYTES),
+    "org.apache.hadoop.io.Text": (_engine.KEY_TEXT, _engine.CMP_TEXT),
+}
+
+_COMPARATORS = {
+    "org.apache.tez.runtime.library.common.comparator.TezBytesComparator":
+        _engine.CMP_TEZBYTES,
+    "org.apache.hadoop.io.Text$Comparator": _engine.CMP_TEXT,
+}
+
+
+def conf_from_tez_properties(props: dict, num_partitions: int):
+    """Build a TzsConf from a flat tez.runtime.* key/value dict (the same keys
+    an OrderedPartitionedKVEdgeConfig would carry)."""
+    def b(key, dflt):
+        v = props.get(key)
+        if v is None:
+            return dflt
+        return str(v).lower() in ("1", "true", "yes")
+
+    key_cls = props.get("tez.runtime.key.class",
+                        "org.apache.hadoop.io.BytesWritable")
+    if key_cls not in _KEY_CLASSES:
+        raise ValueError(f"unsupported key class {key_cls} (round-1 engine "
+                         "supports BytesWritable and Text)")
+    key_type, cmp_default = _KEY_CLASSES[key_cls]
+    cmp_cls = props.get("tez.runtime.key.comparator.class")
+    comparator = _COMPARATORS.get(cmp_cls, cmp_default) if cmp_cls else cmp_default
+
+    return _engine.make_conf(
+        num_partitions,
+        key_type=key_type,
+        comparator=comparator,
+        send_empty_partition_details=int(
+            b("tez.runtime.empty.partitions.info-via-events.enabled", True)),
+        io_sort_factor=int(props.get("tez.runtime.io.sort.factor", 100)),
+        final_merge_enabled=int(b("tez.runtime.enable.final-merge.in.output", True)),
+        sort_buffer_bytes=int(props.get("tez.runtime.io.sort.mb", 100)) << 20,
+    )
