@@ -1,0 +1,216 @@
+#!/usr/bin/env python3
+"""bench.py — BASELINE.json metric: shuffled+sorted KV bytes/sec.
+
+Workload (config.workload): BASELINE.json configs[1] (C2) — the largest
+single-GPU config the metric is quoted on: 1e8 records, 16B random unique key
++ 64B value (BytesWritable serialization), 64 partitions, HashPartitioner +
+TezBytesComparator.  One step = the full reference-equivalent pass over one
+batch: absorb (device-resident input) -> stable radix sort -> IFile emit +
+CRC -> spill index; N>1 adds the RCCL all-to-all-v exchange + reduce-side
+merge (partitions sharded p % N, weak scaling).  Inputs are generated on
+device BEFORE the timed region; no disk I/O in the timed region (spills are
+HBM-resident — DESIGN.md §4).
+
+value = whole-job Σ serialized key+value bytes (the reference's
+TaskCounter.OUTPUT_BYTES semantics, PipelinedSorter.java:466) ÷ max-over-ranks
+wall time per step.  The CPU oracle (kind "port") is timed on a bounded
+sample as cpu_baseline; roofline reports the dominant kernel (radix scatter)
+against the 8 TB/s gfx950 HBM peak.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+# C2 workload constants
+KLEN, VLEN, PARTS = 16, 64, 64
+REC_SER = 4 + KLEN + 4 + VLEN  # serialized record bytes (BytesWritable k+v)
+SEED = 0x7E2C2
+
+
+def run_step_single(tez_amd, conf, d, off, kl, n):
+    s = tez_amd.Sorter(conf)
+    s.write_batch_device(d, off, kl, None, n)
+    s.flush()
+    ctr = s.counters()
+    tms = s.times()
+    s.close()
+    return ctr, tms
+
+
+def run_step_multi(tez_amd, rank, world, device, d, off, kl, n):
+    from tez_amd import exchange as ex
+    conf = tez_amd.make_conf(PARTS, world_size=world, rank=rank)
+    m = tez_amd.Sorter(conf)
+    m.write_batch_device(d, off, kl, None, n)
+    m.flush()
+    ctr = m.counters()
+    d_data, d_off2, d_klen2, rec_ranges, byte_ranges = m.sorted_columnar()
+    plan = ex.plan_send(rec_ranges, byte_ranges, world)
+    sd, sr, sk = ex.pack_send_tensors(m, plan, device)
+    rd, rrl, rkl = ex.exchange(plan, sd, sr, sk)
+    red = ex.reduce_merge(lambda: tez_amd.Sorter(tez_amd.make_conf(PARTS)), rd, rrl, rkl)
+    tms = red.times()
+    red.close()
+    m.close()
+    return ctr, tms
+
+
+def cpu_baseline_line(sample_records=1_000_000):
+    """Oracle (CPU restatement) timed on the same workload shape, bounded
+    sample (~10-30 s)."""
+    import numpy as np
+    import oracle as o
+    rng = np.random.default_rng(SEED)
+    n = sample_records
+    data = np.zeros(n * REC_SER, dtype=np.uint8)
+    view = data.reshape(n, REC_SER)
+    view[:, 0:4] = np.frombuffer(np.int32(KLEN).byteswap().tobytes(), dtype=np.uint8)
+    view[:, 4:4 + KLEN] = rng.integers(0, 256, size=(n, KLEN), dtype=np.uint8)
+    # force uniqueness like the device generator: mix the record id
+    ids = np.arange(n, dtype=np.uint32).view(np.uint8).reshape(n, 4)
+    view[:, 8:12] ^= ids
+    view[:, 4 + KLEN:8 + KLEN] = np.frombuffer(
+        np.int32(VLEN).byteswap().tobytes(), dtype=np.uint8)
+    view[:, 8 + KLEN:] = rng.integers(0, 256, size=(n, VLEN), dtype=np.uint8)
+    offs = np.arange(0, REC_SER * (n + 1), REC_SER, dtype=np.uint64)
+    klens = np.full(n, 4 + KLEN, dtype=np.uint32)
+    t0 = time.perf_counter()
+    o.spill(data, offs, klens, PARTS, key_type=o.KEY_BYTES, comparator=o.CMP_TEZBYTES)
+    dt = time.perf_counter() - t0
+    return {
+        "value": n * REC_SER / dt,
+        "unit": "bytes/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"oracle tzo_spill on {n} records of the C2 shape "
+                  f"({dt:.1f}s, single-threaded qsort restatement)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--records", type=int, default=100_000_000,
+                    help="total records across all ranks (C2 default 1e8)")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--traffic-bytes", type=float, default=None,
+                    help="PMC-measured HBM bytes per scatter launch (from "
+                         "profiles/, rocprofv3 --pmc; null if not passed)")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world)
+
+    import __graft_entry__
+    __graft_entry__.build()
+    import tez_amd
+
+    dist = None
+    device = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        dist = tdist
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        tdist.init_process_group("nccl")
+
+    n_local = args.records // n_gpus
+    conf = tez_amd.make_conf(PARTS)
+    d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local, kind=0,
+                                        klen=KLEN, vlen=VLEN, conf=conf)
+    tez_amd.free_device(part)
+
+    def barrier_sync():
+        if dist:
+            import torch
+            torch.cuda.synchronize()
+            dist.barrier()
+            torch.cuda.synchronize()
+
+    def one_step():
+        if world > 1:
+            return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
+        return run_step_single(tez_amd, conf, d, off, kl, n_local)
+
+    last_ctr = last_tms = None
+    for _ in range(args.warmup):
+        last_ctr, last_tms = one_step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        last_ctr, last_tms = one_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    tez_amd.free_device(d, off, kl)
+
+    total_bytes_per_step = last_ctr["output_bytes"] * n_gpus  # whole-job Σ
+    ms_per_step = elapsed / args.steps * 1e3
+    value = total_bytes_per_step * args.steps / elapsed
+
+    if rank == 0:
+        # roofline: dominant kernel = radix scatter; algorithmic bytes per
+        # element per launch = 24 (read u64 key + u32 idx, write both)
+        dk_ns = max(last_tms["dominant_kernel_ns"], 1)
+        dk_elems = last_tms["dominant_kernel_elems"]
+        achieved_gbps = 24.0 * dk_elems / dk_ns  # bytes/ns == GB/s
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved_gbps, 1),
+            "peak": 8000.0,
+            "unit": "GB/s",
+            "frac": round(achieved_gbps / 8000.0, 4),
+            "traffic": args.traffic_bytes,
+        }
+        cpu = None
+        if not args.skip_cpu_baseline and world == 1:
+            cpu = cpu_baseline_line()
+        out = {
+            "metric": "shuffled+sorted KV bytes/sec",
+            "value": round(value, 1),
+            "unit": "bytes/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": "C2: 1e8 rec x (16B unique key + 64B val), 64 partitions,"
+                            " BytesWritable/TezBytesComparator, ordered shuffle",
+                "records": args.records,
+                "key_bytes": KLEN,
+                "value_bytes": VLEN,
+                "partitions": PARTS,
+                "parallelism": f"shuffle-shard p%%{n_gpus}" if n_gpus > 1 else "single",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+            "phase_ms": {k: round(v / 1e6, 2) for k, v in last_tms.items()
+                         if k.endswith("_ns")},
+        }
+        print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

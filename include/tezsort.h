@@ -113,6 +113,15 @@ int tzs_sorter_spill_output(tzs_sorter* s, int32_t spill_id,
  * <dir>/output/<unique_id>_<spill>/file.out[.index] when final merge is off. */
 int tzs_sorter_write_files(tzs_sorter* s, const char* local_dir, const char* unique_id);
 
+/* Final sort as permuted columnar arrays — the xGMI exchange wire
+ * (DESIGN.md §4): records in sorted order, partition p's records at
+ * rec_ranges[p]..rec_ranges[p+1] / bytes byte_ranges[p]..byte_ranges[p+1].
+ * Pointers are device memory owned by the sorter (valid until close).
+ * rec_ranges/byte_ranges are caller host arrays of num_partitions+1. */
+int tzs_sorter_sorted_columnar(tzs_sorter* s, const void** d_data,
+                               const uint64_t** d_off, const uint32_t** d_klen,
+                               uint64_t* rec_ranges, uint64_t* byte_ranges);
+
 /* Counters mirrored from TaskCounter semantics (ExternalSorter.java:217-225). */
 typedef struct tzs_counters {
   int64_t output_records;   /* OUTPUT_RECORDS */
@@ -160,7 +169,7 @@ void tzs_free_device(void* d_ptr);
 /* ---- introspection ------------------------------------------------------ */
 /* Per-phase HIP-event times of the last flush, nanoseconds. */
 typedef struct tzs_times {
-  int64_t absorb_ns, composite_ns, sort_ns, permute_ns, emit_ns, crc_ns, merge_ns, total_ns;
+  int64_t absorb_ns, composite_ns, sort_ns, permute_ns, emit_ns, crc_ns, dominant_kernel_elems, total_ns;
   int64_t sort_passes;        /* radix passes actually run (incl. refinement) */
   int64_t dominant_kernel_ns; /* scatter total */
 } tzs_times;

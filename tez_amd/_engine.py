@@ -49,7 +49,7 @@ class TzsCounters(ctypes.Structure):
 class TzsTimes(ctypes.Structure):
     _fields_ = [(n, ctypes.c_int64) for n in
                 ("absorb_ns", "composite_ns", "sort_ns", "permute_ns", "emit_ns",
-                 "crc_ns", "merge_ns", "total_ns", "sort_passes",
+                 "crc_ns", "dominant_kernel_elems", "total_ns", "sort_passes",
                  "dominant_kernel_ns")]
 
 
@@ -82,6 +82,12 @@ def lib():
     L.tzs_sorter_spill_output.argtypes = [c.c_void_p, c.c_int32, c.POINTER(c.c_void_p),
                                           c.POINTER(c.c_int64), c.POINTER(TzsIndexRecord)]
     L.tzs_sorter_write_files.argtypes = [c.c_void_p, c.c_char_p, c.c_char_p]
+    L.tzs_sorter_sorted_columnar.argtypes = [c.c_void_p, c.POINTER(c.c_void_p),
+                                             c.POINTER(c.c_void_p),
+                                             c.POINTER(c.c_void_p),
+                                             c.POINTER(c.c_uint64),
+                                             c.POINTER(c.c_uint64)]
+    L.tzs_memcpy_d2d.argtypes = [c.c_void_p, c.c_void_p, c.c_uint64]
     L.tzs_sorter_counters.argtypes = [c.c_void_p, c.POINTER(TzsCounters)]
     L.tzs_sorter_times.argtypes = [c.c_void_p, c.POINTER(TzsTimes)]
     L.tzs_sorter_close.argtypes = [c.c_void_p]
@@ -167,6 +173,21 @@ class Sorter:
             ba = (ctypes.c_char * n.value).from_buffer(buf)
             _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba), p, n.value), "d2h")
         return bytes(buf), [(r.start_offset, r.raw_length, r.part_length) for r in idx]
+
+    def sorted_columnar(self):
+        """Final sort as device columnar arrays + partition ranges
+        (the exchange wire — DESIGN.md §4).  Returns
+        (d_data, d_off, d_klen, rec_ranges: list, byte_ranges: list)."""
+        P = self.conf.num_partitions
+        d = ctypes.c_void_p()
+        o = ctypes.c_void_p()
+        k = ctypes.c_void_p()
+        rr = (ctypes.c_uint64 * (P + 1))()
+        br = (ctypes.c_uint64 * (P + 1))()
+        _ck(lib().tzs_sorter_sorted_columnar(self.h, ctypes.byref(d), ctypes.byref(o),
+                                             ctypes.byref(k), rr, br),
+            "sorted_columnar")
+        return d, o, k, list(rr), list(br)
 
     def write_files(self, local_dir: str, unique_id: str):
         _ck(lib().tzs_sorter_write_files(self.h, local_dir.encode(), unique_id.encode()),

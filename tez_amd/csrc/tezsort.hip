@@ -553,6 +553,31 @@ __global__ void k_emit_records(RecTable rt, const uint32_t* sidx, const uint8_t*
   }
 }
 
+/* permuted-columnar materialization (exchange wire — DESIGN.md §4):
+ * records gathered into sorted order as (data, off, klen) so partition
+ * ranges are contiguous per destination rank. */
+__global__ void k_sorted_reclens(RecTable rt, const uint32_t* sidx, uint64_t* lens,
+                                 uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    RecView v = rt_view(rt, sidx[i]);
+    lens[i] = (uint64_t)v.klen + v.vlen;
+  }
+}
+__global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
+                                  const uint64_t* out_off, uint8_t* out_data,
+                                  uint32_t* out_klen, uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    RecView v = rt_view(rt, sidx[i]);
+    uint8_t* w = out_data + out_off[i];
+    for (uint32_t b = 0; b < v.klen; b++) w[b] = v.key[b];
+    w += v.klen;
+    for (uint32_t b = 0; b < v.vlen; b++) w[b] = v.val[b];
+    out_klen[i] = v.klen;
+  }
+}
+
 /* ---- CRC over emitted segments ----
  * chunk kernel: thread computes CRC32 of one 256-byte chunk of one partition's
  * checksummed range (payload .. EOF); combine kernel: one wave per partition
@@ -764,6 +789,10 @@ static int scan_u64(const uint64_t* d_in, uint64_t* d_out, uint32_t n, uint64_t*
   return 0;
 }
 
+static thread_local int64_t g_scatter_ns = 0;
+static thread_local int64_t g_scatter_launches = 0;
+static thread_local int64_t g_scatter_elems = 0;
+
 /* stable LSD radix over KeyT with payload arrays. Sorts in place (ping-pong,
  * result left in the primary arrays). */
 template <typename KeyT>
@@ -782,6 +811,8 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
   bool has_a1 = d_a1 != nullptr;
   if (has_a1 && ta1.alloc(sizeof(uint32_t) * n)) return -12;
 
+  hipEvent_t evs[16], eve[16];
+  int nev = 0;
   KeyT* kin = d_key;
   KeyT* kout = (KeyT*)tk.p;
   uint32_t* a0in = d_a0;
@@ -796,6 +827,8 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                        (uint32_t*)counts.p, nb, (uint32_t*)offsets.p, (uint32_t*)totals.p);
     hipLaunchKernelGGL(k_radix_scan_digits, dim3(1), dim3(RADIX), 0, 0,
                        (uint32_t*)totals.p, (uint32_t*)bases.p);
+    if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
+                    (void)hipEventRecord(evs[nev]); }
     if (has_a1)
       hipLaunchKernelGGL((k_radix_scatter<KeyT, true>), dim3(nb), dim3(BLOCK), 0, 0,
                          kin, kout, a0in, a0out, a1in, a1out, n, b,
@@ -804,10 +837,20 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
       hipLaunchKernelGGL((k_radix_scatter<KeyT, false>), dim3(nb), dim3(BLOCK), 0, 0,
                          kin, kout, a0in, a0out, nullptr, nullptr, n, b,
                          (uint32_t*)offsets.p, (uint32_t*)bases.p);
+    if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
     std::swap(kin, kout);
     std::swap(a0in, a0out);
     if (has_a1) std::swap(a1in, a1out);
     passes++;
+  }
+  (void)hipDeviceSynchronize();
+  for (int e = 0; e < nev; e++) {
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, evs[e], eve[e]);
+    g_scatter_ns += (int64_t)(ms * 1e6);
+    g_scatter_launches += 1;
+    g_scatter_elems += n;
+    (void)hipEventDestroy(evs[e]); (void)hipEventDestroy(eve[e]);
   }
   if (passes & 1) {
     /* result is in the temp arrays: copy back */
@@ -867,6 +910,11 @@ struct tzs_sorter {
   tzs_times times = {};
   /* scratch kept across calls */
   DBuf skey, sidx, eq, same, sizes, scan, parts_sorted;
+  /* final-sort metadata for the exchange path */
+  std::vector<uint64_t> final_rec_ranges;  /* [P+1] record index ranges */
+  RecTable final_rt = {};
+  uint32_t final_n = 0;
+  DBuf col_data, col_off, col_klen;        /* permuted columnar view */
 };
 
 extern "C" void tzs_conf_default(tzs_conf* c, int32_t num_partitions) {
@@ -1012,6 +1060,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                          const uint8_t* h_spill_rle, int nspills_rle,
                          SpillData* outsp) {
   tzs_times& T = s->times;
+  g_scatter_ns = 0; g_scatter_launches = 0; g_scatter_elems = 0;
   hipEvent_t ev[10];
   for (auto& e : ev) (void)hipEventCreate(&e);
   (void)hipEventRecord(ev[0]);
@@ -1030,7 +1079,6 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   /* 2. base radix sort over the full u64 composite */
   int rc = radix_sort<uint64_t>(d_key, d_idx, nullptr, n, 8);
   if (rc) return rc;
-  T.sort_passes += 8;
   (void)hipEventRecord(ev[2]);
 
   /* 3. refinement levels */
@@ -1104,7 +1152,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     if (rc) return rc;
     rc = radix_sort<uint32_t>((uint32_t*)seg.p, (uint32_t*)pos.p, nullptr, m, segbytes);
     if (rc) return rc;
-    T.sort_passes += 8 + segbytes;
+
     /* wait: after the seg sort, lkey is NOT permuted alongside (we passed pos
        as aux of seg sort, losing lkey alignment).  Redo: sort with lkey as
        aux too.  radix_sort<uint32_t> carries only u32 payloads — carry lkey
@@ -1171,6 +1219,9 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   }
   std::vector<uint64_t> h_prec_start(P + 1, 0);
   for (int p = 0; p < P; p++) h_prec_start[p + 1] = h_prec_start[p] + h_pcount[p];
+  s->final_rec_ranges = h_prec_start;
+  s->final_rt = rt;
+  s->final_n = n;
   /* body bytes per partition = scan[start of next] - scan[start] */
   std::vector<uint64_t> h_scan_at(P + 1, 0);
   {
@@ -1288,6 +1339,9 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   (void)hipEventElapsedTime(&ms, ev[4], ev[5]); T.emit_ns += (int64_t)(ms * 1e6);
   (void)hipEventElapsedTime(&ms, ev[5], ev[6]); T.crc_ns += (int64_t)(ms * 1e6);
   (void)hipEventElapsedTime(&ms, ev[0], ev[6]); T.total_ns += (int64_t)(ms * 1e6);
+  T.dominant_kernel_ns += g_scatter_ns;
+  T.sort_passes += g_scatter_launches;  /* scatter launch count (roofline) */
+  T.dominant_kernel_elems += g_scatter_elems;
   for (auto& e : ev) (void)hipEventDestroy(e);
   return 0;
 }
@@ -1521,6 +1575,50 @@ extern "C" void tzs_sorter_close(tzs_sorter* s) {
   delete s;
 }
 
+/* Materialize the final sort as permuted columnar arrays (exchange wire).
+ * Returns device pointers owned by the sorter and host record ranges
+ * rec_ranges[P+1] / byte_ranges[P+1]. */
+extern "C" int tzs_sorter_sorted_columnar(tzs_sorter* s,
+                                          const void** d_data, const uint64_t** d_off,
+                                          const uint32_t** d_klen,
+                                          uint64_t* rec_ranges, uint64_t* byte_ranges) {
+  if (!s->flushed || s->final_n == 0) {
+    if (!s->flushed) FAIL(-22, "flush first");
+    if (d_data) *d_data = nullptr;
+    int P = s->conf.num_partitions;
+    for (int p = 0; p <= P; p++) { rec_ranges[p] = 0; byte_ranges[p] = 0; }
+    return 0;
+  }
+  uint32_t n = s->final_n;
+  static thread_local DBuf lens;
+  if (lens.alloc(8ull * n)) return -12;
+  hipLaunchKernelGGL(k_sorted_reclens, dim3(grid1d(n)), dim3(BLOCK), 0, 0, s->final_rt,
+                     (const uint32_t*)s->sidx.p, (uint64_t*)lens.p, n);
+  if (s->col_off.alloc(8ull * (n + 1))) return -12;
+  uint64_t total = 0;
+  if (scan_u64((uint64_t*)lens.p, (uint64_t*)s->col_off.p, n, &total)) return -12;
+  HIP_CHECK(hipMemcpy((uint64_t*)s->col_off.p + n, &total, 8, hipMemcpyHostToDevice));
+  if (s->col_data.alloc(total ? total : 1)) return -12;
+  if (s->col_klen.alloc(4ull * n)) return -12;
+  hipLaunchKernelGGL(k_permute_records, dim3(grid1d(n)), dim3(BLOCK), 0, 0, s->final_rt,
+                     (const uint32_t*)s->sidx.p, (const uint64_t*)s->col_off.p,
+                     (uint8_t*)s->col_data.p, (uint32_t*)s->col_klen.p, n);
+  HIP_CHECK(hipDeviceSynchronize());
+  int P = s->conf.num_partitions;
+  for (int p = 0; p <= P; p++) {
+    uint64_t r = s->final_rec_ranges[p];
+    rec_ranges[p] = r;
+    uint64_t b = total;
+    if (r < n)
+      HIP_CHECK(hipMemcpy(&b, (uint64_t*)s->col_off.p + r, 8, hipMemcpyDeviceToHost));
+    byte_ranges[p] = b;
+  }
+  if (d_data) *d_data = s->col_data.p;
+  if (d_off) *d_off = (const uint64_t*)s->col_off.p;
+  if (d_klen) *d_klen = (const uint32_t*)s->col_klen.p;
+  return 0;
+}
+
 /* ---- synthetic generation ---- */
 extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen,
                             int32_t vlen, const tzs_conf* conf, void** d_data,
@@ -1572,6 +1670,10 @@ extern "C" int tzs_merge_segments(const tzs_conf* conf, const tzs_segment* segs,
 /* ---- misc helpers for the Python layer ---- */
 extern "C" int tzs_memcpy_d2h(void* host, const void* dev, uint64_t n) {
   HIP_CHECK(hipMemcpy(host, dev, n, hipMemcpyDeviceToHost));
+  return 0;
+}
+extern "C" int tzs_memcpy_d2d(void* dst, const void* src, uint64_t n) {
+  HIP_CHECK(hipMemcpy(dst, src, n, hipMemcpyDeviceToDevice));
   return 0;
 }
 extern "C" int tzs_memcpy_h2d(void* dev, const void* host, uint64_t n) {

@@ -1,0 +1,131 @@
+"""Intra-node shuffle exchange: RCCL all-to-all-v over xGMI, replacing the
+reference's HTTP fetch (Shuffle/ShuffleScheduler/FetcherOrderedGrouped —
+SURVEY §3b/§8e).  Partition p is owned by rank p % world_size; the per-pair
+byte counts come from the spill index triples (the ShuffleHeader-equivalent
+metadata, ShuffleHeader.java:82-106), exchanged first as a size matrix
+(all_gather), then torch.distributed.all_to_all_single moves the bytes —
+grouped send/recv over the 7 pairwise xGMI links.
+
+Planning functions are torch-free and covered by CPU (gloo) tests; the tensor
+movement uses whatever backend the process group has (nccl=RCCL on the GPU
+box, gloo in CPU tests)."""
+from dataclasses import dataclass
+from typing import List
+
+
+def owner_of(partition: int, world: int) -> int:
+    return partition % world
+
+
+def parts_for_dest(P: int, world: int, dest: int) -> List[int]:
+    return [p for p in range(P) if p % world == dest]
+
+
+@dataclass
+class SendPlan:
+    """Partition segments ordered by (dest rank, partition asc)."""
+    order: List[int]          # partition ids in send order
+    byte_splits: List[int]    # per-dest byte counts
+    rec_splits: List[int]     # per-dest record counts
+    seg_bytes: List[int]      # per ordered segment
+    seg_recs: List[int]
+
+
+def plan_send(rec_ranges, byte_ranges, world: int) -> SendPlan:
+    """rec_ranges/byte_ranges: [P+1] prefix arrays from
+    tzs_sorter_sorted_columnar (partition p = [r[p], r[p+1]))."""
+    P = len(rec_ranges) - 1
+    order, seg_bytes, seg_recs = [], [], []
+    byte_splits, rec_splits = [], []
+    for d in range(world):
+        b = r = 0
+        for p in parts_for_dest(P, world, d):
+            order.append(p)
+            sb = int(byte_ranges[p + 1] - byte_ranges[p])
+            sr = int(rec_ranges[p + 1] - rec_ranges[p])
+            seg_bytes.append(sb)
+            seg_recs.append(sr)
+            b += sb
+            r += sr
+        byte_splits.append(b)
+        rec_splits.append(r)
+    return SendPlan(order, byte_splits, rec_splits, seg_bytes, seg_recs)
+
+
+def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None):
+    """all-to-all-v of (data bytes, per-record lengths, per-record klens).
+    send_* are torch tensors already laid out in plan order (uint8/u32/u32).
+    Returns (recv_data, recv_reclen, recv_klen) tensors."""
+    import torch
+    import torch.distributed as dist
+    world = dist.get_world_size(group)
+    in_b = plan.byte_splits
+    in_r = plan.rec_splits
+    # exchange split sizes first (the size-matrix all-gather; the 24B index
+    # triples' byte counts — SURVEY §8e)
+    my_sizes = torch.tensor([in_b, in_r], dtype=torch.int64,
+                            device=send_data.device)
+    all_sizes = [torch.empty_like(my_sizes) for _ in range(world)]
+    dist.all_gather(all_sizes, my_sizes, group=group)
+    out_b = [int(all_sizes[src][0][dist.get_rank(group)]) for src in range(world)]
+    out_r = [int(all_sizes[src][1][dist.get_rank(group)]) for src in range(world)]
+    recv_data = torch.empty(sum(out_b), dtype=torch.uint8, device=send_data.device)
+    recv_reclen = torch.empty(sum(out_r), dtype=torch.int32, device=send_data.device)
+    recv_klen = torch.empty(sum(out_r), dtype=torch.int32, device=send_data.device)
+    dist.all_to_all_single(recv_data, send_data, out_b, in_b, group=group)
+    dist.all_to_all_single(recv_reclen, send_reclen, out_r, in_r, group=group)
+    dist.all_to_all_single(recv_klen, send_klen, out_r, in_r, group=group)
+    return recv_data, recv_reclen, recv_klen
+
+
+def pack_send_tensors(sorter, plan: SendPlan, device):
+    """Build the send tensors from a flushed sorter's columnar view
+    (device-to-device copies per partition segment)."""
+    import torch
+    from ._engine import lib, _ck
+    d_data, d_off, d_klen, rec_ranges, byte_ranges = sorter.sorted_columnar()
+    total_b = sum(plan.byte_splits)
+    total_r = sum(plan.rec_splits)
+    send_data = torch.empty(max(total_b, 1), dtype=torch.uint8, device=device)
+    send_reclen = torch.empty(max(total_r, 1), dtype=torch.int32, device=device)
+    send_klen = torch.empty(max(total_r, 1), dtype=torch.int32, device=device)
+    # per-record lengths from the off array: copy off into a torch u64 tensor
+    n = rec_ranges[-1]
+    off_t = torch.empty(n + 1, dtype=torch.int64, device=device)
+    if n >= 0 and d_off.value:
+        _ck(lib().tzs_memcpy_d2d(off_t.data_ptr(), d_off, 8 * (n + 1)), "d2d")
+    reclen_all = (off_t[1:] - off_t[:-1]).to(torch.int32) if n > 0 else off_t.to(torch.int32)[:0]
+    klen_all = torch.empty(max(n, 1), dtype=torch.int32, device=device)
+    if n > 0:
+        _ck(lib().tzs_memcpy_d2d(klen_all.data_ptr(), d_klen, 4 * n), "d2d")
+    bpos = rpos = 0
+    for i, p in enumerate(plan.order):
+        sb, sr = plan.seg_bytes[i], plan.seg_recs[i]
+        if sb:
+            _ck(lib().tzs_memcpy_d2d(send_data.data_ptr() + bpos,
+                                     d_data.value + int(byte_ranges[p]), sb), "d2d")
+        if sr:
+            r0 = int(rec_ranges[p])
+            send_reclen[rpos:rpos + sr] = reclen_all[r0:r0 + sr]
+            send_klen[rpos:rpos + sr] = klen_all[r0:r0 + sr]
+        bpos += sb
+        rpos += sr
+    return send_data[:total_b], send_reclen[:total_r], send_klen[:total_r]
+
+
+def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen):
+    """Feed received columnar records into a reduce-side sorter (the
+    MergeManager/TezMerger replacement) and flush: the final merged IFile for
+    this rank's owned partitions."""
+    import torch
+    n = int(recv_reclen.numel())
+    sorter = conf_factory()
+    if n:
+        off = torch.zeros(n + 1, dtype=torch.int64, device=recv_data.device)
+        torch.cumsum(recv_reclen.to(torch.int64), 0, out=off[1:])
+        klen_u32 = recv_klen.contiguous()
+        sorter.write_batch_device(recv_data.data_ptr(), off.data_ptr(),
+                                  klen_u32.data_ptr(), None, n)
+        sorter._exchange_keepalive = (recv_data, off, klen_u32)
+    sorter.flush()
+    return sorter
