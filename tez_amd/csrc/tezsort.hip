@@ -504,12 +504,14 @@ __global__ void k_writer_same(RecTable rt, const uint32_t* sidx, const uint8_t* 
   }
 }
 __global__ void k_emit_sizes(RecTable rt, const uint32_t* sidx, const uint8_t* same,
-                             uint64_t* sizes, uint32_t n) {
+                             const uint32_t* parts, uint64_t* sizes, uint32_t n) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, sidx[i]);
     uint64_t sz;
-    uint8_t prev_same = (i > 0) ? same[i - 1] : 0;
+    /* a run never crosses a partition segment: the previous record's RLE
+       state is invisible to this partition's IFile stream */
+    uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
     if (same[i]) {
       sz = (prev_same ? 0 : 1) /* RLE marker */ + d_vint_size(v.vlen) + v.vlen;
     } else {
@@ -526,30 +528,42 @@ __global__ void k_sorted_parts(const uint64_t* skeys, int pbits, uint32_t* parts
        i += gridDim.x * blockDim.x)
     parts[i] = pbits ? (uint32_t)(skeys[i] >> (64 - pbits)) : 0;
 }
-/* per-record emit: stream_off[i] = seg_payload_start[part] + (scan[i] - part_scan_base[part]) */
+/* per-record emit: stream_off[i] = seg_payload_start[part] + (scan[i] - part_scan_base[part]).
+ * One WAVE per record: lane-parallel payload copy (the serialized key and
+ * value are contiguous in the record, so the payload is one span); lane 0
+ * writes the marker/vint header.  Coalesced on both sides. */
 __global__ void k_emit_records(RecTable rt, const uint32_t* sidx, const uint8_t* same,
                                const uint64_t* scan, const uint32_t* parts,
                                const uint64_t* seg_payload_start,
                                const uint64_t* part_scan_base,
                                uint8_t* out, uint32_t n) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x) {
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (uint32_t i = wave; i < n; i += nwaves) {
     RecView v = rt_view(rt, sidx[i]);
     uint32_t p = parts[i];
     uint8_t* w = out + seg_payload_start[p] + (scan[i] - part_scan_base[p]);
-    uint8_t prev_same = (i > 0) ? same[i - 1] : 0;
+    uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
+    const uint8_t* src;
+    uint32_t len;
+    uint32_t hdr = 0;
+    uint8_t hdrbuf[12];
     if (same[i]) {
-      if (!prev_same) *w++ = 0xFE; /* RLE_MARKER -2 */
-      w += d_vint_write(w, v.vlen);
-      for (uint32_t b = 0; b < v.vlen; b++) w[b] = v.val[b];
+      if (!prev_same) hdrbuf[hdr++] = 0xFE; /* RLE_MARKER -2 */
+      hdr += d_vint_write(hdrbuf + hdr, v.vlen);
+      src = v.val;
+      len = v.vlen;
     } else {
-      if (prev_same) *w++ = 0xFD; /* V_END_MARKER -3 */
-      w += d_vint_write(w, v.klen);
-      w += d_vint_write(w, v.vlen);
-      for (uint32_t b = 0; b < v.klen; b++) w[b] = v.key[b];
-      w += v.klen;
-      for (uint32_t b = 0; b < v.vlen; b++) w[b] = v.val[b];
+      if (prev_same) hdrbuf[hdr++] = 0xFD; /* V_END_MARKER -3 */
+      hdr += d_vint_write(hdrbuf + hdr, v.klen);
+      hdr += d_vint_write(hdrbuf + hdr, v.vlen);
+      src = v.key;                /* key ‖ val are contiguous in the record */
+      len = v.klen + v.vlen;
     }
+    if (lane < hdr) w[lane] = hdrbuf[lane];
+    w += hdr;
+    for (uint32_t b = lane; b < len; b += WAVE) w[b] = src[b];
   }
 }
 
@@ -567,14 +581,15 @@ __global__ void k_sorted_reclens(RecTable rt, const uint32_t* sidx, uint64_t* le
 __global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
                                   const uint64_t* out_off, uint8_t* out_data,
                                   uint32_t* out_klen, uint32_t n) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x) {
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (uint32_t i = wave; i < n; i += nwaves) {
     RecView v = rt_view(rt, sidx[i]);
     uint8_t* w = out_data + out_off[i];
-    for (uint32_t b = 0; b < v.klen; b++) w[b] = v.key[b];
-    w += v.klen;
-    for (uint32_t b = 0; b < v.vlen; b++) w[b] = v.val[b];
-    out_klen[i] = v.klen;
+    uint32_t len = v.klen + v.vlen;  /* key ‖ val contiguous */
+    for (uint32_t b = lane; b < len; b += WAVE) w[b] = v.key[b];
+    if (lane == 0) out_klen[i] = v.klen;
   }
 }
 
@@ -624,22 +639,70 @@ __global__ void k_crc_chunks(const uint8_t* stream, const uint64_t* range_start,
   }
 }
 
-__global__ void k_crc_combine(const uint64_t* range_len, const uint64_t* chunk_base,
-                              const uint32_t* chunk_crc, uint32_t nparts,
-                              uint32_t* part_crc) {
+/* Tree combine, two levels.  Group = up to 2048 chunks (512 KB of payload):
+ * thread t serially folds its 8 consecutive chunks, then an LDS tree folds
+ * the 256 threads; a final kernel folds each partition's groups.  Every fold
+ * is combine(L,R) = shift(crc_L, len_R) ^ crc_R with the generic
+ * bit-decomposed shift so ragged tails are exact. */
+#define CRC_GROUP_CHUNKS 2048
+__global__ void k_crc_combine_groups(const uint64_t* range_len, const uint64_t* chunk_base,
+                                     const uint64_t* group_base /* [P+1] */,
+                                     const uint32_t* chunk_crc, uint32_t nparts,
+                                     uint32_t total_groups,
+                                     uint32_t* group_crc, uint64_t* group_len) {
+  __shared__ uint32_t s_crc[BLOCK];
+  __shared__ uint64_t s_len[BLOCK];
+  for (uint32_t g = blockIdx.x; g < total_groups; g += gridDim.x) {
+    /* find partition via binary search over group_base */
+    uint32_t lo = 0, hi = nparts;
+    while (lo + 1 < hi) {
+      uint32_t mid = (lo + hi) / 2;
+      if (group_base[mid] <= g) lo = mid; else hi = mid;
+    }
+    uint32_t p = lo;
+    uint64_t glocal = g - group_base[p];
+    uint64_t len = range_len[p];
+    uint64_t nchunks = (len + CRC_CHUNK - 1) / CRC_CHUNK;
+    uint64_t c0 = glocal * CRC_GROUP_CHUNKS;
+    uint64_t cend = min(c0 + (uint64_t)CRC_GROUP_CHUNKS, nchunks);
+    /* thread-serial fold of 8 consecutive chunks */
+    uint64_t t0 = c0 + (uint64_t)threadIdx.x * 8;
+    uint32_t crc = 0;
+    uint64_t mylen = 0;
+    for (uint64_t c = t0; c < min(t0 + 8, cend); c++) {
+      uint64_t clen = len - c * CRC_CHUNK;
+      if (clen > CRC_CHUNK) clen = CRC_CHUNK;
+      crc = d_crc_shift(crc, clen) ^ chunk_crc[chunk_base[p] + c];
+      mylen += clen;
+    }
+    s_crc[threadIdx.x] = crc;
+    s_len[threadIdx.x] = mylen;
+    __syncthreads();
+    /* LDS tree: combine thread t with t+stride (left=t, right=t+stride) */
+    for (int stride = BLOCK / 2; stride > 0; stride >>= 1) {
+      if ((int)threadIdx.x < stride) {
+        uint64_t rl = s_len[threadIdx.x + stride];
+        if (rl) {
+          s_crc[threadIdx.x] = d_crc_shift(s_crc[threadIdx.x], rl)
+                               ^ s_crc[threadIdx.x + stride];
+          s_len[threadIdx.x] += rl;
+        }
+      }
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) { group_crc[g] = s_crc[0]; group_len[g] = s_len[0]; }
+    __syncthreads();
+  }
+}
+__global__ void k_crc_combine_final(const uint64_t* group_base, const uint32_t* group_crc,
+                                    const uint64_t* group_len, uint32_t nparts,
+                                    uint32_t* part_crc) {
   uint32_t p = blockIdx.x * blockDim.x + threadIdx.x;
   if (p >= nparts) return;
-  uint64_t len = range_len[p];
-  if (len == 0) { part_crc[p] = 0; return; }
-  uint64_t nchunks = (len + CRC_CHUNK - 1) / CRC_CHUNK;
-  uint64_t b0 = chunk_base[p];
-  uint32_t crc = chunk_crc[b0];
-  uint64_t done = (len < CRC_CHUNK) ? len : CRC_CHUNK;
-  for (uint64_t c = 1; c < nchunks; c++) {
-    uint64_t clen = len - c * CRC_CHUNK;
-    if (clen > CRC_CHUNK) clen = CRC_CHUNK;
-    crc = d_crc_shift(crc, clen) ^ chunk_crc[b0 + c];
-    done += clen;
+  uint32_t crc = 0;
+  for (uint64_t g = group_base[p]; g < group_base[p + 1]; g++) {
+    if (group_len[g])
+      crc = d_crc_shift(crc, group_len[g]) ^ group_crc[g];
   }
   part_crc[p] = crc;
 }
@@ -1197,15 +1260,16 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                      writer_rle, (const uint8_t*)d_sprle.p, (uint8_t*)s->same.p, n);
 
   /* 5. emit sizes + partition layout */
-  if (s->sizes.alloc(sizeof(uint64_t) * n)) return -12;
-  if (s->scan.alloc(sizeof(uint64_t) * n)) return -12;
-  hipLaunchKernelGGL(k_emit_sizes, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
-                     (const uint8_t*)s->same.p, (uint64_t*)s->sizes.p, n);
-  uint64_t total_body = 0;
-  if (scan_u64((uint64_t*)s->sizes.p, (uint64_t*)s->scan.p, n, &total_body)) return -12;
   if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
   hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
                      (uint32_t*)s->parts_sorted.p, n);
+  if (s->sizes.alloc(sizeof(uint64_t) * n)) return -12;
+  if (s->scan.alloc(sizeof(uint64_t) * n)) return -12;
+  hipLaunchKernelGGL(k_emit_sizes, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
+                     (const uint8_t*)s->same.p, (const uint32_t*)s->parts_sorted.p,
+                     (uint64_t*)s->sizes.p, n);
+  uint64_t total_body = 0;
+  if (scan_u64((uint64_t*)s->sizes.p, (uint64_t*)s->scan.p, n, &total_body)) return -12;
   /* partition record ranges: host-side from a partition histogram */
   std::vector<uint32_t> h_pcount(P, 0);
   {
@@ -1321,9 +1385,26 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                        (const uint64_t*)d_rstart.p, (const uint64_t*)d_rlen.p,
                        (const uint64_t*)d_chunkbase.p, P, (uint32_t)total_chunks,
                        (uint32_t*)d_chunkcrc.p);
-  hipLaunchKernelGGL(k_crc_combine, dim3((P + BLOCK - 1) / BLOCK), dim3(BLOCK), 0, 0,
-                     (const uint64_t*)d_rlen.p, (const uint64_t*)d_chunkbase.p,
-                     (const uint32_t*)d_chunkcrc.p, P, (uint32_t*)d_partcrc.p);
+  std::vector<uint64_t> h_groupbase(P + 1, 0);
+  for (int p = 0; p < P; p++) {
+    uint64_t nchunks = h_chunkbase[p + 1] - h_chunkbase[p];
+    h_groupbase[p + 1] = h_groupbase[p] + (nchunks + CRC_GROUP_CHUNKS - 1) / CRC_GROUP_CHUNKS;
+  }
+  uint64_t total_groups = h_groupbase[P];
+  static thread_local DBuf d_groupbase, d_groupcrc, d_grouplen;
+  if (up(d_groupbase, h_groupbase.data(), 8 * (P + 1))) return -12;
+  if (d_groupcrc.alloc(4 * (total_groups ? total_groups : 1))) return -12;
+  if (d_grouplen.alloc(8 * (total_groups ? total_groups : 1))) return -12;
+  if (total_groups)
+    hipLaunchKernelGGL(k_crc_combine_groups,
+                       dim3((uint32_t)min(total_groups, (uint64_t)2048)), dim3(BLOCK), 0, 0,
+                       (const uint64_t*)d_rlen.p, (const uint64_t*)d_chunkbase.p,
+                       (const uint64_t*)d_groupbase.p, (const uint32_t*)d_chunkcrc.p, P,
+                       (uint32_t)total_groups, (uint32_t*)d_groupcrc.p,
+                       (uint64_t*)d_grouplen.p);
+  hipLaunchKernelGGL(k_crc_combine_final, dim3((P + BLOCK - 1) / BLOCK), dim3(BLOCK), 0, 0,
+                     (const uint64_t*)d_groupbase.p, (const uint32_t*)d_groupcrc.p,
+                     (const uint64_t*)d_grouplen.p, P, (uint32_t*)d_partcrc.p);
   /* re-patch to write CRC trailers (header/EOF rewrite is idempotent) */
   hipLaunchKernelGGL(k_patch_segments, dim3((P + BLOCK - 1) / BLOCK), dim3(BLOCK), 0, 0,
                      d_out, (const uint64_t*)d_segstart.p, (const uint64_t*)d_body.p,
