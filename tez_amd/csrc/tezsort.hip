@@ -678,9 +678,10 @@ __global__ void k_crc_combine_groups(const uint64_t* range_len, const uint64_t* 
     s_crc[threadIdx.x] = crc;
     s_len[threadIdx.x] = mylen;
     __syncthreads();
-    /* LDS tree: combine thread t with t+stride (left=t, right=t+stride) */
-    for (int stride = BLOCK / 2; stride > 0; stride >>= 1) {
-      if ((int)threadIdx.x < stride) {
+    /* LDS tree over ADJACENT spans: CRC combine is associative but NOT
+     * commutative, so thread t merges [t, t+stride) with [t+stride, t+2s). */
+    for (int stride = 1; stride < BLOCK; stride <<= 1) {
+      if ((threadIdx.x & (2 * stride - 1)) == 0 && threadIdx.x + stride < BLOCK) {
         uint64_t rl = s_len[threadIdx.x + stride];
         if (rl) {
           s_crc[threadIdx.x] = d_crc_shift(s_crc[threadIdx.x], rl)
@@ -800,22 +801,57 @@ static inline uint32_t grid1d(uint64_t n) {
   if (b == 0) b = 1;
   return (uint32_t)b;
 }
+/* wave-per-record kernels are latency-bound on per-record setup loads:
+ * give them more waves */
+static inline uint32_t grid_waves(uint64_t nrec) {
+  uint64_t b = (nrec + WPB - 1) / WPB; /* one record per wave per iteration */
+  if (b > 8192) b = 8192;
+  if (b == 0) b = 1;
+  return (uint32_t)b;
+}
+
+/* Pooled device allocator: size-class (next power of two) free lists.
+ * hipMalloc/hipFree of multi-GB buffers costs hundreds of ms; the shuffle
+ * engine's buffer sizes recur every spill/step, so pooling removes that
+ * entirely (288 GB HBM makes holding the pool cheap). */
+#include <unordered_map>
+static std::unordered_map<size_t, std::vector<void*>>& pool_map() {
+  static std::unordered_map<size_t, std::vector<void*>> m;
+  return m;
+}
+static size_t pool_class(size_t n) {
+  size_t c = 1 << 16;
+  while (c < n) c <<= 1;
+  return c;
+}
+static int pool_alloc(size_t n, void** out, size_t* cls_out) {
+  size_t cls = pool_class(n);
+  auto& fl = pool_map()[cls];
+  if (!fl.empty()) { *out = fl.back(); fl.pop_back(); *cls_out = cls; return 0; }
+  if (hipMalloc(out, cls) != hipSuccess) {
+    /* under pressure: drop the whole pool and retry once */
+    for (auto& kv : pool_map())
+      for (void* q : kv.second) (void)hipFree(q);
+    pool_map().clear();
+    if (hipMalloc(out, cls) != hipSuccess) {
+      snprintf(g_err, sizeof(g_err), "hipMalloc(%zu) failed", cls);
+      return -12;
+    }
+  }
+  *cls_out = cls;
+  return 0;
+}
+static void pool_free(void* p, size_t cls) { if (p) pool_map()[cls].push_back(p); }
 
 struct DBuf {
   void* p = nullptr;
-  size_t sz = 0;
+  size_t sz = 0;   /* size class */
   int alloc(size_t n) {
     if (n <= sz) return 0;
-    if (p) (void)hipFree(p);
-    p = nullptr; sz = 0;
-    if (hipMalloc(&p, n) != hipSuccess) {
-      snprintf(g_err, sizeof(g_err), "hipMalloc(%zu) failed", n);
-      return -12;
-    }
-    sz = n;
-    return 0;
+    release();
+    return pool_alloc(n, &p, &sz);
   }
-  void release() { if (p) { (void)hipFree(p); p = nullptr; sz = 0; } }
+  void release() { if (p) { pool_free(p, sz); p = nullptr; sz = 0; } }
 };
 
 /* exclusive scan of u64 array (device), returns total via last+add trick */
@@ -1360,7 +1396,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   if (up(d_rlen, h_range_len.data(), 8 * P)) return -12;
 
   /* 6. emit records */
-  hipLaunchKernelGGL(k_emit_records, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
+  hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0, rt, d_idx,
                      (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
                      (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
                      (const uint64_t*)d_scanbase.p, d_out, n);
@@ -1681,7 +1717,7 @@ extern "C" int tzs_sorter_sorted_columnar(tzs_sorter* s,
   HIP_CHECK(hipMemcpy((uint64_t*)s->col_off.p + n, &total, 8, hipMemcpyHostToDevice));
   if (s->col_data.alloc(total ? total : 1)) return -12;
   if (s->col_klen.alloc(4ull * n)) return -12;
-  hipLaunchKernelGGL(k_permute_records, dim3(grid1d(n)), dim3(BLOCK), 0, 0, s->final_rt,
+  hipLaunchKernelGGL(k_permute_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0, s->final_rt,
                      (const uint32_t*)s->sidx.p, (const uint64_t*)s->col_off.p,
                      (uint8_t*)s->col_data.p, (uint32_t*)s->col_klen.p, n);
   HIP_CHECK(hipDeviceSynchronize());
