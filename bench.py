@@ -33,12 +33,21 @@ SEED = 0x7E2C2
 
 
 def run_step_single(tez_amd, conf, d, off, kl, n):
+    t0 = time.perf_counter()
     s = tez_amd.Sorter(conf)
+    t1 = time.perf_counter()
     s.write_batch_device(d, off, kl, None, n)
+    t2 = time.perf_counter()
     s.flush()
+    t3 = time.perf_counter()
     ctr = s.counters()
     tms = s.times()
     s.close()
+    t4 = time.perf_counter()
+    tms["host_create_ns"] = int((t1 - t0) * 1e9)
+    tms["host_absorb_ns"] = int((t2 - t1) * 1e9)
+    tms["host_flush_ns"] = int((t3 - t2) * 1e9)
+    tms["host_close_ns"] = int((t4 - t3) * 1e9)
     return ctr, tms
 
 
