@@ -888,6 +888,15 @@ static void pool_free(void* p, size_t cls) { if (p) pool_map()[cls].push_back(p)
 struct DBuf {
   void* p = nullptr;
   size_t sz = 0;   /* size class */
+  DBuf() = default;
+  DBuf(const DBuf&) = delete;
+  DBuf& operator=(const DBuf&) = delete;
+  DBuf(DBuf&& o) noexcept : p(o.p), sz(o.sz) { o.p = nullptr; o.sz = 0; }
+  DBuf& operator=(DBuf&& o) noexcept {
+    if (this != &o) { release(); p = o.p; sz = o.sz; o.p = nullptr; o.sz = 0; }
+    return *this;
+  }
+  ~DBuf() { release(); }
   int alloc(size_t n) {
     if (n <= sz) return 0;
     release();
