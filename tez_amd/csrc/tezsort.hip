@@ -224,9 +224,14 @@ __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32
   }
 }
 
-/* composite = (part << (64-pbits)) | (first 8 content bytes BE >> pbits) */
+/* composite = (part << (64-pbits)) | (first content bytes BE >> pbits),
+ * masked to the top sort_bytes bytes: the radix sort only orders those
+ * (adaptive pass count — DESIGN.md §4); rarer-than-1% ties are resolved by
+ * the refinement levels, whose equality test must see the same mask. */
 __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int pbits,
-                                  uint64_t* d_key, uint32_t* d_idx, uint32_t n) {
+                                  int sort_bytes, uint64_t* d_key, uint32_t* d_idx,
+                                  uint32_t n) {
+  uint64_t mask = (sort_bytes >= 8) ? ~0ull : ~0ull << (8 * (8 - sort_bytes));
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, i);
@@ -234,7 +239,7 @@ __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int pbits,
     uint32_t m = v.clen < 8 ? v.clen : 8;
     for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
     uint64_t key = pbits ? (((uint64_t)(uint32_t)d_part[i] << (64 - pbits)) | (c >> pbits)) : c;
-    d_key[i] = key;
+    d_key[i] = key & mask;
     d_idx[i] = i;
   }
 }
@@ -654,30 +659,59 @@ __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
   return crc;
 }
 
-__global__ void k_crc_chunks(const uint8_t* stream, const uint64_t* range_start,
-                             const uint64_t* range_len, const uint64_t* chunk_base,
-                             uint32_t nparts, uint32_t total_chunks, uint32_t* chunk_crc) {
+/* Super-chunk staged CRC: one block stages 256 chunks (64 KB) of a
+ * partition's checksummed range into LDS with coalesced u32 loads (the
+ * per-thread strided byte reads of the naive version thrash L1), then each
+ * thread CRCs its 256-byte chunk from LDS.  Chunk rows are padded to 260 B so
+ * lane t's u32 reads land on distinct banks. */
+#define CRC_SC_CHUNKS 256
+#define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 64 KiB */
+__global__ __launch_bounds__(BLOCK) void k_crc_chunks(
+    const uint8_t* stream, const uint64_t* range_start, const uint64_t* range_len,
+    const uint64_t* chunk_base, const uint64_t* sc_base /* [P+1] */,
+    uint32_t nparts, uint32_t total_sc, uint32_t* chunk_crc) {
   __shared__ uint32_t tab[256];
+  __shared__ uint8_t stage[CRC_SC_CHUNKS * (CRC_CHUNK + 4)];
   for (int i = threadIdx.x; i < 256; i += blockDim.x) tab[i] = c_crc_table[i];
-  __syncthreads();
-  for (uint32_t c = blockIdx.x * blockDim.x + threadIdx.x; c < total_chunks;
-       c += gridDim.x * blockDim.x) {
-    /* find partition: linear scan ok for small P; binary search otherwise */
+  for (uint32_t sc = blockIdx.x; sc < total_sc; sc += gridDim.x) {
     uint32_t lo = 0, hi = nparts;
     while (lo + 1 < hi) {
       uint32_t mid = (lo + hi) / 2;
-      if (chunk_base[mid] <= c) lo = mid; else hi = mid;
+      if (sc_base[mid] <= sc) lo = mid; else hi = mid;
     }
     uint32_t p = lo;
-    uint64_t ci = c - chunk_base[p];
-    uint64_t off = range_start[p] + ci * CRC_CHUNK;
-    uint64_t len = range_len[p] - ci * CRC_CHUNK;
-    if (len > CRC_CHUNK) len = CRC_CHUNK;
-    uint32_t crc = 0xFFFFFFFFu;
-    const uint8_t* ptr = stream + off;
-    for (uint64_t b = 0; b < len; b++)
-      crc = tab[(crc ^ ptr[b]) & 0xFF] ^ (crc >> 8);
-    chunk_crc[c] = crc ^ 0xFFFFFFFFu;
+    uint64_t sc_local = sc - sc_base[p];
+    uint64_t byte0 = sc_local * CRC_SC_BYTES;
+    uint64_t avail = range_len[p] - byte0;
+    if (avail > CRC_SC_BYTES) avail = CRC_SC_BYTES;
+    uint64_t gbase = range_start[p] + byte0;
+    /* stage: aligned u32 loads; per-byte LDS scatter into padded rows */
+    uint32_t a = (uint32_t)(gbase & 3);
+    const uint32_t* wsrc = (const uint32_t*)(stream + gbase - a);
+    uint32_t nwords = (uint32_t)((a + avail + 3) / 4);
+    __syncthreads();
+    for (uint32_t wi = threadIdx.x; wi < nwords; wi += blockDim.x) {
+      uint32_t w = wsrc[wi];
+      for (int j = 0; j < 4; j++) {
+        int64_t g = (int64_t)wi * 4 + j - a;
+        if (g >= 0 && g < (int64_t)avail) {
+          uint32_t ch = (uint32_t)(g / CRC_CHUNK), o = (uint32_t)(g % CRC_CHUNK);
+          stage[ch * (CRC_CHUNK + 4) + o] = (uint8_t)(w >> (8 * j));
+        }
+      }
+    }
+    __syncthreads();
+    uint32_t t = threadIdx.x;
+    uint64_t c0 = t * (uint64_t)CRC_CHUNK;
+    if (c0 < avail) {
+      uint64_t len = avail - c0;
+      if (len > CRC_CHUNK) len = CRC_CHUNK;
+      uint32_t crc = 0xFFFFFFFFu;
+      const uint8_t* ptr = stage + t * (CRC_CHUNK + 4);
+      for (uint64_t b = 0; b < len; b++)
+        crc = tab[(crc ^ ptr[b]) & 0xFF] ^ (crc >> 8);
+      chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + t] = crc ^ 0xFFFFFFFFu;
+    }
   }
 }
 
@@ -1217,17 +1251,25 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
 
   int P = s->conf.num_partitions;
   int pbits = s->pbits;
+  /* adaptive radix width: enough composite bits that expected tie-involved
+     records stay below ~1% of n (ties go through refinement anyway):
+     bits = pbits + log2(n) + 6.  Tests at small n exercise refinement hard. */
+  int needed_bits = pbits + 6;
+  for (uint64_t v = n; v; v >>= 1) needed_bits++;
+  int SB = (needed_bits + 7) / 8;
+  if (SB < 2) SB = 2;
+  if (SB > 8) SB = 8;
   /* 1. composites */
   if (s->skey.alloc(sizeof(uint64_t) * n)) return -12;
   if (s->sidx.alloc(sizeof(uint32_t) * n)) return -12;
   uint64_t* d_key = (uint64_t*)s->skey.p;
   uint32_t* d_idx = (uint32_t*)s->sidx.p;
   hipLaunchKernelGGL(k_build_composite, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                     d_part_unsorted, pbits, d_key, d_idx, n);
+                     d_part_unsorted, pbits, SB, d_key, d_idx, n);
   (void)hipEventRecord(ev[1]);
 
-  /* 2. base radix sort over the full u64 composite */
-  int rc = radix_sort<uint64_t>(d_key, d_idx, nullptr, n, 8);
+  /* 2. base radix sort over the top SB bytes of the composite */
+  int rc = radix_sort<uint64_t>(d_key, d_idx, nullptr, n, 8, 8 - SB);
   if (rc) return rc;
   (void)hipEventRecord(ev[2]);
 
@@ -1236,7 +1278,8 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   uint8_t* d_eq = (uint8_t*)s->eq.p;
   hipLaunchKernelGGL(k_eq_init, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, d_eq, n);
   /* max content length: for refinement level count */
-  int c0 = (64 - pbits) / 8;
+  int c0 = (8 * SB - pbits) / 8;
+  if (c0 < 0) c0 = 0;
   /* determine max clen lazily: use a safe cap by scanning klen on host?  We
      compute it from the conf: key_type BYTES => clen = klen-4 (max over
      spills).  For TEXT, clen <= klen-1.  Host keeps max_klen per spill. */
@@ -1468,11 +1511,20 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                      d_out, (const uint64_t*)d_segstart.p, (const uint64_t*)d_body.p,
                      (const uint8_t*)d_lastsame.p, (const uint32_t*)d_partcrc.p,
                      (const uint8_t*)d_present.p, P);
-  if (total_chunks)
-    hipLaunchKernelGGL(k_crc_chunks, dim3(grid1d(total_chunks)), dim3(BLOCK), 0, 0, d_out,
-                       (const uint64_t*)d_rstart.p, (const uint64_t*)d_rlen.p,
-                       (const uint64_t*)d_chunkbase.p, P, (uint32_t)total_chunks,
-                       (uint32_t*)d_chunkcrc.p);
+  std::vector<uint64_t> h_scbase(P + 1, 0);
+  for (int p = 0; p < P; p++) {
+    uint64_t nchunks = h_chunkbase[p + 1] - h_chunkbase[p];
+    h_scbase[p + 1] = h_scbase[p] + (nchunks + CRC_SC_CHUNKS - 1) / CRC_SC_CHUNKS;
+  }
+  uint64_t total_sc = h_scbase[P];
+  static thread_local DBuf d_scbase;
+  if (up(d_scbase, h_scbase.data(), 8 * (P + 1))) return -12;
+  if (total_sc)
+    hipLaunchKernelGGL(k_crc_chunks,
+                       dim3((uint32_t)min(total_sc, (uint64_t)4096)), dim3(BLOCK), 0, 0,
+                       d_out, (const uint64_t*)d_rstart.p, (const uint64_t*)d_rlen.p,
+                       (const uint64_t*)d_chunkbase.p, (const uint64_t*)d_scbase.p, P,
+                       (uint32_t)total_sc, (uint32_t*)d_chunkcrc.p);
   std::vector<uint64_t> h_groupbase(P + 1, 0);
   for (int p = 0; p < P; p++) {
     uint64_t nchunks = h_chunkbase[p + 1] - h_chunkbase[p];
