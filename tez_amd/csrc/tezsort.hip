@@ -26,7 +26,7 @@
 #define BLOCK 256
 #define WPB (BLOCK / WAVE)
 #define RADIX 256
-#define TILE_ROUNDS 8
+#define TILE_ROUNDS 16
 #define TILE (TILE_ROUNDS * BLOCK) /* elements per scatter block */
 
 /* ------------------------------------------------------------------ */
@@ -158,8 +158,17 @@ __device__ __forceinline__ int d_vint_decoded_size(int8_t first) {
 }
 
 __device__ __forceinline__ int32_t d_hash_bytes(const uint8_t* p, int32_t n) {
+  /* h = 31*h + signed(byte) — fetch 8 bytes at a time (byte loads cost an
+     instruction each and serialize the chain on memory latency) */
   int32_t h = 1;
-  for (int32_t i = 0; i < n; i++) h = (int32_t)((uint32_t)h * 31u) + (int8_t)p[i];
+  int32_t i = 0;
+  for (; i + 8 <= n; i += 8) {
+    uint64_t w;
+    __builtin_memcpy(&w, p + i, 8);
+    for (int j = 0; j < 8; j++)
+      h = (int32_t)((uint32_t)h * 31u) + (int8_t)(uint8_t)(w >> (8 * j));
+  }
+  for (; i < n; i++) h = (int32_t)((uint32_t)h * 31u) + (int8_t)p[i];
   return h;
 }
 
@@ -214,6 +223,9 @@ __global__ void k_max_u32(const uint32_t* a, uint32_t n, uint32_t* out);
 __global__ void k_part_hist(const uint32_t* parts, uint32_t n, uint32_t* counts, int P);
 __global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* sidx,
                               int lb0, int use_len, uint64_t* lkey, uint32_t m);
+__global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
+                                   const uint8_t* same, uint64_t total_body,
+                                   uint32_t n, int P, uint64_t* out);
 
 /* ---- partition + composite ---- */
 __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32_t n) {
@@ -307,33 +319,48 @@ __global__ void k_radix_scan_digits(uint32_t* totals, uint32_t* bases) {
   bases[t] = lds[t] - totals[t];
 }
 
-/* stable scatter with up to two u32 payloads */
+/* Stable scatter with up to two u32 payloads.  Two-phase LDS staging:
+ * (1) per-tile stable ranking (wave ballots + wave histograms), elements
+ * reordered into LDS grouped by digit; (2) digit-contiguous cooperative
+ * global writes — a direct per-element scatter write-allocates a 64B line
+ * per 12B element (PMC: 22 GB written for 6 GB of payload); digit runs of
+ * TILE/256 elements restore coalescing. */
 template <typename KeyT, bool HAS_A1>
-__global__ void k_radix_scatter(const KeyT* keys_in, KeyT* keys_out,
-                                const uint32_t* a0_in, uint32_t* a0_out,
-                                const uint32_t* a1_in, uint32_t* a1_out,
-                                uint32_t n, int byte_idx,
-                                const uint32_t* offsets, const uint32_t* bases) {
-  __shared__ uint32_t running[RADIX];
+__global__ __launch_bounds__(BLOCK) void k_radix_scatter(
+    const KeyT* keys_in, KeyT* keys_out,
+    const uint32_t* a0_in, uint32_t* a0_out,
+    const uint32_t* a1_in, uint32_t* a1_out,
+    uint32_t n, int byte_idx,
+    const uint32_t* offsets, const uint32_t* bases) {
+  __shared__ uint32_t tilecnt[RADIX];       /* per-digit running count in tile */
   __shared__ uint32_t wavehist[WPB][RADIX];
-  for (int i = threadIdx.x; i < RADIX; i += blockDim.x)
-    running[i] = offsets[blockIdx.x * RADIX + i] + bases[i];
+  __shared__ uint32_t tileoff[RADIX];       /* exclusive scan of final counts */
+  __shared__ KeyT ls_key[TILE];
+  __shared__ uint32_t ls_a0[TILE];
+  __shared__ uint32_t ls_a1[HAS_A1 ? TILE : 1];
+  __shared__ uint8_t ls_dig[TILE];
+  for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
   uint32_t start = blockIdx.x * TILE;
   uint32_t end = min(start + TILE, n);
+  uint32_t count = end - start;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wv = threadIdx.x / WAVE;
   const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
-  for (uint32_t r0 = start; r0 < end; r0 += blockDim.x) {
+  uint32_t my_seq[TILE_ROUNDS];             /* rank within (tile, digit) */
+  uint32_t my_dig[TILE_ROUNDS];
+  KeyT my_key[TILE_ROUNDS];
+  uint32_t my_a0[TILE_ROUNDS];
+  uint32_t my_a1v[TILE_ROUNDS];
+  int round = 0;
+  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
     uint32_t i = r0 + threadIdx.x;
     bool active = i < end;
     KeyT key = active ? keys_in[i] : (KeyT)0;
     uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0xFFFFFFFFu;
-    /* zero wave hists */
     __syncthreads();
     for (int j = threadIdx.x; j < WPB * RADIX; j += blockDim.x)
       ((uint32_t*)wavehist)[j] = 0;
     __syncthreads();
-    /* wave match: mask of lanes with same digit */
     uint64_t m = ~0ull;
     for (int b = 0; b < 8; b++) {
       uint64_t bb = __ballot((d >> b) & 1);
@@ -344,20 +371,62 @@ __global__ void k_radix_scatter(const KeyT* keys_in, KeyT* keys_out,
     uint32_t lane_rank = (uint32_t)__popcll(m & lt_mask);
     if (active && lane_rank == 0) wavehist[wv][d] = (uint32_t)__popcll(m);
     __syncthreads();
+    my_key[round] = key;
+    my_dig[round] = d;
+    my_a0[round] = active ? a0_in[i] : 0;
+    if (HAS_A1) my_a1v[round] = active ? a1_in[i] : 0;
     if (active) {
       uint32_t prior = 0;
       for (int w = 0; w < wv; w++) prior += wavehist[w][d];
-      uint32_t pos = running[d] + prior + lane_rank;
-      keys_out[pos] = key;
-      a0_out[pos] = a0_in[i];
-      if (HAS_A1) a1_out[pos] = a1_in[i];
+      my_seq[round] = tilecnt[d] + prior + lane_rank;
+    } else {
+      my_seq[round] = 0;
     }
     __syncthreads();
     for (int j = threadIdx.x; j < RADIX; j += blockDim.x) {
-      uint32_t s = 0;
-      for (int w = 0; w < WPB; w++) s += wavehist[w][j];
-      running[j] += s;
+      uint32_t sum = 0;
+      for (int w = 0; w < WPB; w++) sum += wavehist[w][j];
+      tilecnt[j] += sum;
     }
+  }
+  __syncthreads();
+  /* exclusive scan of tilecnt -> tileoff (256 entries, one block) */
+  {
+    int t = threadIdx.x;
+    uint32_t v = (t < RADIX) ? tilecnt[t] : 0;
+    /* reuse wavehist row 0 as scan scratch */
+    uint32_t* sc = (uint32_t*)wavehist;
+    if (t < RADIX) sc[t] = v;
+    __syncthreads();
+    for (int st = 1; st < RADIX; st <<= 1) {
+      uint32_t add = (t >= st && t < RADIX) ? sc[t - st] : 0;
+      __syncthreads();
+      if (t < RADIX) sc[t] += add;
+      __syncthreads();
+    }
+    if (t < RADIX) tileoff[t] = sc[t] - v;
+    __syncthreads();
+  }
+  /* place elements into LDS grouped by digit */
+  round = 0;
+  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+    uint32_t i = r0 + threadIdx.x;
+    if (i < end) {
+      uint32_t slot = tileoff[my_dig[round]] + my_seq[round];
+      ls_key[slot] = my_key[round];
+      ls_a0[slot] = my_a0[round];
+      if (HAS_A1) ls_a1[slot] = my_a1v[round];
+      ls_dig[slot] = (uint8_t)my_dig[round];
+    }
+  }
+  __syncthreads();
+  /* digit-contiguous global writes */
+  for (uint32_t j = threadIdx.x; j < count; j += blockDim.x) {
+    uint32_t d = ls_dig[j];
+    uint32_t pos = offsets[blockIdx.x * RADIX + d] + bases[d] + (j - tileoff[d]);
+    keys_out[pos] = ls_key[j];
+    a0_out[pos] = ls_a0[j];
+    if (HAS_A1) a1_out[pos] = ls_a1[j];
   }
 }
 
@@ -487,6 +556,26 @@ __global__ void k_count_nonzero_u8(const uint8_t* a, uint32_t n, uint32_t* out) 
 }
 
 /* ---- emit ---- */
+/* Sorted record descriptors: one gather pass per sort; all later emit-side
+ * kernels read these coalesced instead of re-gathering off/klen per record
+ * (PMC showed 13+ GB of over-fetch per re-gather at n=1e8). */
+struct RecDesc {
+  uint64_t src;   /* absolute device address of the serialized record */
+  uint32_t klen;
+  uint32_t vlen;
+};
+__global__ void k_build_desc(RecTable rt, const uint32_t* sidx, RecDesc* desc,
+                             uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    RecView v = rt_view(rt, sidx[i]);
+    RecDesc d;
+    d.src = (uint64_t)(uintptr_t)v.key;
+    d.klen = v.klen;
+    d.vlen = v.vlen;
+    desc[i] = d;
+  }
+}
 /* same-as-prev full-key flags are exactly the final eq[] array.
  * writer-sameness (what the IFile stream encodes as RLE) additionally
  * depends on the rle flag and merge provenance:
@@ -508,11 +597,11 @@ __global__ void k_writer_same(RecTable rt, const uint32_t* sidx, const uint8_t* 
     same[i] = s;
   }
 }
-__global__ void k_emit_sizes(RecTable rt, const uint32_t* sidx, const uint8_t* same,
+__global__ void k_emit_sizes(const RecDesc* desc, const uint8_t* same,
                              const uint32_t* parts, uint64_t* sizes, uint32_t n) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
-    RecView v = rt_view(rt, sidx[i]);
+    RecDesc v = desc[i];
     uint64_t sz;
     /* a run never crosses a partition segment: the previous record's RLE
        state is invisible to this partition's IFile stream */
@@ -537,7 +626,7 @@ __global__ void k_sorted_parts(const uint64_t* skeys, int pbits, uint32_t* parts
  * One WAVE per record: lane-parallel payload copy (the serialized key and
  * value are contiguous in the record, so the payload is one span); lane 0
  * writes the marker/vint header.  Coalesced on both sides. */
-__global__ void k_emit_records(RecTable rt, const uint32_t* sidx, const uint8_t* same,
+__global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
                                const uint64_t* scan, const uint32_t* parts,
                                const uint64_t* seg_payload_start,
                                const uint64_t* part_scan_base,
@@ -552,47 +641,29 @@ __global__ void k_emit_records(RecTable rt, const uint32_t* sidx, const uint8_t*
   for (uint64_t base = (uint64_t)wave * WAVE; base < n;
        base += (uint64_t)nwaves * WAVE) {
     uint32_t i = (uint32_t)base + lane;
-    uint64_t my_src = 0, my_dst = 0, my_h0 = 0;
+    uint64_t my_src = 0, my_dst = 0, my_h0 = 0, my_h1 = 0;
     uint32_t my_len = 0, my_hdr = 0;
     if (i < n) {
-      RecView v = rt_view(rt, sidx[i]);
+      RecDesc v = desc[i];
       uint32_t p = parts[i];
       my_dst = seg_payload_start[p] + (scan[i] - part_scan_base[p]);
       uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
-      uint8_t hdrbuf[12] = {0};
+      uint8_t hdrbuf[16] = {0};
       uint32_t hdr = 0;
       if (same[i]) {
         if (!prev_same) hdrbuf[hdr++] = 0xFE; /* RLE_MARKER -2 */
         hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-        my_src = (uint64_t)(uintptr_t)v.val;
+        my_src = v.src + v.klen;
         my_len = v.vlen;
       } else {
         if (prev_same) hdrbuf[hdr++] = 0xFD; /* V_END_MARKER -3 */
         hdr += d_vint_write(hdrbuf + hdr, v.klen);
         hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-        my_src = (uint64_t)(uintptr_t)v.key; /* key ‖ val contiguous */
+        my_src = v.src;               /* key ‖ val contiguous */
         my_len = v.klen + v.vlen;
       }
       my_hdr = hdr;
       for (int b = 0; b < 8; b++) my_h0 |= (uint64_t)hdrbuf[b] << (8 * b);
-      /* hdr <= 12 but vints of sane lens are <= 5+5+1: pack overflow into len
-         high bits is not needed; bytes 8..11 handled below via second word */
-    }
-    uint64_t my_h1 = 0;
-    if (i < n && my_hdr > 8) {
-      /* rebuild high header bytes (rare: huge klen/vlen vints) */
-      RecView v = rt_view(rt, sidx[i]);
-      uint8_t hdrbuf[12] = {0};
-      uint32_t hdr = 0;
-      uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
-      if (same[i]) {
-        if (!prev_same) hdrbuf[hdr++] = 0xFE;
-        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-      } else {
-        if (prev_same) hdrbuf[hdr++] = 0xFD;
-        hdr += d_vint_write(hdrbuf + hdr, v.klen);
-        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-      }
       for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
     }
     uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
@@ -1393,9 +1464,14 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
   hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
                      (uint32_t*)s->parts_sorted.p, n);
+  static thread_local DBuf descbuf;
+  if (descbuf.alloc(sizeof(RecDesc) * n)) return -12;
+  hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
+                     (RecDesc*)descbuf.p, n);
   if (s->sizes.alloc(sizeof(uint64_t) * n)) return -12;
   if (s->scan.alloc(sizeof(uint64_t) * n)) return -12;
-  hipLaunchKernelGGL(k_emit_sizes, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
+  hipLaunchKernelGGL(k_emit_sizes, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                     (const RecDesc*)descbuf.p,
                      (const uint8_t*)s->same.p, (const uint32_t*)s->parts_sorted.p,
                      (uint64_t*)s->sizes.p, n);
   uint64_t total_body = 0;
@@ -1418,25 +1494,25 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   s->final_rt = rt;
   s->final_n = n;
   /* body bytes per partition = scan[start of next] - scan[start] */
+  /* gather per-partition scan bases + last-same flags in one kernel +
+     one D2H (was 2P+1 synchronous 8-byte copies) */
   std::vector<uint64_t> h_scan_at(P + 1, 0);
-  {
-    /* gather scan at partition record starts (host: P+1 copies — fine) */
-    for (int p = 0; p <= P; p++) {
-      uint64_t v = total_body;
-      if (h_prec_start[p] < n)
-        HIP_CHECK(hipMemcpy(&v, (uint64_t*)s->scan.p + h_prec_start[p], 8,
-                            hipMemcpyDeviceToHost));
-      h_scan_at[p] = v;
-    }
-  }
-  /* last_same flag per partition (V_END before EOF) */
   std::vector<uint8_t> h_last_same(P, 0);
-  for (int p = 0; p < P; p++) {
-    if (h_pcount[p] == 0) continue;
-    uint64_t last = h_prec_start[p + 1] - 1;
-    uint8_t v = 0;
-    HIP_CHECK(hipMemcpy(&v, (uint8_t*)s->same.p + last, 1, hipMemcpyDeviceToHost));
-    h_last_same[p] = v;
+  {
+    static thread_local DBuf d_pstart, d_gathered;
+    if (d_pstart.alloc(8 * (P + 1))) return -12;
+    if (d_gathered.alloc(16 * (P + 1))) return -12;
+    HIP_CHECK(hipMemcpyAsync(d_pstart.p, h_prec_start.data(), 8 * (P + 1),
+                             hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_gather_part_meta, dim3((P + BLOCK) / BLOCK), dim3(BLOCK), 0, 0,
+                       (const uint64_t*)d_pstart.p, (const uint64_t*)s->scan.p,
+                       (const uint8_t*)s->same.p, total_body, n, P,
+                       (uint64_t*)d_gathered.p);
+    std::vector<uint64_t> tmp(2 * (P + 1));
+    HIP_CHECK(hipMemcpy(tmp.data(), d_gathered.p, 16 * (P + 1), hipMemcpyDeviceToHost));
+    for (int p = 0; p <= P; p++) h_scan_at[p] = tmp[2 * p];
+    for (int p = 0; p < P; p++)
+      h_last_same[p] = (h_pcount[p] > 0) ? (uint8_t)tmp[2 * p + 1] : 0;
   }
   /* segment layout */
   int send_empty = s->conf.send_empty_partition_details;
@@ -1491,7 +1567,8 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   if (up(d_rlen, h_range_len.data(), 8 * P)) return -12;
 
   /* 6. emit records */
-  hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0, rt, d_idx,
+  hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
+                     (const RecDesc*)descbuf.p,
                      (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
                      (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
                      (const uint64_t*)d_scanbase.p, d_out, n);
@@ -1587,6 +1664,20 @@ __global__ void k_part_hist(const uint32_t* parts, uint32_t n, uint32_t* counts,
   for (int j = threadIdx.x; j < P; j += blockDim.x)
     if (lh[j]) atomicAdd(&counts[j], lh[j]);
 }
+__global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
+                                   const uint8_t* same, uint64_t total_body,
+                                   uint32_t n, int P, uint64_t* out) {
+  int p = blockIdx.x * blockDim.x + threadIdx.x;
+  if (p > P) return;
+  uint64_t r = pstart[p];
+  out[2 * p] = (r < n) ? scan[r] : total_body;
+  /* last-same of partition p (p < P): same[] at pstart[p+1]-1 */
+  if (p < P) {
+    uint64_t e = pstart[p + 1];
+    out[2 * p + 1] = (e > 0 && e <= n && e > r) ? (uint64_t)same[e - 1] : 0;
+  }
+}
+
 __global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* sidx,
                               int lb0, int use_len, uint64_t* lkey, uint32_t m) {
   for (uint32_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m;
