@@ -26,7 +26,7 @@
 #define BLOCK 256
 #define WPB (BLOCK / WAVE)
 #define RADIX 256
-#define TILE_ROUNDS 16
+#define TILE_ROUNDS 8
 #define TILE (TILE_ROUNDS * BLOCK) /* elements per scatter block */
 
 /* ------------------------------------------------------------------ */
@@ -667,7 +667,7 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
     }
     uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
-    for (uint32_t r = 0; r < nvalid; r++) {
+    auto emit_one = [&](uint32_t r) {
       uint64_t src = __shfl(my_src, r);
       uint64_t dsto = __shfl(my_dst, r);
       uint64_t h0 = __shfl(my_h0, r);
@@ -681,6 +681,12 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       w += hdr;
       const uint8_t* sp = (const uint8_t*)(uintptr_t)src;
       for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp[b];
+    };
+    if (nvalid == WAVE) {
+#pragma unroll 4
+      for (uint32_t r = 0; r < WAVE; r++) emit_one(r);
+    } else {
+      for (uint32_t r = 0; r < nvalid; r++) emit_one(r);
     }
   }
 }
@@ -717,6 +723,7 @@ __global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
  * reduces its chunk list sequentially with the 256-byte shift matrix. */
 #define CRC_CHUNK 256
 __constant__ uint32_t c_crc_table[256];
+__constant__ uint32_t c_crc_table4[4][256]; /* slice-by-4: T0 = standard */
 __constant__ uint32_t c_crc_mats[CRC_MATS][32];
 
 __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
@@ -730,20 +737,23 @@ __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
   return crc;
 }
 
-/* Super-chunk staged CRC: one block stages 256 chunks (64 KB) of a
- * partition's checksummed range into LDS with coalesced u32 loads (the
- * per-thread strided byte reads of the naive version thrash L1), then each
- * thread CRCs its 256-byte chunk from LDS.  Chunk rows are padded to 260 B so
- * lane t's u32 reads land on distinct banks. */
-#define CRC_SC_CHUNKS 256
-#define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 64 KiB */
+/* Super-chunk staged CRC v3: one block stages 128 chunks (32 KB) of a
+ * partition's checksummed range into LDS via coalesced aligned-u32 loads
+ * with a funnel shift for misaligned range starts, then each thread CRCs its
+ * 256-byte chunk from LDS with slice-by-4 tables (4 lookups per word, 1/4 the
+ * dependent-chain length of bytewise).  Rows padded one word so lane t's
+ * reads land on distinct banks. */
+#define CRC_SC_CHUNKS 128
+#define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 32 KiB */
+#define CRC_ROW_W (CRC_CHUNK / 4 + 1)            /* 65 words per chunk row */
 __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     const uint8_t* stream, const uint64_t* range_start, const uint64_t* range_len,
     const uint64_t* chunk_base, const uint64_t* sc_base /* [P+1] */,
     uint32_t nparts, uint32_t total_sc, uint32_t* chunk_crc) {
-  __shared__ uint32_t tab[256];
-  __shared__ uint8_t stage[CRC_SC_CHUNKS * (CRC_CHUNK + 4)];
-  for (int i = threadIdx.x; i < 256; i += blockDim.x) tab[i] = c_crc_table[i];
+  __shared__ uint32_t tab4[4][256];
+  __shared__ uint32_t stage32[CRC_SC_CHUNKS * CRC_ROW_W];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x)
+    ((uint32_t*)tab4)[i] = ((const uint32_t*)c_crc_table4)[i];
   for (uint32_t sc = blockIdx.x; sc < total_sc; sc += gridDim.x) {
     uint32_t lo = 0, hi = nparts;
     while (lo + 1 < hi) {
@@ -756,40 +766,41 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     uint64_t avail = range_len[p] - byte0;
     if (avail > CRC_SC_BYTES) avail = CRC_SC_BYTES;
     uint64_t gbase = range_start[p] + byte0;
-    /* stage: aligned u32 loads; per-byte LDS scatter into padded rows */
     uint32_t a = (uint32_t)(gbase & 3);
     const uint32_t* wsrc = (const uint32_t*)(stream + gbase - a);
-    uint32_t nwords = (uint32_t)((a + avail + 3) / 4);
+    uint32_t nwords = (uint32_t)((avail + 3) / 4);
     __syncthreads();
     for (uint32_t wi = threadIdx.x; wi < nwords; wi += blockDim.x) {
       uint32_t w = wsrc[wi];
-      for (int j = 0; j < 4; j++) {
-        int64_t g = (int64_t)wi * 4 + j - a;
-        if (g >= 0 && g < (int64_t)avail) {
-          uint32_t ch = (uint32_t)(g / CRC_CHUNK), o = (uint32_t)(g % CRC_CHUNK);
-          stage[ch * (CRC_CHUNK + 4) + o] = (uint8_t)(w >> (8 * j));
-        }
-      }
+      if (a) w = (w >> (8 * a)) | (wsrc[wi + 1] << (32 - 8 * a));
+      stage32[(wi >> 6) * CRC_ROW_W + (wi & 63)] = w;
     }
     __syncthreads();
     uint32_t t = threadIdx.x;
-    uint64_t c0 = t * (uint64_t)CRC_CHUNK;
-    if (c0 < avail) {
+    uint64_t c0 = (uint64_t)t * CRC_CHUNK;
+    if (t < CRC_SC_CHUNKS && c0 < avail) {
       uint64_t len = avail - c0;
       if (len > CRC_CHUNK) len = CRC_CHUNK;
       uint32_t crc = 0xFFFFFFFFu;
-      const uint8_t* ptr = stage + t * (CRC_CHUNK + 4);
-      for (uint64_t b = 0; b < len; b++)
-        crc = tab[(crc ^ ptr[b]) & 0xFF] ^ (crc >> 8);
+      const uint32_t* row = stage32 + t * CRC_ROW_W;
+      uint32_t full = (uint32_t)(len / 4);
+      for (uint32_t w = 0; w < full; w++) {
+        uint32_t c = crc ^ row[w];
+        crc = tab4[3][c & 0xFF] ^ tab4[2][(c >> 8) & 0xFF]
+            ^ tab4[1][(c >> 16) & 0xFF] ^ tab4[0][c >> 24];
+      }
+      for (uint32_t b = full * 4; b < len; b++) {
+        uint8_t byte = (uint8_t)(row[b / 4] >> (8 * (b & 3)));
+        crc = (crc >> 8) ^ tab4[0][(crc ^ byte) & 0xFF];
+      }
       chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + t] = crc ^ 0xFFFFFFFFu;
     }
   }
 }
 
-/* Tree combine, two levels.  Group = up to 2048 chunks (512 KB of payload):
- * thread t serially folds its 8 consecutive chunks, then an LDS tree folds
- * the 256 threads; a final kernel folds each partition's groups.  Every fold
- * is combine(L,R) = shift(crc_L, len_R) ^ crc_R with the generic
+/* Tree combine, two levels: thread-serial fold of 8 chunks, then an LDS tree
+ * over ADJACENT spans; a final kernel folds each partition's groups.  Every
+ * fold is combine(L,R) = shift(crc_L, len_R) ^ crc_R with the generic
  * bit-decomposed shift so ragged tails are exact. */
 #define CRC_GROUP_CHUNKS 2048
 __global__ void k_crc_combine_groups(const uint64_t* range_len, const uint64_t* chunk_base,
@@ -932,6 +943,14 @@ static int ensure_device_constants() {
   h_crc_init();
   h_build_crc_mats();
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table), h_crc_table, sizeof(h_crc_table)));
+  {
+    static uint32_t t4[4][256];
+    for (int i = 0; i < 256; i++) t4[0][i] = h_crc_table[i];
+    for (int k = 1; k < 4; k++)
+      for (int i = 0; i < 256; i++)
+        t4[k][i] = (t4[k - 1][i] >> 8) ^ h_crc_table[t4[k - 1][i] & 0xFF];
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table4), t4, sizeof(t4)));
+  }
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_mats), h_crc_shift_mat, sizeof(h_crc_shift_mat)));
   done = true;
   return 0;
@@ -1543,7 +1562,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     outsp->index[p].part_length = partl;
   }
   uint64_t stream_len = cursor;
-  if (outsp->ifile.alloc(stream_len ? stream_len : 1)) return -12;
+  if (outsp->ifile.alloc(stream_len ? stream_len + 16 : 1)) return -12;
   uint8_t* d_out = (uint8_t*)outsp->ifile.p;
   outsp->ifile_len = (int64_t)stream_len;
   outsp->rle = (uint8_t)writer_rle;
