@@ -667,26 +667,67 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
     }
     uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
-    auto emit_one = [&](uint32_t r) {
-      uint64_t src = __shfl(my_src, r);
-      uint64_t dsto = __shfl(my_dst, r);
-      uint64_t h0 = __shfl(my_h0, r);
-      uint64_t h1 = __shfl(my_h1, r);
-      uint32_t len = __shfl(my_len, r);
-      uint32_t hdr = __shfl(my_hdr, r);
-      uint8_t* w = out + dsto;
-      if (lane < hdr)
-        w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
-                             : (uint8_t)(h1 >> (8 * (lane - 8)));
-      w += hdr;
-      const uint8_t* sp = (const uint8_t*)(uintptr_t)src;
-      for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp[b];
-    };
-    if (nvalid == WAVE) {
-#pragma unroll 4
-      for (uint32_t r = 0; r < WAVE; r++) emit_one(r);
+    /* payload-length cap for the software-pipelined path: each lane buffers
+       PIPE_B bytes of a record (covers records up to 64*PIPE_B bytes) */
+    constexpr int PIPE_B = 2;
+    uint32_t maxlen = my_len;
+    for (int sh = 32; sh >= 1; sh >>= 1) {
+      uint32_t o = __shfl_xor(maxlen, sh);
+      if (o > maxlen) maxlen = o;
+    }
+    if (nvalid == WAVE && maxlen <= WAVE * PIPE_B) {
+      /* 2-deep pipeline: issue record r+1's gather while storing record r —
+         the plain loop serializes on one load->store chain per record and
+         runs at ~10% duty cycle */
+      uint64_t src0 = __shfl(my_src, 0), dst0 = __shfl(my_dst, 0);
+      uint64_t h00 = __shfl(my_h0, 0), h10 = __shfl(my_h1, 0);
+      uint32_t len0 = __shfl(my_len, 0), hdr0 = __shfl(my_hdr, 0);
+      uint8_t b0[PIPE_B];
+      const uint8_t* sp0 = (const uint8_t*)(uintptr_t)src0;
+#pragma unroll
+      for (int k = 0; k < PIPE_B; k++)
+        b0[k] = (lane + k * WAVE < len0) ? sp0[lane + k * WAVE] : 0;
+      for (uint32_t r = 0; r < WAVE; r++) {
+        uint64_t src1 = 0, dst1 = 0, h01 = 0, h11 = 0;
+        uint32_t len1 = 0, hdr1 = 0;
+        uint8_t b1[PIPE_B] = {0};
+        if (r + 1 < WAVE) {
+          src1 = __shfl(my_src, r + 1); dst1 = __shfl(my_dst, r + 1);
+          h01 = __shfl(my_h0, r + 1); h11 = __shfl(my_h1, r + 1);
+          len1 = __shfl(my_len, r + 1); hdr1 = __shfl(my_hdr, r + 1);
+          const uint8_t* sp1 = (const uint8_t*)(uintptr_t)src1;
+#pragma unroll
+          for (int k = 0; k < PIPE_B; k++)
+            b1[k] = (lane + k * WAVE < len1) ? sp1[lane + k * WAVE] : 0;
+        }
+        uint8_t* w = out + dst0;
+        if (lane < hdr0)
+          w[lane] = (lane < 8) ? (uint8_t)(h00 >> (8 * lane))
+                               : (uint8_t)(h10 >> (8 * (lane - 8)));
+        w += hdr0;
+#pragma unroll
+        for (int k = 0; k < PIPE_B; k++)
+          if (lane + k * WAVE < len0) w[lane + k * WAVE] = b0[k];
+        src0 = src1; dst0 = dst1; h00 = h01; h10 = h11; len0 = len1; hdr0 = hdr1;
+#pragma unroll
+        for (int k = 0; k < PIPE_B; k++) b0[k] = b1[k];
+      }
     } else {
-      for (uint32_t r = 0; r < nvalid; r++) emit_one(r);
+      for (uint32_t r = 0; r < nvalid; r++) {
+        uint64_t src = __shfl(my_src, r);
+        uint64_t dsto = __shfl(my_dst, r);
+        uint64_t h0 = __shfl(my_h0, r);
+        uint64_t h1 = __shfl(my_h1, r);
+        uint32_t len = __shfl(my_len, r);
+        uint32_t hdr = __shfl(my_hdr, r);
+        uint8_t* w = out + dsto;
+        if (lane < hdr)
+          w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
+                               : (uint8_t)(h1 >> (8 * (lane - 8)));
+        w += hdr;
+        const uint8_t* sp = (const uint8_t*)(uintptr_t)src;
+        for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp[b];
+      }
     }
   }
 }
