@@ -1,0 +1,103 @@
+"""Product-side IFile stream parser (KeyValuesReader support).
+
+Restates the reader semantics of IFile.Reader
+(IFile.java:877-1001: positionToNextRecord / RLE & V_END handling) and the
+vint codec (hadoop WritableUtils).  Pure Python — used by the plugin input
+to ingest fetched segments; the hot path (sort/merge/emit) never goes
+through here.
+"""
+import zlib
+
+
+def vint_read(b, pos):
+    first = b[pos]
+    first_s = first - 256 if first >= 128 else first
+    if first_s >= -112:
+        return first_s, pos + 1
+    ln = (-119 - first_s) if first_s < -120 else (-111 - first_s)
+    v = 0
+    for k in range(1, ln):
+        v = (v << 8) | b[pos + k]
+    if first_s < -120:
+        v = ~v
+    return v, pos + ln
+
+
+def vint_write(v):
+    if -112 <= v <= 127:
+        return bytes([v & 0xFF])
+    neg = v < 0
+    if neg:
+        v = ~v
+    nbytes = 0
+    t = v
+    while t:
+        t >>= 8
+        nbytes += 1
+    marker = (-120 - nbytes) if neg else (-112 - nbytes)
+    return bytes([marker & 0xFF]) + v.to_bytes(nbytes, "big")
+
+
+def read_stream(stream, with_header=True, verify_crc=True):
+    """Parse one IFile stream; returns list of (key_ser, val_ser, same_key)."""
+    if with_header:
+        if stream[:3] != b"TIF":
+            raise ValueError("bad IFile magic")
+        if stream[3] != 0:
+            raise ValueError("compressed IFile segments are out of scope (SURVEY §8f)")
+        body = stream[4:-4]
+        if verify_crc and zlib.crc32(stream[4:-4]) != int.from_bytes(stream[-4:], "big"):
+            raise ValueError("IFile CRC mismatch")
+    else:
+        body = stream
+    pos = 0
+    out = []
+    cur_key = b""
+    prev_rle = False
+    while True:
+        if prev_rle:
+            vlen, pos = vint_read(body, pos)
+            if vlen == -3:  # V_END: fresh lengths follow
+                klen, pos = vint_read(body, pos)
+                vlen, pos = vint_read(body, pos)
+            else:
+                klen = -2
+        else:
+            klen, pos = vint_read(body, pos)
+            vlen, pos = vint_read(body, pos)
+        if klen == -1 and vlen == -1:
+            break
+        same = klen == -2
+        if not same:
+            cur_key = bytes(body[pos: pos + klen])
+            pos += klen
+        val = bytes(body[pos: pos + vlen])
+        pos += vlen
+        out.append((cur_key, val, same))
+        prev_rle = same
+    return out
+
+
+def serialize_bytes_writable(content: bytes) -> bytes:
+    return len(content).to_bytes(4, "big") + content
+
+
+def deserialize_bytes_writable(ser: bytes) -> bytes:
+    return ser[4:]
+
+
+def serialize_text(content: bytes) -> bytes:
+    return vint_write(len(content)) + content
+
+
+def deserialize_text(ser: bytes) -> bytes:
+    n, pos = vint_read(ser, 0)
+    return bytes(ser[pos:pos + n])
+
+
+def serialize_int_writable(v: int) -> bytes:
+    return int(v).to_bytes(4, "big", signed=True)
+
+
+def deserialize_int_writable(ser: bytes) -> int:
+    return int.from_bytes(ser[:4], "big", signed=True)
