@@ -960,6 +960,76 @@ __global__ void k_generate_fixed(uint64_t seed, int64_t n, int32_t klen, int32_t
     }
   }
 }
+/* kind 2 (C5, TeraSort-shaped): 10B key + 90B value, range partitions
+ * (TotalOrderPartitioner-style splits on the leading 2 key bytes). */
+__global__ void k_range_partition(const uint8_t* data, uint64_t rec_bytes, int32_t P,
+                                  int32_t* d_part, int64_t n) {
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* k = data + (uint64_t)i * rec_bytes + 4;
+    uint32_t v = ((uint32_t)k[0] << 8) | k[1];
+    d_part[i] = (int32_t)((uint64_t)v * (uint32_t)P >> 16);
+  }
+}
+
+/* kind 1 (C3): Text keys, content = zipf(1.1)-drawn dict word + unique
+ * base-36 record-id suffix, total length in [4,32]; 64B BytesWritable
+ * values.  Lengths are computed twice (len pass + fill pass) from the same
+ * deterministic draw. */
+__device__ __forceinline__ void d_c3_key(uint64_t seed, int64_t i,
+                                         uint8_t* out /* <=32 */, int* out_len) {
+  uint64_t r = d_splitmix64(seed ^ (uint64_t)i * 0x9E3779B97F4A7C15ull);
+  /* zipf-ish rank via inverse tail cdf u^(-1/(s-1)), s=1.1, clamped to 1e6 */
+  double u = ((double)(r >> 11) + 1.0) / 9007199254740992.0;
+  double z = exp(-10.0 * log(u));
+  uint32_t word = (z >= 1e6) ? 999999u : (uint32_t)z;
+  uint64_t wh = d_splitmix64(0xC3C3ull ^ word);
+  int wl = 3 + (int)(wh % 8); /* 3..10 word chars */
+  int n = 0;
+  for (; n < wl; n++) out[n] = 'a' + (uint8_t)((wh >> (5 * (n % 12))) % 26);
+  /* unique suffix: '#' + base36(i) */
+  out[n++] = '#';
+  uint64_t v = (uint64_t)i;
+  uint8_t tmp[14];
+  int t = 0;
+  do { uint64_t d = v % 36; tmp[t++] = (uint8_t)(d < 10 ? '0' + d : 'a' + d - 10); v /= 36; } while (v);
+  while (t && n < 32) out[n++] = tmp[--t];
+  if (n < 4) { while (n < 4) out[n++] = '_'; }
+  *out_len = n;
+}
+__global__ void k_gen_text_lens(uint64_t seed, int64_t n, int32_t vlen, uint64_t* lens) {
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t kb[32];
+    int kl;
+    d_c3_key(seed, i, kb, &kl);
+    /* Text key: 1-byte vint (len<=32) + content; value: 4B BE len + vlen */
+    lens[i] = (uint64_t)(1 + kl) + 4 + (uint64_t)vlen;
+  }
+}
+__global__ void k_gen_text_fill(uint64_t seed, int64_t n, int32_t vlen,
+                                const uint64_t* off, uint8_t* data, uint32_t* klen_arr) {
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t kb[32];
+    int kl;
+    d_c3_key(seed, i, kb, &kl);
+    uint8_t* r = data + off[i];
+    r[0] = (uint8_t)kl; /* vint: 0<len<=32 is a single byte */
+    for (int b = 0; b < kl; b++) r[1 + b] = kb[b];
+    uint8_t* v = r + 1 + kl;
+    v[0] = (uint8_t)(vlen >> 24); v[1] = (uint8_t)(vlen >> 16);
+    v[2] = (uint8_t)(vlen >> 8); v[3] = (uint8_t)vlen;
+    uint64_t sv = d_splitmix64(seed ^ 0x77ull ^ (uint64_t)i);
+    for (int32_t c = 0; c < vlen; c += 8) {
+      uint64_t w = d_splitmix64(sv + (uint64_t)(c >> 3));
+      for (int32_t k2 = 0; k2 < 8 && c + k2 < vlen; k2++)
+        v[4 + c + k2] = (uint8_t)(w >> (8 * k2));
+    }
+    klen_arr[i] = (uint32_t)(1 + kl);
+  }
+}
+
 __global__ void k_fill_fixed_offsets(uint64_t* off, uint32_t* klen_arr, int64_t n,
                                      uint64_t rec_bytes, uint32_t klen_ser) {
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i <= n;
@@ -2004,30 +2074,47 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
                             int32_t vlen, const tzs_conf* conf, void** d_data,
                             uint64_t** d_off, uint32_t** d_klen, int32_t** d_part) {
   if (ensure_device_constants()) return -70;
-  if (kind != 0) FAIL(-22, "generator kind %d not implemented yet", kind);
-  uint64_t rec = 4 + (uint64_t)klen + 4 + (uint64_t)vlen;
+  if (kind < 0 || kind > 2) FAIL(-22, "generator kind %d not implemented", kind);
   void* dd = nullptr;
   uint64_t* doff = nullptr;
   uint32_t* dkl = nullptr;
-  HIP_CHECK(hipMalloc(&dd, rec * n));
   HIP_CHECK(hipMalloc((void**)&doff, 8 * (n + 1)));
   HIP_CHECK(hipMalloc((void**)&dkl, 4 * n));
-  hipLaunchKernelGGL(k_generate_fixed, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n, klen,
-                     vlen, (uint8_t*)dd, rec);
-  hipLaunchKernelGGL(k_fill_fixed_offsets, dim3(grid1d(n + 1)), dim3(BLOCK), 0, 0, doff,
-                     dkl, n, rec, (uint32_t)(4 + klen));
+  uint64_t rec = 0;
+  if (kind == 0 || kind == 2) {
+    rec = 4 + (uint64_t)klen + 4 + (uint64_t)vlen;
+    HIP_CHECK(hipMalloc(&dd, rec * n));
+    hipLaunchKernelGGL(k_generate_fixed, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n,
+                       klen, vlen, (uint8_t*)dd, rec);
+    hipLaunchKernelGGL(k_fill_fixed_offsets, dim3(grid1d(n + 1)), dim3(BLOCK), 0, 0,
+                       doff, dkl, n, rec, (uint32_t)(4 + klen));
+  } else { /* kind 1: Text keys (C3 shape) */
+    hipLaunchKernelGGL(k_gen_text_lens, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n,
+                       vlen, doff);
+    uint64_t total = 0;
+    if (scan_u64(doff, doff, (uint32_t)n, &total)) return -12;
+    HIP_CHECK(hipMemcpy(doff + n, &total, 8, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMalloc(&dd, total ? total : 1));
+    hipLaunchKernelGGL(k_gen_text_fill, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n,
+                       vlen, doff, (uint8_t*)dd, dkl);
+  }
   if (d_part) {
     int32_t* dp = nullptr;
     HIP_CHECK(hipMalloc((void**)&dp, 4 * n));
-    RecTable rt = {};
-    rt.nspills = 1;
-    rt.data[0] = (const uint8_t*)dd;
-    rt.off[0] = doff;
-    rt.klen[0] = dkl;
-    rt.base[0] = 0; rt.base[1] = (uint32_t)n;
-    rt.key_type = conf ? conf->key_type : TZS_KEY_BYTES;
-    hipLaunchKernelGGL(k_hash_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                       conf ? conf->num_partitions : 1, dp, (uint32_t)n);
+    if (kind == 2) {
+      hipLaunchKernelGGL(k_range_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                         (const uint8_t*)dd, rec, conf ? conf->num_partitions : 1, dp, n);
+    } else {
+      RecTable rt = {};
+      rt.nspills = 1;
+      rt.data[0] = (const uint8_t*)dd;
+      rt.off[0] = doff;
+      rt.klen[0] = dkl;
+      rt.base[0] = 0; rt.base[1] = (uint32_t)n;
+      rt.key_type = (kind == 1) ? TZS_KEY_TEXT : (conf ? conf->key_type : TZS_KEY_BYTES);
+      hipLaunchKernelGGL(k_hash_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                         conf ? conf->num_partitions : 1, dp, (uint32_t)n);
+    }
     *d_part = dp;
   }
   HIP_CHECK(hipDeviceSynchronize());
