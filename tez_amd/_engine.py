@@ -65,6 +65,15 @@ def lib():
             f"tez_amd native engine missing: {_LIB} not built. "
             "Run __graft_entry__.build() (hipcc --offload-arch=gfx950). "
             "There is no CPU fallback.")
+    # Load torch (and its bundled libamdhip64, SONAME libamdhip64.so.7) FIRST:
+    # if this .so pulls /opt/rocm's runtime in first, torch later loads a
+    # second HSA runtime in-process and sees zero GPUs.  With torch resident,
+    # the dynamic linker resolves our libamdhip64.so.7 dependency to torch's
+    # already-loaded copy — one runtime for both.
+    try:
+        import torch  # noqa: F401
+    except Exception:
+        pass
     L = ctypes.CDLL(_LIB)
     c = ctypes
     L.tzs_last_error.restype = c.c_char_p
