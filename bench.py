@@ -109,6 +109,9 @@ def main():
     ap.add_argument("--records", type=int, default=100_000_000,
                     help="total records across all ranks (C2 default 1e8)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--force-exchange", action="store_true",
+                    help="run the all-to-all-v exchange + reduce-merge path even "
+                         "at world_size=1 (bench-code validation)")
     ap.add_argument("--traffic-bytes", type=float, default=None,
                     help="PMC-measured HBM bytes per scatter launch (from "
                          "profiles/, rocprofv3 --pmc; null if not passed)")
@@ -125,13 +128,17 @@ def main():
 
     dist = None
     device = None
-    if world > 1:
+    use_exchange = world > 1 or args.force_exchange
+    if use_exchange:
         import torch
         import torch.distributed as tdist
         dist = tdist
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
-        tdist.init_process_group("nccl")
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        if not tdist.is_initialized():
+            tdist.init_process_group("nccl", rank=rank, world_size=world)
 
     n_local = args.records // n_gpus
     conf = tez_amd.make_conf(PARTS)
@@ -147,7 +154,7 @@ def main():
             torch.cuda.synchronize()
 
     def one_step():
-        if world > 1:
+        if use_exchange:
             return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
         return run_step_single(tez_amd, conf, d, off, kl, n_local)
 

@@ -1,0 +1,127 @@
+"""Full-BASELINE-size property checks on GPU (at sizes the
+oracle cannot replay, verify size-independent invariants): the C2 workload at
+its real 1e8-record size — accounting identities, CRC verification, sorted
+order, and partition placement on sampled segments."""
+import zlib
+
+import pytest
+
+import oracle as o
+
+pytestmark = pytest.mark.gpu
+
+N = 100_000_000
+P = 64
+KLEN, VLEN = 16, 64
+REC = 4 + KLEN + 4 + VLEN
+
+
+@pytest.fixture(scope="module")
+def engine():
+    import __graft_entry__
+    __graft_entry__.build()
+    import tez_amd
+    if not tez_amd.device_available():
+        pytest.skip("no GPU")
+    return tez_amd
+
+
+def bounded_parse(body, max_records):
+    """Parse the first max_records of an IFile body (after header, no CRC)."""
+    from tez_amd import ifile
+    pos = 0
+    recs = []
+    while len(recs) < max_records:
+        klen, pos = ifile.vint_read(body, pos)
+        vlen, pos = ifile.vint_read(body, pos)
+        if klen == -1 and vlen == -1:
+            break
+        assert klen >= 0 and vlen >= 0, "unexpected RLE in unique-key C2"
+        recs.append((bytes(body[pos: pos + klen]),
+                     bytes(body[pos + klen: pos + klen + vlen])))
+        pos += klen + vlen
+    return recs
+
+
+def test_c2_full_size_properties(engine):
+    conf = engine.make_conf(P)
+    d, off, kl, part = engine.generate(seed=0x7E2C2, n=N, kind=0, klen=KLEN,
+                                       vlen=VLEN, conf=conf)
+    s = engine.Sorter(conf)
+    s.write_batch_device(d, off, kl, None, N)
+    engine.free_device(d, off, kl, part)
+    s.flush()
+    ctr = s.counters()
+    assert ctr["output_records"] == N
+    assert ctr["output_bytes"] == N * REC
+    dev, total, idx = s.output_meta()
+
+    # a9 accounting identities over the whole index
+    cursor = 0
+    total_raw = 0
+    for st, raw, cl in idx:
+        assert st == cursor
+        if cl:
+            assert cl == raw + 4          # uncompressed partLength = raw + CRC
+            cursor += cl
+        else:
+            assert raw == 0
+        total_raw += raw
+    assert cursor == total
+    assert ctr["output_bytes_with_overhead"] == total_raw
+    # every byte of payload appears exactly once: sum of (raw - framing) =
+    # records * (rec + 2 vints); framing per record = 2 vints (1B each at C2
+    # sizes), header 4 + EOF 2 per partition
+    nonempty = sum(1 for _s, r, c in idx if c)
+    assert total_raw == N * (REC + 2) + nonempty * 6
+
+    # deep-verify two partitions: CRC + bounded sorted-order/placement parse
+    checked = 0
+    for p, (st, raw, cl) in enumerate(idx):
+        if cl == 0 or checked >= 2:
+            continue
+        seg = engine.read_device(dev, st, cl)
+        assert seg[:4] == b"TIF\x00"
+        assert zlib.crc32(seg[4:-4]) == int.from_bytes(seg[-4:], "big")
+        recs = bounded_parse(seg[4:-4], 50_000)
+        assert len(recs) == 50_000
+        prev = None
+        for kk, vv in recs:
+            assert len(kk) == 4 + KLEN and len(vv) == 4 + VLEN
+            content = kk[4:]
+            assert (o.hash_bytes(content) & 0x7FFFFFFF) % P == p
+            if prev is not None:
+                assert prev < kk  # unique keys: strictly increasing
+            prev = kk
+        checked += 1
+    assert checked == 2
+    s.close()
+
+
+def test_merge_segments_cabi(engine):
+    """tzs_merge_segments over two device-resident columnar segments."""
+    import random
+    rng = random.Random(3)
+    import numpy as np
+
+    def seg(nrec, seed):
+        r = random.Random(seed)
+        pairs = [(o.serialize_bytes_writable(bytes(r.randrange(256) for _ in range(8))),
+                  o.serialize_bytes_writable(b"v%04d" % i)) for i in range(nrec)]
+        data, off, klen = o.build_records(pairs)
+        d, ofp, kp, _ = engine.upload_records(data.tobytes(), off, klen)
+        return pairs, (d, ofp, kp, nrec)
+
+    p1, s1 = seg(500, 1)
+    p2, s2 = seg(400, 2)
+    conf = engine.make_conf(1)
+    data, rec = engine.merge_segments(conf, [s1, s2])
+    st, raw, cl = rec
+    assert cl == len(data)
+    got = o.ifile_read(data, with_header=True)
+    # oracle merge of the same two segments
+    allp = p1 + p2
+    d2, o2, k2 = o.build_records(allp)
+    want = o.spill(d2, o2, k2, 1, partitions=np.zeros(len(allp), dtype=np.int32))
+    widx = o.index_decode(want["index"], 1)
+    assert data == want["data"][: widx[0][2]]

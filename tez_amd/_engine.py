@@ -40,6 +40,11 @@ class TzsIndexRecord(ctypes.Structure):
                 ("part_length", ctypes.c_int64)]
 
 
+class TzsSegment(ctypes.Structure):
+    _fields_ = [("d_data", ctypes.c_void_p), ("d_off", ctypes.c_void_p),
+                ("d_klen", ctypes.c_void_p), ("n", ctypes.c_int64)]
+
+
 class TzsCounters(ctypes.Structure):
     _fields_ = [(n, ctypes.c_int64) for n in
                 ("output_records", "output_bytes", "output_bytes_with_overhead",
@@ -97,6 +102,9 @@ def lib():
                                              c.POINTER(c.c_uint64),
                                              c.POINTER(c.c_uint64)]
     L.tzs_memcpy_d2d.argtypes = [c.c_void_p, c.c_void_p, c.c_uint64]
+    L.tzs_merge_segments.argtypes = [c.POINTER(TzsConf), c.POINTER(TzsSegment),
+                                     c.c_int32, c.POINTER(c.c_void_p),
+                                     c.POINTER(c.c_int64), c.POINTER(TzsIndexRecord)]
     L.tzs_sorter_counters.argtypes = [c.c_void_p, c.POINTER(TzsCounters)]
     L.tzs_sorter_times.argtypes = [c.c_void_p, c.POINTER(TzsTimes)]
     L.tzs_sorter_close.argtypes = [c.c_void_p]
@@ -182,6 +190,15 @@ class Sorter:
             ba = (ctypes.c_char * n.value).from_buffer(buf)
             _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba), p, n.value), "d2h")
         return bytes(buf), [(r.start_offset, r.raw_length, r.part_length) for r in idx]
+
+    def output_meta(self):
+        """(device_ptr, nbytes, index) without copying the stream to host."""
+        p = ctypes.c_void_p()
+        n = ctypes.c_int64()
+        idx = (TzsIndexRecord * self.conf.num_partitions)()
+        _ck(lib().tzs_sorter_output(self.h, ctypes.byref(p), ctypes.byref(n), idx),
+            "output")
+        return p, n.value, [(r.start_offset, r.raw_length, r.part_length) for r in idx]
 
     def sorted_columnar(self):
         """Final sort as device columnar arrays + partition ranges
@@ -269,6 +286,38 @@ def free_device(*ptrs):
     for p in ptrs:
         if p:
             lib().tzs_free_device(p)
+
+
+def read_device(dev_ptr, offset, nbytes):
+    """Copy [offset, offset+nbytes) of a device buffer to host bytes."""
+    buf = bytearray(nbytes)
+    if nbytes:
+        ba = (ctypes.c_char * nbytes).from_buffer(buf)
+        _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba),
+                                 ctypes.c_void_p((dev_ptr.value if hasattr(dev_ptr, "value")
+                                                  else dev_ptr) + offset), nbytes), "d2h")
+    return bytes(buf)
+
+
+def merge_segments(conf, segments):
+    """tzs_merge_segments: segments = list of (d_data, d_off, d_klen, n)
+    device-pointer tuples (one partition).  Returns (bytes, (start,raw,part))."""
+    n = len(segments)
+    arr = (TzsSegment * max(n, 1))()
+    for i, (d, o, k, cnt) in enumerate(segments):
+        arr[i].d_data = d if isinstance(d, int) else d.value
+        arr[i].d_off = o if isinstance(o, int) else o.value
+        arr[i].d_klen = k if isinstance(k, int) else k.value
+        arr[i].n = cnt
+    out = ctypes.c_void_p()
+    ln = ctypes.c_int64()
+    rec = TzsIndexRecord()
+    _ck(lib().tzs_merge_segments(ctypes.byref(conf), arr, n, ctypes.byref(out),
+                                 ctypes.byref(ln), ctypes.byref(rec)), "merge_segments")
+    data = read_device(out, 0, ln.value)
+    if out.value:
+        lib().tzs_free_device(out)
+    return data, (rec.start_offset, rec.raw_length, rec.part_length)
 
 
 def device_available():

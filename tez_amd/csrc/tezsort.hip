@@ -2130,8 +2130,40 @@ extern "C" void tzs_free_device(void* p) { if (p) (void)hipFree(p); }
 extern "C" int tzs_merge_segments(const tzs_conf* conf, const tzs_segment* segs,
                                   int32_t nsegs, void** d_out, int64_t* out_bytes,
                                   tzs_index_record* rec) {
-  (void)conf; (void)segs; (void)nsegs; (void)d_out; (void)out_bytes; (void)rec;
-  FAIL(-38, "tzs_merge_segments: implemented via tzs_sorter flush path in round 1");
+  /* Reduce-side k-way merge of sorted columnar segments of ONE partition
+   * (TezMerger.MergeQueue restated as a stable re-sort — DESIGN.md §4).
+   * Each segment is absorbed (appended + offset-rebased) and flush() runs
+   * the union sort + IFile emit. */
+  if (!conf || nsegs < 0) FAIL(-22, "bad args");
+  tzs_conf c1 = *conf;
+  c1.num_partitions = 1;
+  tzs_sorter* s = nullptr;
+  int rc2 = tzs_sorter_create(&c1, &s);
+  if (rc2) return rc2;
+  for (int i = 0; i < nsegs; i++) {
+    rc2 = tzs_sorter_write_batch_device(s, segs[i].d_data, segs[i].d_off,
+                                        segs[i].d_klen, nullptr, segs[i].n);
+    if (rc2) { tzs_sorter_close(s); return rc2; }
+  }
+  rc2 = tzs_sorter_flush(s);
+  if (rc2) { tzs_sorter_close(s); return rc2; }
+  const void* db = nullptr;
+  int64_t nb = 0;
+  tzs_index_record ix;
+  rc2 = tzs_sorter_output(s, &db, &nb, &ix);
+  if (rc2) { tzs_sorter_close(s); return rc2; }
+  void* own = nullptr;
+  if (nb) {
+    if (hipMalloc(&own, (size_t)nb) != hipSuccess) { tzs_sorter_close(s); FAIL(-12, "oom"); }
+    if (hipMemcpy(own, db, (size_t)nb, hipMemcpyDeviceToDevice) != hipSuccess) {
+      (void)hipFree(own); tzs_sorter_close(s); FAIL(-70, "copy");
+    }
+  }
+  *d_out = own;
+  *out_bytes = nb;
+  if (rec) *rec = ix;
+  tzs_sorter_close(s);
+  return 0;
 }
 
 /* ---- misc helpers for the Python layer ---- */
