@@ -179,6 +179,11 @@ struct RecTable {
   const uint64_t* off[MAX_SPILLS];
   const uint32_t* klen[MAX_SPILLS];
   uint32_t base[MAX_SPILLS + 1]; /* global id g in [base[s], base[s+1]) */
+  /* uniform-record fast path: when every record of spill s has the same
+     serialized size/klen (C2/C5 shapes), offsets become arithmetic and the
+     off/klen gathers (PMC: 12.7 GB per descriptor pass at n=1e8) vanish. */
+  uint32_t rec_u[MAX_SPILLS];   /* record bytes, 0 = non-uniform */
+  uint32_t klen_u[MAX_SPILLS];
   int32_t nspills;
   int32_t key_type; /* 0 bytes, 1 text */
 };
@@ -200,11 +205,18 @@ __device__ __forceinline__ RecView rt_view(const RecTable& rt, uint32_t g) {
   int s = rt_spill_of(rt, g);
   uint32_t r = g - rt.base[s];
   RecView v;
-  uint64_t o = rt.off[s][r];
-  v.key = rt.data[s] + o;
-  v.klen = rt.klen[s][r];
-  v.val = v.key + v.klen;
-  v.vlen = (uint32_t)(rt.off[s][r + 1] - o - v.klen);
+  if (rt.rec_u[s]) {
+    v.key = rt.data[s] + (uint64_t)r * rt.rec_u[s];
+    v.klen = rt.klen_u[s];
+    v.val = v.key + v.klen;
+    v.vlen = rt.rec_u[s] - v.klen;
+  } else {
+    uint64_t o = rt.off[s][r];
+    v.key = rt.data[s] + o;
+    v.klen = rt.klen[s][r];
+    v.val = v.key + v.klen;
+    v.vlen = (uint32_t)(rt.off[s][r + 1] - o - v.klen);
+  }
   if (rt.key_type == 1) {
     int n = d_vint_decoded_size((int8_t)v.key[0]);
     v.content = v.key + n;
@@ -226,6 +238,8 @@ __global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* 
 __global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
                                    const uint8_t* same, uint64_t total_body,
                                    uint32_t n, int P, uint64_t* out);
+__global__ void k_check_uniform(const uint64_t* off, const uint32_t* klen, int64_t n,
+                                uint64_t* mm /* {min_diff,max_diff,min_klen,max_klen} */);
 
 /* ---- partition + composite ---- */
 __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32_t n) {
@@ -240,17 +254,22 @@ __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32
  * masked to the top sort_bytes bytes: the radix sort only orders those
  * (adaptive pass count — DESIGN.md §4); rarer-than-1% ties are resolved by
  * the refinement levels, whose equality test must see the same mask. */
-__global__ void k_build_composite(RecTable rt, const int32_t* d_part, int pbits,
-                                  int sort_bytes, uint64_t* d_key, uint32_t* d_idx,
-                                  uint32_t n) {
+__global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
+                                  int pbits, int sort_bytes, uint64_t* d_key,
+                                  uint32_t* d_idx, uint32_t n) {
+  /* d_part == nullptr: compute the HashPartitioner placement here (fused —
+     a separate hash kernel re-reads all content bytes, ~10 GB at n=1e8). */
   uint64_t mask = (sort_bytes >= 8) ? ~0ull : ~0ull << (8 * (8 - sort_bytes));
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, i);
+    uint32_t part;
+    if (d_part) part = (uint32_t)d_part[i];
+    else part = (uint32_t)((d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P);
     uint64_t c = 0;
     uint32_t m = v.clen < 8 ? v.clen : 8;
     for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
-    uint64_t key = pbits ? (((uint64_t)(uint32_t)d_part[i] << (64 - pbits)) | (c >> pbits)) : c;
+    uint64_t key = pbits ? (((uint64_t)part << (64 - pbits)) | (c >> pbits)) : c;
     d_key[i] = key & mask;
     d_idx[i] = i;
   }
@@ -1262,6 +1281,7 @@ struct SpillData {
   DBuf klen;   /* u32 [n] */
   int64_t n = 0;
   uint8_t rle = 0;
+  uint32_t rec_u = 0, klen_u = 0;
   /* emitted IFile bytes + host index */
   DBuf ifile;
   int64_t ifile_len = 0;
@@ -1278,6 +1298,10 @@ struct tzs_sorter {
   DBuf cur_data, cur_off, cur_klen, cur_part;
   int64_t cur_n = 0;
   uint64_t cur_bytes = 0;
+  uint32_t cur_rec_u = 0;   /* uniform record bytes (0 = not uniform) */
+  uint32_t cur_klen_u = 0;
+  bool cur_first_batch = true;
+  bool have_explicit_parts = false;
   /* host-path staging */
   std::vector<uint8_t> host_data;
   std::vector<uint64_t> host_off;
@@ -1405,25 +1429,36 @@ extern "C" int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
   std::swap(s->cur_data, nd); nd.release();
   std::swap(s->cur_off, noff); noff.release();
   std::swap(s->cur_klen, nkl); nkl.release();
-  /* partitions */
+  /* partitions: explicit ones are stored; HashPartitioner placement is
+     computed fused into the composite build at sort time */
   if (d_part) {
     HIP_CHECK(hipMemcpyAsync((int32_t*)npart.p + old_n, d_part, sizeof(int32_t) * n,
                              hipMemcpyDeviceToDevice));
-  } else {
-    RecTable rt = {};
-    rt.nspills = 1;
-    rt.data[0] = (const uint8_t*)s->cur_data.p;
-    rt.off[0] = (const uint64_t*)s->cur_off.p;
-    rt.klen[0] = (const uint32_t*)s->cur_klen.p;
-    rt.base[0] = 0; rt.base[1] = (uint32_t)(old_n + n);
-    rt.key_type = s->conf.key_type;
-    /* note: offsets for new records are at index old_n.. — RecTable spans the
-       whole buffer; we only fill new records */
-    hipLaunchKernelGGL(k_hash_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                       s->conf.num_partitions, (int32_t*)npart.p, (uint32_t)(old_n + n));
-    /* (recomputes all; fine: idempotent) */
+    s->have_explicit_parts = true;
+  } else if (s->have_explicit_parts) {
+    FAIL(-22, "cannot mix explicit and computed partitions in one sorter");
   }
   std::swap(s->cur_part, npart); npart.release();
+  {
+    static thread_local DBuf mm;
+    if (mm.alloc(32)) return -12;
+    uint64_t init[4] = {~0ull, 0, ~0ull, 0};
+    HIP_CHECK(hipMemcpyAsync(mm.p, init, 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_check_uniform, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                       d_off, d_klen, n, (uint64_t*)mm.p);
+    uint64_t res[4];
+    HIP_CHECK(hipMemcpy(res, mm.p, 32, hipMemcpyDeviceToHost));
+    bool uni = (res[0] == res[1]) && (res[2] == res[3]) && res[0] <= 0xFFFFFFFFull;
+    if (s->cur_first_batch) {
+      s->cur_rec_u = uni ? (uint32_t)res[0] : 0;
+      s->cur_klen_u = uni ? (uint32_t)res[2] : 0;
+      s->cur_first_batch = false;
+    } else if (!uni || s->cur_rec_u != (uint32_t)res[0] ||
+               s->cur_klen_u != (uint32_t)res[2]) {
+      s->cur_rec_u = 0;
+      s->cur_klen_u = 0;
+    }
+  }
   s->cur_n = old_n + n;
   s->cur_bytes = old_bytes + nbytes;
   s->ctr.output_records += n;
@@ -1466,7 +1501,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   uint64_t* d_key = (uint64_t*)s->skey.p;
   uint32_t* d_idx = (uint32_t*)s->sidx.p;
   hipLaunchKernelGGL(k_build_composite, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                     d_part_unsorted, pbits, SB, d_key, d_idx, n);
+                     d_part_unsorted, P, pbits, SB, d_key, d_idx, n);
   (void)hipEventRecord(ev[1]);
 
   /* 2. base radix sort over the top SB bytes of the composite */
@@ -1794,6 +1829,22 @@ __global__ void k_part_hist(const uint32_t* parts, uint32_t n, uint32_t* counts,
   for (int j = threadIdx.x; j < P; j += blockDim.x)
     if (lh[j]) atomicAdd(&counts[j], lh[j]);
 }
+__global__ void k_check_uniform(const uint64_t* off, const uint32_t* klen, int64_t n,
+                                uint64_t* mm) {
+  uint64_t dmin = ~0ull, dmax = 0, kmin = ~0ull, kmax = 0;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t d = off[i + 1] - off[i];
+    dmin = min(dmin, d); dmax = max(dmax, d);
+    uint64_t k = klen[i];
+    kmin = min(kmin, k); kmax = max(kmax, k);
+  }
+  atomicMin((unsigned long long*)&mm[0], (unsigned long long)dmin);
+  atomicMax((unsigned long long*)&mm[1], (unsigned long long)dmax);
+  atomicMin((unsigned long long*)&mm[2], (unsigned long long)kmin);
+  atomicMax((unsigned long long*)&mm[3], (unsigned long long)kmax);
+}
+
 __global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
                                    const uint8_t* same, uint64_t total_body,
                                    uint32_t n, int P, uint64_t* out) {
@@ -1838,16 +1889,24 @@ extern "C" int tzs_sorter_spill(tzs_sorter* s) {
   std::swap(sp->off, s->cur_off);
   std::swap(sp->klen, s->cur_klen);
   sp->n = s->cur_n;
+  sp->rec_u = s->cur_rec_u;
+  sp->klen_u = s->cur_klen_u;
+  s->cur_first_batch = true;
+  s->cur_rec_u = 0;
+  s->cur_klen_u = 0;
   RecTable rt = {};
   rt.nspills = 1;
   rt.data[0] = (const uint8_t*)sp->data.p;
   rt.off[0] = (const uint64_t*)sp->off.p;
   rt.klen[0] = (const uint32_t*)sp->klen.p;
+  rt.rec_u[0] = sp->rec_u;
+  rt.klen_u[0] = sp->klen_u;
   rt.base[0] = 0; rt.base[1] = (uint32_t)sp->n;
   rt.key_type = s->conf.key_type;
   uint8_t dummy_rle = 0;
-  rc = sort_and_emit(s, rt, (uint32_t)sp->n, (const int32_t*)s->cur_part.p, &dummy_rle,
-                     1, sp);
+  rc = sort_and_emit(s, rt, (uint32_t)sp->n,
+                     s->have_explicit_parts ? (const int32_t*)s->cur_part.p : nullptr,
+                     &dummy_rle, 1, sp);
   s->cur_part.release();
   s->cur_n = 0;
   s->cur_bytes = 0;
@@ -1925,23 +1984,21 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
     rt.data[i] = (const uint8_t*)sp->data.p;
     rt.off[i] = (const uint64_t*)sp->off.p;
     rt.klen[i] = (const uint32_t*)sp->klen.p;
+    rt.rec_u[i] = sp->rec_u;
+    rt.klen_u[i] = sp->klen_u;
     rt.base[i] = (uint32_t)total_n;
     total_n += (uint64_t)sp->n;
     spill_rle[i] = sp->rle;
   }
   rt.base[nsp] = (uint32_t)total_n;
   if (total_n >> 32) FAIL(-22, "too many records for final merge");
-  /* partitions for the union: recompute on device (cheap vs sort) */
-  static thread_local DBuf d_part_union;
-  if (d_part_union.alloc(sizeof(int32_t) * (total_n ? total_n : 1))) return -12;
-  hipLaunchKernelGGL(k_hash_partition, dim3(grid1d(total_n)), dim3(BLOCK), 0, 0, rt,
-                     s->conf.num_partitions, (int32_t*)d_part_union.p, (uint32_t)total_n);
-  /* NOTE: if the caller supplied explicit partitions, they were used for the
-     per-spill sort but are not retained; recomputing assumes HashPartitioner.
-     Explicit-partition users must keep final merge to 1 spill for now (round-1
-     limitation, documented). */
+  /* partitions for the union are recomputed inside the composite build.
+     NOTE: explicit-partition users must keep final merge to 1 spill (the
+     per-record placement is not retained across spills — round-1 limit). */
+  if (s->have_explicit_parts)
+    FAIL(-22, "explicit partitions require a single spill in round 1");
   SpillData finalsp;
-  rc = sort_and_emit(s, rt, (uint32_t)total_n, (const int32_t*)d_part_union.p,
+  rc = sort_and_emit(s, rt, (uint32_t)total_n, nullptr,
                      spill_rle.data(), nsp, &finalsp);
   if (rc) return rc;
   s->final_index = finalsp.index;
