@@ -695,46 +695,41 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       if (o > maxlen) maxlen = o;
     }
     if (nvalid == WAVE && maxlen <= WAVE * PIPE_B) {
-      /* Pipelined: issue record r+DEPTH-1's gather while storing record r.
-         Buffers are parity-indexed (NOT rotated — a register rotate makes
-         the compiler wait vmcnt on the freshly issued loads every iteration,
-         which measured as 67% parked waves). */
-      constexpr int DEPTH = 4;
-      uint64_t psrc[DEPTH], pdst[DEPTH], ph0[DEPTH], ph1[DEPTH];
-      uint32_t plen[DEPTH], phdr[DEPTH];
-      uint8_t pb[DEPTH][PIPE_B];
+      /* 2-deep pipeline: issue record r+1's gather while storing record r —
+         the plain loop serializes on one load->store chain per record and
+         runs at ~10% duty cycle */
+      uint64_t src0 = __shfl(my_src, 0), dst0 = __shfl(my_dst, 0);
+      uint64_t h00 = __shfl(my_h0, 0), h10 = __shfl(my_h1, 0);
+      uint32_t len0 = __shfl(my_len, 0), hdr0 = __shfl(my_hdr, 0);
+      uint8_t b0[PIPE_B];
+      const uint8_t* sp0 = (const uint8_t*)(uintptr_t)src0;
 #pragma unroll
-      for (int q = 0; q < DEPTH - 1; q++) {
-        psrc[q] = __shfl(my_src, q); pdst[q] = __shfl(my_dst, q);
-        ph0[q] = __shfl(my_h0, q); ph1[q] = __shfl(my_h1, q);
-        plen[q] = __shfl(my_len, q); phdr[q] = __shfl(my_hdr, q);
-        const uint8_t* sp = (const uint8_t*)(uintptr_t)psrc[q];
-#pragma unroll
-        for (int k = 0; k < PIPE_B; k++)
-          pb[q][k] = (lane + k * WAVE < plen[q]) ? sp[lane + k * WAVE] : 0;
-      }
-#pragma unroll 4
+      for (int k = 0; k < PIPE_B; k++)
+        b0[k] = (lane + k * WAVE < len0) ? sp0[lane + k * WAVE] : 0;
       for (uint32_t r = 0; r < WAVE; r++) {
-        const int cur = r % DEPTH;
-        const int nxt = (r + DEPTH - 1) % DEPTH;
-        uint32_t rn = r + DEPTH - 1;
-        if (rn < WAVE) {
-          psrc[nxt] = __shfl(my_src, rn); pdst[nxt] = __shfl(my_dst, rn);
-          ph0[nxt] = __shfl(my_h0, rn); ph1[nxt] = __shfl(my_h1, rn);
-          plen[nxt] = __shfl(my_len, rn); phdr[nxt] = __shfl(my_hdr, rn);
-          const uint8_t* sp = (const uint8_t*)(uintptr_t)psrc[nxt];
+        uint64_t src1 = 0, dst1 = 0, h01 = 0, h11 = 0;
+        uint32_t len1 = 0, hdr1 = 0;
+        uint8_t b1[PIPE_B] = {0};
+        if (r + 1 < WAVE) {
+          src1 = __shfl(my_src, r + 1); dst1 = __shfl(my_dst, r + 1);
+          h01 = __shfl(my_h0, r + 1); h11 = __shfl(my_h1, r + 1);
+          len1 = __shfl(my_len, r + 1); hdr1 = __shfl(my_hdr, r + 1);
+          const uint8_t* sp1 = (const uint8_t*)(uintptr_t)src1;
 #pragma unroll
           for (int k = 0; k < PIPE_B; k++)
-            pb[nxt][k] = (lane + k * WAVE < plen[nxt]) ? sp[lane + k * WAVE] : 0;
+            b1[k] = (lane + k * WAVE < len1) ? sp1[lane + k * WAVE] : 0;
         }
-        uint8_t* w = out + pdst[cur];
-        if (lane < phdr[cur])
-          w[lane] = (lane < 8) ? (uint8_t)(ph0[cur] >> (8 * lane))
-                               : (uint8_t)(ph1[cur] >> (8 * (lane - 8)));
-        w += phdr[cur];
+        uint8_t* w = out + dst0;
+        if (lane < hdr0)
+          w[lane] = (lane < 8) ? (uint8_t)(h00 >> (8 * lane))
+                               : (uint8_t)(h10 >> (8 * (lane - 8)));
+        w += hdr0;
 #pragma unroll
         for (int k = 0; k < PIPE_B; k++)
-          if (lane + k * WAVE < plen[cur]) w[lane + k * WAVE] = pb[cur][k];
+          if (lane + k * WAVE < len0) w[lane + k * WAVE] = b0[k];
+        src0 = src1; dst0 = dst1; h00 = h01; h10 = h11; len0 = len1; hdr0 = hdr1;
+#pragma unroll
+        for (int k = 0; k < PIPE_B; k++) b0[k] = b1[k];
       }
     } else {
       for (uint32_t r = 0; r < nvalid; r++) {
