@@ -803,7 +803,7 @@ __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
  * 256-byte chunk from LDS with slice-by-4 tables (4 lookups per word, 1/4 the
  * dependent-chain length of bytewise).  Rows padded one word so lane t's
  * reads land on distinct banks. */
-#define CRC_SC_CHUNKS 128
+#define CRC_SC_CHUNKS 64
 #define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 32 KiB */
 #define CRC_ROW_W (CRC_CHUNK / 4 + 1)            /* 65 words per chunk row */
 __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
@@ -1199,9 +1199,13 @@ static thread_local int64_t g_scatter_elems = 0;
 
 /* stable LSD radix over KeyT with payload arrays. Sorts in place (ping-pong,
  * result left in the primary arrays). */
+/* dbuf_key/dbuf_a0/dbuf_a1 may be passed so an odd pass count SWAPS the
+ * buffer handles with the ping-pong temps instead of copying ~1.2 GB back. */
 template <typename KeyT>
 static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
-                      int nbytes_key, int first_byte = 0) {
+                      int nbytes_key, int first_byte = 0,
+                      DBuf* dbuf_key = nullptr, DBuf* dbuf_a0 = nullptr,
+                      DBuf* dbuf_a1 = nullptr) {
   if (n <= 1) return 0;
   uint32_t nb = nblocks_for(n, TILE);
   static thread_local DBuf counts, offsets, totals, bases, tk64, tk32, ta0, ta1;
@@ -1257,11 +1261,18 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     (void)hipEventDestroy(evs[e]); (void)hipEventDestroy(eve[e]);
   }
   if (passes & 1) {
-    /* result is in the temp arrays: copy back */
-    HIP_CHECK(hipMemcpyAsync(d_key, kin, sizeof(KeyT) * n, hipMemcpyDeviceToDevice));
-    HIP_CHECK(hipMemcpyAsync(d_a0, a0in, sizeof(uint32_t) * n, hipMemcpyDeviceToDevice));
-    if (has_a1)
-      HIP_CHECK(hipMemcpyAsync(d_a1, a1in, sizeof(uint32_t) * n, hipMemcpyDeviceToDevice));
+    if (dbuf_key && dbuf_a0 && (!has_a1 || dbuf_a1)) {
+      /* hand the temp buffers to the caller; keep the old primaries as temps */
+      std::swap(*dbuf_key, tk);
+      std::swap(*dbuf_a0, ta0);
+      if (has_a1) std::swap(*dbuf_a1, ta1);
+    } else {
+      HIP_CHECK(hipMemcpyAsync(d_key, kin, sizeof(KeyT) * n, hipMemcpyDeviceToDevice));
+      HIP_CHECK(hipMemcpyAsync(d_a0, a0in, sizeof(uint32_t) * n, hipMemcpyDeviceToDevice));
+      if (has_a1)
+        HIP_CHECK(hipMemcpyAsync(d_a1, a1in, sizeof(uint32_t) * n,
+                                 hipMemcpyDeviceToDevice));
+    }
   }
   return 0;
 }
@@ -1505,8 +1516,11 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   (void)hipEventRecord(ev[1]);
 
   /* 2. base radix sort over the top SB bytes of the composite */
-  int rc = radix_sort<uint64_t>(d_key, d_idx, nullptr, n, 8, 8 - SB);
+  int rc = radix_sort<uint64_t>(d_key, d_idx, nullptr, n, 8, 8 - SB,
+                                &s->skey, &s->sidx, nullptr);
   if (rc) return rc;
+  d_key = (uint64_t*)s->skey.p;
+  d_idx = (uint32_t*)s->sidx.p;
   (void)hipEventRecord(ev[2]);
 
   /* 3. refinement levels */
