@@ -78,6 +78,11 @@ def test_c3_text_zipf_multi_spill(engine):
         engine.free_device(d, off, kl, part)
         spills.append(o.spill(data, offs, klen, P, key_type=o.KEY_TEXT,
                               comparator=o.CMP_TEXT))
+        # per-spill segment view (pipelined shuffle):
+        # spill files must byte-match the oracle spill too
+        sdata, sidx = s.spill_output(k)
+        assert sdata == spills[-1]["data"]
+        assert sidx == o.index_decode(spills[-1]["index"], P)
     s.flush()
     got, gidx = s.output()
     ctr = s.counters()
