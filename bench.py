@@ -112,9 +112,12 @@ def main():
     ap.add_argument("--force-exchange", action="store_true",
                     help="run the all-to-all-v exchange + reduce-merge path even "
                          "at world_size=1 (bench-code validation)")
-    ap.add_argument("--traffic-bytes", type=float, default=None,
-                    help="PMC-measured HBM bytes per scatter launch (from "
-                         "profiles/, rocprofv3 --pmc; null if not passed)")
+    ap.add_argument("--traffic-bytes", type=float, default=3.10e9,
+                    help="PMC-measured HBM bytes per dominant-kernel (radix "
+                         "scatter) launch at the default 1e8-record workload: "
+                         "2xFETCH_SIZE + WRITE_SIZE per rocprofv3 --pmc with "
+                         "the gfx950 FETCH calibration (profiles/"
+                         "r01_pmc_final.txt). Pass 0 to report null.")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -191,7 +194,9 @@ def main():
             "peak": 8000.0,
             "unit": "GB/s",
             "frac": round(achieved_gbps / 8000.0, 4),
-            "traffic": args.traffic_bytes,
+            "traffic": (args.traffic_bytes
+                        if args.traffic_bytes and args.records == 100_000_000
+                        and world == 1 else None),
         }
         cpu = None
         if not args.skip_cpu_baseline and world == 1:
