@@ -783,7 +783,7 @@ __global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
  * reduces its chunk list sequentially with the 256-byte shift matrix. */
 #define CRC_CHUNK 256
 __constant__ uint32_t c_crc_table[256];
-__constant__ uint32_t c_crc_table8[8][256]; /* slice-by-8: T0 = standard */
+__constant__ uint32_t c_crc_table4[4][256]; /* slice-by-4: T0 = standard */
 __constant__ uint32_t c_crc_mats[CRC_MATS][32];
 
 __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
@@ -803,17 +803,17 @@ __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
  * 256-byte chunk from LDS with slice-by-4 tables (4 lookups per word, 1/4 the
  * dependent-chain length of bytewise).  Rows padded one word so lane t's
  * reads land on distinct banks. */
-#define CRC_SC_CHUNKS 256
+#define CRC_SC_CHUNKS 64
 #define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 32 KiB */
 #define CRC_ROW_W (CRC_CHUNK / 4 + 1)            /* 65 words per chunk row */
 __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     const uint8_t* stream, const uint64_t* range_start, const uint64_t* range_len,
     const uint64_t* chunk_base, const uint64_t* sc_base /* [P+1] */,
     uint32_t nparts, uint32_t total_sc, uint32_t* chunk_crc) {
-  __shared__ uint32_t tab8[8][256];
+  __shared__ uint32_t tab4[4][256];
   __shared__ uint32_t stage32[CRC_SC_CHUNKS * CRC_ROW_W];
-  for (int i = threadIdx.x; i < 2048; i += blockDim.x)
-    ((uint32_t*)tab8)[i] = ((const uint32_t*)c_crc_table8)[i];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x)
+    ((uint32_t*)tab4)[i] = ((const uint32_t*)c_crc_table4)[i];
   for (uint32_t sc = blockIdx.x; sc < total_sc; sc += gridDim.x) {
     uint32_t lo = 0, hi = nparts;
     while (lo + 1 < hi) {
@@ -843,18 +843,15 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
       if (len > CRC_CHUNK) len = CRC_CHUNK;
       uint32_t crc = 0xFFFFFFFFu;
       const uint32_t* row = stage32 + t * CRC_ROW_W;
-      uint32_t full8 = (uint32_t)(len / 8);
-      for (uint32_t w = 0; w < full8; w++) {
-        uint32_t lo = crc ^ row[2 * w];
-        uint32_t hi = row[2 * w + 1];
-        crc = tab8[7][lo & 0xFF] ^ tab8[6][(lo >> 8) & 0xFF]
-            ^ tab8[5][(lo >> 16) & 0xFF] ^ tab8[4][lo >> 24]
-            ^ tab8[3][hi & 0xFF] ^ tab8[2][(hi >> 8) & 0xFF]
-            ^ tab8[1][(hi >> 16) & 0xFF] ^ tab8[0][hi >> 24];
+      uint32_t full = (uint32_t)(len / 4);
+      for (uint32_t w = 0; w < full; w++) {
+        uint32_t c = crc ^ row[w];
+        crc = tab4[3][c & 0xFF] ^ tab4[2][(c >> 8) & 0xFF]
+            ^ tab4[1][(c >> 16) & 0xFF] ^ tab4[0][c >> 24];
       }
-      for (uint32_t b = full8 * 8; b < len; b++) {
+      for (uint32_t b = full * 4; b < len; b++) {
         uint8_t byte = (uint8_t)(row[b / 4] >> (8 * (b & 3)));
-        crc = (crc >> 8) ^ tab8[0][(crc ^ byte) & 0xFF];
+        crc = (crc >> 8) ^ tab4[0][(crc ^ byte) & 0xFF];
       }
       chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + t] = crc ^ 0xFFFFFFFFu;
     }
@@ -1077,12 +1074,12 @@ static int ensure_device_constants() {
   h_build_crc_mats();
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table), h_crc_table, sizeof(h_crc_table)));
   {
-    static uint32_t t8[8][256];
-    for (int i = 0; i < 256; i++) t8[0][i] = h_crc_table[i];
-    for (int k = 1; k < 8; k++)
+    static uint32_t t4[4][256];
+    for (int i = 0; i < 256; i++) t4[0][i] = h_crc_table[i];
+    for (int k = 1; k < 4; k++)
       for (int i = 0; i < 256; i++)
-        t8[k][i] = (t8[k - 1][i] >> 8) ^ h_crc_table[t8[k - 1][i] & 0xFF];
-    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table8), t8, sizeof(t8)));
+        t4[k][i] = (t4[k - 1][i] >> 8) ^ h_crc_table[t4[k - 1][i] & 0xFF];
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table4), t4, sizeof(t4)));
   }
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_mats), h_crc_shift_mat, sizeof(h_crc_shift_mat)));
   done = true;
