@@ -777,115 +777,6 @@ __global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
   }
 }
 
-/* emit v2 (env TZS_EMIT_V=2): group-of-G prefetch — issue all G records'
- * gathers, then store them while the next group's loads are in flight.
- * Trades VGPRs (two G-record payload groups) for memory-level parallelism. */
-__global__ void k_emit_records_v2(const RecDesc* desc, const uint8_t* same,
-                                  const uint64_t* scan, const uint32_t* parts,
-                                  const uint64_t* seg_payload_start,
-                                  const uint64_t* part_scan_base,
-                                  uint8_t* out, uint32_t n) {
-  constexpr int PIPE_B = 2;
-  constexpr int G = 8;
-  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
-  uint32_t lane = threadIdx.x & (WAVE - 1);
-  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
-  for (uint64_t base = (uint64_t)wave * WAVE; base < n;
-       base += (uint64_t)nwaves * WAVE) {
-    uint32_t i = (uint32_t)base + lane;
-    uint64_t my_src = 0, my_dst = 0, my_h0 = 0, my_h1 = 0;
-    uint32_t my_len = 0, my_hdr = 0;
-    if (i < n) {
-      RecDesc v = desc[i];
-      uint32_t p = parts[i];
-      my_dst = seg_payload_start[p] + (scan[i] - part_scan_base[p]);
-      uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
-      uint8_t hdrbuf[16] = {0};
-      uint32_t hdr = 0;
-      if (same[i]) {
-        if (!prev_same) hdrbuf[hdr++] = 0xFE;
-        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-        my_src = v.src + v.klen;
-        my_len = v.vlen;
-      } else {
-        if (prev_same) hdrbuf[hdr++] = 0xFD;
-        hdr += d_vint_write(hdrbuf + hdr, v.klen);
-        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-        my_src = v.src;
-        my_len = v.klen + v.vlen;
-      }
-      my_hdr = hdr;
-      for (int b = 0; b < 8; b++) my_h0 |= (uint64_t)hdrbuf[b] << (8 * b);
-      for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
-    }
-    uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
-    uint32_t maxlen = my_len;
-    for (int sh = 32; sh >= 1; sh >>= 1) {
-      uint32_t o = __shfl_xor(maxlen, sh);
-      if (o > maxlen) maxlen = o;
-    }
-    if (nvalid == WAVE && maxlen <= WAVE * PIPE_B) {
-      uint8_t pb[2][G][PIPE_B];
-      uint64_t gsrc[2][G];
-      uint32_t glen[2][G];
-      auto load_group = [&](int slot, uint32_t g0) {
-#pragma unroll
-        for (int j = 0; j < G; j++) {
-          uint64_t src = __shfl(my_src, g0 + j);
-          uint32_t len = __shfl(my_len, g0 + j);
-          gsrc[slot][j] = src;
-          glen[slot][j] = len;
-          const uint8_t* sp = (const uint8_t*)(uintptr_t)src;
-#pragma unroll
-          for (int k = 0; k < PIPE_B; k++)
-            pb[slot][j][k] = (lane + k * WAVE < len) ? sp[lane + k * WAVE] : 0;
-        }
-      };
-      auto store_group = [&](int slot, uint32_t g0) {
-#pragma unroll
-        for (int j = 0; j < G; j++) {
-          uint32_t r = g0 + j;
-          uint64_t dsto = __shfl(my_dst, r);
-          uint64_t h0 = __shfl(my_h0, r);
-          uint64_t h1 = __shfl(my_h1, r);
-          uint32_t hdr = __shfl(my_hdr, r);
-          uint32_t len = glen[slot][j];
-          uint8_t* w = out + dsto;
-          if (lane < hdr)
-            w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
-                                 : (uint8_t)(h1 >> (8 * (lane - 8)));
-          w += hdr;
-#pragma unroll
-          for (int k = 0; k < PIPE_B; k++)
-            if (lane + k * WAVE < len) w[lane + k * WAVE] = pb[slot][j][k];
-        }
-      };
-      load_group(0, 0);
-      for (uint32_t g0 = 0; g0 < WAVE; g0 += G) {
-        int slot = (g0 / G) & 1;
-        if (g0 + G < WAVE) load_group(slot ^ 1, g0 + G);
-        store_group(slot, g0);
-      }
-    } else {
-      for (uint32_t r = 0; r < nvalid; r++) {
-        uint64_t src = __shfl(my_src, r);
-        uint64_t dsto = __shfl(my_dst, r);
-        uint64_t h0 = __shfl(my_h0, r);
-        uint64_t h1 = __shfl(my_h1, r);
-        uint32_t len = __shfl(my_len, r);
-        uint32_t hdr = __shfl(my_hdr, r);
-        uint8_t* w = out + dsto;
-        if (lane < hdr)
-          w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
-                               : (uint8_t)(h1 >> (8 * (lane - 8)));
-        w += hdr;
-        const uint8_t* sp = (const uint8_t*)(uintptr_t)src;
-        for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp[b];
-      }
-    }
-  }
-}
-
 /* ---- CRC over emitted segments ----
  * chunk kernel: thread computes CRC32 of one 256-byte chunk of one partition's
  * checksummed range (payload .. EOF); combine kernel: one wave per partition
@@ -1855,18 +1746,6 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   if (up(d_rlen, h_range_len.data(), 8 * P)) return -12;
 
   /* 6. emit records */
-  static int emit_v = -1;
-  if (emit_v < 0) {
-    const char* e = getenv("TZS_EMIT_V");
-    emit_v = (e && e[0] == '2') ? 2 : 1;
-  }
-  if (emit_v == 2)
-    hipLaunchKernelGGL(k_emit_records_v2, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
-                       (const RecDesc*)descbuf.p,
-                       (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
-                       (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
-                       (const uint64_t*)d_scanbase.p, d_out, n);
-  else
   hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
                      (const RecDesc*)descbuf.p,
                      (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
