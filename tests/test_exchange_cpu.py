@@ -79,3 +79,50 @@ def _worker(rank, world, tmpdir):
 def test_exchange_gloo_world2():
     import torch.multiprocessing as mp
     mp.spawn(_worker, args=(2, None), nprocs=2, join=True)
+
+
+def test_exchange_gloo_world4():
+    import torch.multiprocessing as mp
+    import os
+    os.environ["MASTER_PORT"] = "29518"
+    mp.spawn(_worker4, args=(4,), nprocs=4, join=True)
+
+
+def _worker4(rank, world):
+    import torch
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ.setdefault("MASTER_PORT", "29518")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    P = 11
+    recs = {p: [f"w4r{rank}p{p}i{i}".encode() for i in range((rank + p) % 3 + 1)]
+            for p in range(P)}
+    blob = b""
+    rec_ranges, byte_ranges, reclens, klens = [0], [0], [], []
+    for p in range(P):
+        for r in recs[p]:
+            blob += r
+            reclens.append(len(r))
+            klens.append(2)
+        rec_ranges.append(len(reclens))
+        byte_ranges.append(len(blob))
+    plan = ex.plan_send(rec_ranges, byte_ranges, world)
+    sd, srl, skl = b"", [], []
+    for i, p in enumerate(plan.order):
+        sd += blob[byte_ranges[p]:byte_ranges[p + 1]]
+        srl += reclens[rec_ranges[p]:rec_ranges[p + 1]]
+        skl += klens[rec_ranges[p]:rec_ranges[p + 1]]
+    rd, rrl, rkl = ex.exchange(
+        plan,
+        torch.from_numpy(np.frombuffer(sd, dtype=np.uint8).copy()),
+        torch.tensor(srl, dtype=torch.int32),
+        torch.tensor(skl, dtype=torch.int32))
+    want = b""
+    for src in range(world):
+        srecs = {p: [f"w4r{src}p{p}i{i}".encode() for i in range((src + p) % 3 + 1)]
+                 for p in range(P)}
+        for p in ex.parts_for_dest(P, world, rank):
+            for r in srecs[p]:
+                want += r
+    assert bytes(rd.numpy().tobytes()) == want, f"rank {rank}"
+    dist.destroy_process_group()

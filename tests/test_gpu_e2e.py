@@ -118,3 +118,49 @@ def test_exchange_world1_on_device(engine):
     assert got_idx == want_idx
     assert got_data == want_data
     dist.destroy_process_group()
+
+
+def test_pipelined_shuffle_events_and_segments(engine):
+    """final merge disabled: per-spill DMEs with spill_id/last_event
+    (PipelinedSorter.java:374-385,709-726); reduce side merges the per-spill
+    partition segments."""
+    from tez_amd.ordered_output import OrderedPartitionedKVOutput
+    from tez_amd.ordered_input import OrderedGroupedKVInput
+    from tez_amd import events as ev
+
+    P = 2
+    props = {"tez.runtime.key.class": "org.apache.hadoop.io.Text",
+             "tez.runtime.value.class": "org.apache.hadoop.io.IntWritable",
+             "tez.runtime.enable.final-merge.in.output": "false"}
+    out = OrderedPartitionedKVOutput(P, props, unique_id="attempt_p0").start()
+    w = out.get_writer()
+    rng = random.Random(7)
+    words1 = [WORDS[rng.randrange(len(WORDS))] for _ in range(3000)]
+    for word in words1:
+        w.write(word.encode(), 1)
+    mid_ev = out.spill()
+    d0 = ev.parse_dme_payload(mid_ev.payload)
+    assert d0["spill_id"] == 0 and d0["last_event"] is False
+    assert d0["path_component"] == "attempt_p0_0"
+    words2 = [WORDS[rng.randrange(len(WORDS))] for _ in range(3000)]
+    for word in words2:
+        w.write(word.encode(), 1)
+    evs = out.close()
+    # VM event + one DME per spill; the last carries last_event=true
+    dmes = [e for e in evs if isinstance(e, ev.CompositeDataMovementEvent)]
+    assert len(dmes) == 2
+    dlast = ev.parse_dme_payload(dmes[-1].payload)
+    assert dlast["spill_id"] == 1 and dlast["last_event"] is True
+
+    counted = {}
+    for p in range(P):
+        inp = OrderedGroupedKVInput(p, props)
+        for sid in range(2):
+            seg, _ = out.spill_segment(sid, p)
+            inp.add_segment(seg)
+        inp.start()
+        for key, vals in inp.get_reader():
+            counted[key.decode()] = counted.get(key.decode(), 0) + sum(
+                int.from_bytes(v, "big") for v in vals)
+    want = collections.Counter(words1 + words2)
+    assert counted == dict(want)

@@ -34,6 +34,11 @@ def conf_from_tez_properties(props: dict, num_partitions: int):
             return dflt
         return str(v).lower() in ("1", "true", "yes")
 
+    sorter_cls = props.get("tez.runtime.sorter.class", "PIPELINED")
+    if str(sorter_cls).upper() not in ("PIPELINED", "LEGACY"):
+        raise ValueError(f"unknown sorter.class {sorter_cls}")
+    # PIPELINED and LEGACY (DefaultSorter) both map to the GPU engine
+    # (one sorter serves both; TezRuntimeConfiguration.java:170)
     key_cls = props.get("tez.runtime.key.class",
                         "org.apache.hadoop.io.BytesWritable")
     if key_cls not in _KEY_CLASSES:

@@ -57,10 +57,46 @@ class OrderedPartitionedKVOutput:
     def get_writer(self):
         return KeyValuesWriter(self)
 
+    def spill(self):
+        """Force a spill; in pipelined-shuffle mode
+        (tez.runtime.enable.final-merge.in.output=false +
+        tez.runtime.pipelined-shuffle.enabled=true) returns the per-spill
+        CompositeDataMovementEvent (sendPipelinedShuffleEvents,
+        PipelinedSorter.java:374-385)."""
+        sid = self._sorter.spill()
+        _, index = self._sorter.spill_output(sid)
+        dme = ev.build_dme_payload(
+            index, self.host, self.port, f"{self.unique_id}_{sid}",
+            final_merge_enabled=False, spill_id=sid, last_event=False)
+        return ev.CompositeDataMovementEvent(0, self.num_partitions, dme)
+
     def close(self):
         """flush + final merge; returns the List<Event> equivalent
-        (VertexManagerEvent + CompositeDataMovementEvent)."""
+        (VertexManagerEvent + CompositeDataMovementEvent).  With final merge
+        disabled, one DME per spill is returned instead
+        (PipelinedSorter.flush :709-726)."""
         self._sorter.flush()
+        if not self._final_merge:
+            ctr = self._sorter.counters()
+            nspills = self._sorter.num_spills()
+            events = []
+            for sid in range(nspills):
+                _, index = self._sorter.spill_output(sid)
+                dme = ev.build_dme_payload(
+                    index, self.host, self.port, f"{self.unique_id}_{sid}",
+                    final_merge_enabled=False, spill_id=sid,
+                    last_event=(sid == nspills - 1))
+                events.append(ev.CompositeDataMovementEvent(
+                    0, self.num_partitions, dme))
+            events.insert(0, ev.VertexManagerEvent(
+                "<dest>", ev.build_vm_payload(ctr["output_bytes"],
+                                              ctr["output_records"])))
+            self._events = events
+            self._spill_data = [self._sorter.spill_output(s2)
+                                for s2 in range(nspills)]
+            self._sorter.close()
+            self._sorter = None
+            return events
         data, index = self._sorter.output()
         ctr = self._sorter.counters()
         self._events = ev.events_on_flush(
@@ -78,3 +114,9 @@ class OrderedPartitionedKVOutput:
     def segment(self, partition):
         s, r, c = self._index[partition]
         return self._data[s: s + c], r
+
+    def spill_segment(self, spill_id, partition):
+        """Pipelined mode: partition slice of one spill's IFile stream."""
+        data, index = self._spill_data[spill_id]
+        s, r, c = index[partition]
+        return data[s: s + c], r
