@@ -34,6 +34,9 @@ class OrderedPartitionedKVOutput:
         self._sorter = None
         self._records = 0
         self._events = None
+        self._final_merge = str(self.props.get(
+            "tez.runtime.enable.final-merge.in.output", "true")).lower() \
+            in ("1", "true", "yes")
         key_cls = self.props.get("tez.runtime.key.class",
                                  "org.apache.hadoop.io.BytesWritable")
         if key_cls == "org.apache.hadoop.io.Text":
