@@ -53,6 +53,12 @@ typedef struct tzs_conf {
   int32_t device;                      /* HIP device ordinal; -1 = current */
   int32_t world_size;                  /* GPUs sharing the shuffle (partition owner = p % world_size) */
   int32_t rank;                        /* this process's rank */
+  int32_t combiner;                    /* 0 none; 1 SUM_INT: fold equal-key runs,
+                                          summing 4B BE IntWritable values
+                                          (runCombineProcessor call sites,
+                                          PipelinedSorter.java:602-609,816-821) */
+  int32_t min_spills_for_combine;      /* tez.runtime.combine.min.spills, default 3
+                                          (PipelinedSorter.java:244) */
   int32_t reserved0;
 } tzs_conf;
 

@@ -68,14 +68,14 @@ def _load():
     lib.tzo_spill.restype = c.c_int
     lib.tzo_spill.argtypes = [
         u8p, c.POINTER(c.c_uint64), c.POINTER(c.c_uint32), c.POINTER(c.c_int32),
-        c.c_int64, c.c_int32, c.c_int, c.c_int, c.c_int, c.c_int,
+        c.c_int64, c.c_int32, c.c_int, c.c_int, c.c_int, c.c_int, c.c_int,
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
         c.POINTER(c.c_int64), c.POINTER(c.c_int)]
     lib.tzo_final_merge.restype = c.c_int
     lib.tzo_final_merge.argtypes = [
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64), c.POINTER(c.c_void_p),
-        c.c_int32, c.c_int32, c.c_int, c.c_int, c.c_int, c.c_int32,
+        c.c_int32, c.c_int32, c.c_int, c.c_int, c.c_int, c.c_int32, c.c_int,
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64)]
     lib.tzo_free.argtypes = [c.c_void_p]
@@ -171,7 +171,7 @@ def build_records(serialized_pairs):
 
 def spill(data, off, klen, num_partitions, key_type=KEY_BYTES,
           comparator=CMP_TEZBYTES, rle_mode=-1, send_empty=True,
-          partitions=None, want_order=False):
+          partitions=None, want_order=False, combiner=0):
     """One PipelinedSorter spill (see tzo_spill). Returns dict with
     data/index bytes, rle flag, and optionally the sorted order."""
     n = len(klen)
@@ -186,6 +186,7 @@ def spill(data, off, klen, num_partitions, key_type=KEY_BYTES,
         _u8p(data), off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
         klen.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)), pp,
         n, num_partitions, key_type, comparator, rle_mode, int(send_empty),
+        combiner,
         ctypes.byref(od), ctypes.byref(odl), ctypes.byref(oi), ctypes.byref(oil),
         order.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)) if want_order else None,
         ctypes.byref(rle))
@@ -197,7 +198,7 @@ def spill(data, off, klen, num_partitions, key_type=KEY_BYTES,
 
 
 def final_merge(spills, num_partitions, comparator=CMP_TEZBYTES, rle_mode=-1,
-                send_empty=True, factor=100):
+                send_empty=True, factor=100, combiner=0):
     """PipelinedSorter.flush final merge over spill outputs
     (list of dicts with 'data'/'index'). numSpills==1 is a rename
     (PipelinedSorter.java:731-757): returns the spill unchanged."""
@@ -217,7 +218,7 @@ def final_merge(spills, num_partitions, comparator=CMP_TEZBYTES, rle_mode=-1,
     oi = ctypes.c_void_p()
     oil = ctypes.c_int64()
     rc = _lib.tzo_final_merge(dptr, dlen, iptr, n, num_partitions, comparator,
-                              rle_mode, int(send_empty), factor,
+                              rle_mode, int(send_empty), factor, combiner,
                               ctypes.byref(od), ctypes.byref(odl),
                               ctypes.byref(oi), ctypes.byref(oil))
     assert rc == 0, f"tzo_final_merge rc={rc}"

@@ -428,12 +428,27 @@ static int spill_cmp(const void* A, const void* B) {
   return ia < ib ? -1 : (ia > ib ? 1 : 0); /* deterministic; unpinned vs reference */
 }
 
+/* Combiner restatement (runCombineProcessor call sites:
+ * PipelinedSorter.java:602-609,816-821): fold runs of comparator-equal keys
+ * within a partition, summing 4-byte big-endian IntWritable values (two's
+ * complement wrap, as java int addition does).  combiner: 0 = none,
+ * 1 = SUM_INT. */
+static int32_t be32(const uint8_t* p) {
+  return (int32_t)(((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) |
+                   ((uint32_t)p[2] << 8) | (uint32_t)p[3]);
+}
+static void put_be32(uint8_t* p, int32_t v) {
+  p[0] = (uint8_t)((uint32_t)v >> 24); p[1] = (uint8_t)((uint32_t)v >> 16);
+  p[2] = (uint8_t)((uint32_t)v >> 8); p[3] = (uint8_t)v;
+}
+
 TZO_API int tzo_spill(
     /* records: record i = serialized key ‖ serialized value at off[i]..off[i+1] */
     const uint8_t* data, const uint64_t* off, const uint32_t* klen,
     const int32_t* part_in /* may be NULL => HashPartitioner */,
     int64_t n,
     int32_t P, int key_type, int comparator, int rle_mode, int send_empty,
+    int combiner,
     uint8_t** out_data, int64_t* out_data_len,
     uint8_t** out_index, int64_t* out_index_len,
     int64_t* out_order /* [n] or NULL */, int* out_rle) {
@@ -482,12 +497,36 @@ TZO_API int tzo_spill(
     int64_t rawl = 0, partl = 0;
     if (has || !send_empty) {
       tzo_writer* w = tzo_writer_new(rle);
-      for (int64_t i = lo; i < pos; i++) {
-        int64_t r = order[i];
-        const uint8_t* kb = data + off[r];
-        const uint8_t* vb = kb + klen[r];
-        int32_t vl = (int32_t)(off[r + 1] - off[r] - klen[r]);
-        tzo_writer_append(w, kb, (int32_t)klen[r], vb, vl);
+      if (combiner == 1) {
+        /* fold equal-key runs, emit (key, sum) */
+        int64_t i = lo;
+        while (i < pos) {
+          int64_t r0 = order[i];
+          const uint8_t* kb = data + off[r0];
+          int32_t kl = (int32_t)klen[r0];
+          int32_t sum = 0;
+          int64_t j = i;
+          for (; j < pos; j++) {
+            int64_t r = order[j];
+            if (j > i && tzo_compare_key(comparator, kb, kl, data + off[r],
+                                         (int32_t)klen[r]) != 0)
+              break;
+            sum = (int32_t)((uint32_t)sum +
+                            (uint32_t)be32(data + off[r] + klen[r]));
+          }
+          uint8_t vb[4];
+          put_be32(vb, sum);
+          tzo_writer_append(w, kb, kl, vb, 4);
+          i = j;
+        }
+      } else {
+        for (int64_t i = lo; i < pos; i++) {
+          int64_t r = order[i];
+          const uint8_t* kb = data + off[r];
+          const uint8_t* vb = kb + klen[r];
+          int32_t vl = (int32_t)(off[r + 1] - off[r] - klen[r]);
+          tzo_writer_append(w, kb, (int32_t)klen[r], vb, vl);
+        }
       }
       uint8_t* seg; int64_t seglen;
       tzo_writer_close(w, &seg, &seglen, &rawl, &partl);
@@ -669,7 +708,7 @@ TZO_API int tzo_final_merge(
     const uint8_t* const* spill_data, const int64_t* spill_len,
     const uint8_t* const* spill_index /* 24P+8 each; big-endian triples */,
     int32_t nspills, int32_t P, int comparator, int rle_mode, int send_empty,
-    int32_t factor,
+    int32_t factor, int combiner /* 0 none, 1 SUM_INT (gated by caller) */,
     uint8_t** out_data, int64_t* out_data_len,
     uint8_t** out_index, int64_t* out_index_len) {
   (void)spill_len;
@@ -744,7 +783,41 @@ TZO_API int tzo_final_merge(
       for (int i = 0; i < ns; i++) segs[i]->seq = i;
       if (should_write) {
         uint8_t* seg; int64_t seglen;
-        merge_records(segs, ns, comparator, rle_mode, 1, &seg, &seglen, &rawl, &partl);
+        if (combiner == 1) {
+          /* fold across the merged stream (runCombineProcessor at final
+             merge, PipelinedSorter.java:816-821): flatten the merge, fold,
+             re-emit */
+          uint8_t* tmp; int64_t tl, trl, tpl;
+          merge_records(segs, ns, comparator, 0, 1, &tmp, &tl, &trl, &tpl);
+          tzo_records* R = NULL;
+          tzo_ifile_read(tmp, tl, 1, &R);
+          free(tmp);
+          tzo_writer* w = tzo_writer_new(rle_mode == 1);
+          int64_t i = 0;
+          while (i < R->n) {
+            const uint8_t* kb = R->keys + R->key_off[i];
+            int32_t kl = (int32_t)(R->key_off[i + 1] - R->key_off[i]);
+            int32_t sum = 0;
+            int64_t j = i;
+            for (; j < R->n; j++) {
+              const uint8_t* kb2 = R->keys + R->key_off[j];
+              int32_t kl2 = (int32_t)(R->key_off[j + 1] - R->key_off[j]);
+              if (j > i && tzo_compare_key(comparator, kb, kl, kb2, kl2) != 0)
+                break;
+              sum = (int32_t)((uint32_t)sum +
+                              (uint32_t)be32(R->vals + R->val_off[j]));
+            }
+            uint8_t vb[4];
+            put_be32(vb, sum);
+            tzo_writer_append(w, kb, kl, vb, 4);
+            i = j;
+          }
+          tzo_writer_close(w, &seg, &seglen, &rawl, &partl);
+          tzo_writer_free(w);
+          tzo_records_free(R);
+        } else {
+          merge_records(segs, ns, comparator, rle_mode, 1, &seg, &seglen, &rawl, &partl);
+        }
         if (seg) { buf_put(&file, seg, (size_t)seglen); free(seg); }
       }
     }

@@ -30,6 +30,8 @@ class TzsConf(ctypes.Structure):
         ("device", ctypes.c_int32),
         ("world_size", ctypes.c_int32),
         ("rank", ctypes.c_int32),
+        ("combiner", ctypes.c_int32),
+        ("min_spills_for_combine", ctypes.c_int32),
         ("reserved0", ctypes.c_int32),
     ]
 

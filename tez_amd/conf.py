@@ -48,10 +48,21 @@ def conf_from_tez_properties(props: dict, num_partitions: int):
     cmp_cls = props.get("tez.runtime.key.comparator.class")
     comparator = _COMPARATORS.get(cmp_cls, cmp_default) if cmp_cls else cmp_default
 
+    combiner = 0
+    comb_cls = props.get("tez.runtime.combiner.class")
+    if comb_cls:
+        val_cls = props.get("tez.runtime.value.class", "")
+        if val_cls != "org.apache.hadoop.io.IntWritable":
+            raise ValueError("round-1 combiner supports IntWritable-sum only "
+                             "(SURVEY §8f row 1)")
+        combiner = 1  # SUM_INT
+
     return _engine.make_conf(
         num_partitions,
         key_type=key_type,
         comparator=comparator,
+        combiner=combiner,
+        min_spills_for_combine=int(props.get("tez.runtime.combine.min.spills", 3)),
         send_empty_partition_details=int(
             b("tez.runtime.empty.partitions.info-via-events.enabled", True)),
         io_sort_factor=int(props.get("tez.runtime.io.sort.factor", 100)),

@@ -574,6 +574,65 @@ __global__ void k_count_nonzero_u8(const uint8_t* a, uint32_t n, uint32_t* out) 
   if (threadIdx.x == 0) atomicAdd(out, s);
 }
 
+/* ---- combiner (SUM_INT) ----
+ * Fold runs of equal keys (eq flags) into one record with the 4-byte
+ * big-endian IntWritable sum (java int wrap) — runCombineProcessor
+ * restated (PipelinedSorter.java:602-609,816-821). */
+__global__ void k_combine_mark(const uint8_t* eq, uint64_t* runstart, uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x)
+    runstart[i] = eq[i] ? 0 : 1;
+}
+__global__ void k_combine_pos(const uint8_t* eq, const uint64_t* rs_scan,
+                              uint32_t* pos, uint64_t* lens2, RecTable rt,
+                              const uint32_t* sidx, uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    if (!eq[i]) {
+      uint32_t j = (uint32_t)rs_scan[i];
+      pos[j] = i;
+      RecView v = rt_view(rt, sidx[i]);
+      lens2[j] = (uint64_t)v.klen + 4;
+    }
+  }
+}
+__global__ void k_combine_fold(RecTable rt, const uint32_t* sidx, const uint32_t* pos,
+                               const uint64_t* off2, const uint32_t* parts_in,
+                               uint8_t* data2, uint32_t* klen2, uint32_t* parts2,
+                               uint32_t M, uint32_t n) {
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (uint32_t j = wave; j < M; j += nwaves) {
+    uint32_t i0 = pos[j];
+    uint32_t i1 = (j + 1 < M) ? pos[j + 1] : n;
+    RecView k0 = rt_view(rt, sidx[i0]);
+    /* lane-parallel partial sums over the run */
+    uint32_t sum = 0;
+    for (uint32_t i = i0 + lane; i < i1; i += WAVE) {
+      RecView v = rt_view(rt, sidx[i]);
+      sum += ((uint32_t)v.val[0] << 24) | ((uint32_t)v.val[1] << 16) |
+             ((uint32_t)v.val[2] << 8) | (uint32_t)v.val[3];
+    }
+    for (int sh = 32; sh >= 1; sh >>= 1) sum += __shfl_xor(sum, sh);
+    uint8_t* w = data2 + off2[j];
+    for (uint32_t b = lane; b < k0.klen; b += WAVE) w[b] = k0.key[b];
+    if (lane == 0) {
+      w[k0.klen + 0] = (uint8_t)(sum >> 24);
+      w[k0.klen + 1] = (uint8_t)(sum >> 16);
+      w[k0.klen + 2] = (uint8_t)(sum >> 8);
+      w[k0.klen + 3] = (uint8_t)sum;
+      klen2[j] = k0.klen;
+      parts2[j] = parts_in[i0];
+    }
+  }
+}
+__global__ void k_iota(uint32_t* a, uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x)
+    a[i] = i;
+}
+
 /* ---- emit ---- */
 /* Sorted record descriptors: one gather pass per sort; all later emit-side
  * kernels read these coalesced instead of re-gathering off/klen per record
@@ -1313,6 +1372,7 @@ struct tzs_sorter {
   uint32_t cur_klen_u = 0;
   bool cur_first_batch = true;
   bool have_explicit_parts = false;
+  bool combined_parts_valid = false;
   /* host-path staging */
   std::vector<uint8_t> host_data;
   std::vector<uint64_t> host_off;
@@ -1351,6 +1411,8 @@ extern "C" void tzs_conf_default(tzs_conf* c, int32_t num_partitions) {
   c->device = -1;
   c->world_size = 1;
   c->rank = 0;
+  c->combiner = 0;
+  c->min_spills_for_combine = 3; /* TEZ_RUNTIME_COMBINE_MIN_SPILLS, PipelinedSorter.java:244 */
 }
 
 extern "C" int tzs_sorter_create(const tzs_conf* conf, tzs_sorter** out) {
@@ -1489,7 +1551,7 @@ __global__ void k_shift_offsets(const uint64_t* src, uint64_t* dst, uint64_t shi
 static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                          const int32_t* d_part_unsorted,
                          const uint8_t* h_spill_rle, int nspills_rle,
-                         SpillData* outsp) {
+                         SpillData* outsp, bool apply_combine = false) {
   tzs_times& T = s->times;
   g_scatter_ns = 0; g_scatter_launches = 0; g_scatter_elems = 0;
   hipEvent_t ev[10];
@@ -1631,6 +1693,60 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   int writer_rle;
   if (s->conf.rle >= 0) writer_rle = s->conf.rle;
   else writer_rle = ((uint64_t)neq_final * 10 > n) ? 1 : 0;
+
+  /* combiner stage: replace the sorted view with folded records */
+  static thread_local SpillData combined;
+  if (apply_combine && s->conf.combiner == 1) {
+    /* partitions of the sorted records (needed for parts2) */
+    if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
+    hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
+                       (uint32_t*)s->parts_sorted.p, n);
+    static thread_local DBuf rs, rs_scan, pos2, lens2, off2, parts2, idx2;
+    if (rs.alloc(8ull * n) || rs_scan.alloc(8ull * n)) return -12;
+    hipLaunchKernelGGL(k_combine_mark, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq,
+                       (uint64_t*)rs.p, n);
+    uint64_t M64 = 0;
+    if (scan_u64((uint64_t*)rs.p, (uint64_t*)rs_scan.p, n, &M64)) return -12;
+    uint32_t M = (uint32_t)M64;
+    if (pos2.alloc(4ull * M) || lens2.alloc(8ull * M) || off2.alloc(8ull * (M + 1)))
+      return -12;
+    hipLaunchKernelGGL(k_combine_pos, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq,
+                       (const uint64_t*)rs_scan.p, (uint32_t*)pos2.p,
+                       (uint64_t*)lens2.p, rt, d_idx, n);
+    uint64_t bytes2 = 0;
+    if (scan_u64((uint64_t*)lens2.p, (uint64_t*)off2.p, M, &bytes2)) return -12;
+    HIP_CHECK(hipMemcpy((uint64_t*)off2.p + M, &bytes2, 8, hipMemcpyHostToDevice));
+    if (combined.data.alloc(bytes2 ? bytes2 : 1)) return -12;
+    if (combined.klen.alloc(4ull * M)) return -12;
+    if (parts2.alloc(4ull * M) || idx2.alloc(4ull * M)) return -12;
+    hipLaunchKernelGGL(k_combine_fold, dim3(grid_waves(M)), dim3(BLOCK), 0, 0, rt,
+                       d_idx, (const uint32_t*)pos2.p, (const uint64_t*)off2.p,
+                       (const uint32_t*)s->parts_sorted.p, (uint8_t*)combined.data.p,
+                       (uint32_t*)combined.klen.p, (uint32_t*)parts2.p, M, n);
+    /* swap the view: single-"spill" table over the folded records */
+    std::swap(combined.off, off2);
+    RecTable rt2 = {};
+    rt2.nspills = 1;
+    rt2.data[0] = (const uint8_t*)combined.data.p;
+    rt2.off[0] = (const uint64_t*)combined.off.p;
+    rt2.klen[0] = (const uint32_t*)combined.klen.p;
+    rt2.base[0] = 0; rt2.base[1] = M;
+    rt2.key_type = rt.key_type;
+    rt = rt2;
+    n = M;
+    hipLaunchKernelGGL(k_iota, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                       (uint32_t*)idx2.p, n);
+    std::swap(s->sidx, idx2);
+    d_idx = (uint32_t*)s->sidx.p;
+    if (s->eq.alloc(n)) return -12;
+    d_eq = (uint8_t*)s->eq.p;
+    HIP_CHECK(hipMemsetAsync(d_eq, 0, n));
+    std::swap(s->parts_sorted, parts2);
+    writer_rle = 0; /* folded keys are unique; rle never triggers */
+    s->combined_parts_valid = true;
+  } else {
+    s->combined_parts_valid = false;
+  }
   if (s->same.alloc(n)) return -12;
   static thread_local DBuf d_sprle;
   if (d_sprle.alloc(nspills_rle ? nspills_rle : 1)) return -12;
@@ -1640,9 +1756,11 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                      writer_rle, (const uint8_t*)d_sprle.p, (uint8_t*)s->same.p, n);
 
   /* 5. emit sizes + partition layout */
-  if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
-  hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
-                     (uint32_t*)s->parts_sorted.p, n);
+  if (!s->combined_parts_valid) {
+    if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
+    hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
+                       (uint32_t*)s->parts_sorted.p, n);
+  }
   static thread_local DBuf descbuf;
   if (descbuf.alloc(sizeof(RecDesc) * n)) return -12;
   hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
@@ -1920,7 +2038,7 @@ extern "C" int tzs_sorter_spill(tzs_sorter* s) {
   uint8_t dummy_rle = 0;
   rc = sort_and_emit(s, rt, (uint32_t)sp->n,
                      s->have_explicit_parts ? (const int32_t*)s->cur_part.p : nullptr,
-                     &dummy_rle, 1, sp);
+                     &dummy_rle, 1, sp, s->conf.combiner != 0);
   s->cur_part.release();
   s->cur_n = 0;
   s->cur_bytes = 0;
@@ -2012,8 +2130,10 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
   if (s->have_explicit_parts)
     FAIL(-22, "explicit partitions require a single spill in round 1");
   SpillData finalsp;
+  bool combine_at_merge = s->conf.combiner != 0 &&
+                          nsp >= s->conf.min_spills_for_combine;
   rc = sort_and_emit(s, rt, (uint32_t)total_n, nullptr,
-                     spill_rle.data(), nsp, &finalsp);
+                     spill_rle.data(), nsp, &finalsp, combine_at_merge);
   if (rc) return rc;
   s->final_index = finalsp.index;
   std::swap(s->final_ifile, finalsp.ifile);
