@@ -1373,6 +1373,12 @@ struct tzs_sorter {
   bool cur_first_batch = true;
   bool have_explicit_parts = false;
   bool combined_parts_valid = false;
+  /* combiner output of the last sort_and_emit (adopted by spill(): the
+     reference's later merges read the COMBINED spill files, so the spill's
+     stored record set must be the folded one) */
+  DBuf combine_data, combine_off, combine_klen;
+  uint32_t combine_n = 0;
+  bool combine_applied = false;
   /* host-path staging */
   std::vector<uint8_t> host_data;
   std::vector<uint64_t> host_off;
@@ -1695,7 +1701,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   else writer_rle = ((uint64_t)neq_final * 10 > n) ? 1 : 0;
 
   /* combiner stage: replace the sorted view with folded records */
-  static thread_local SpillData combined;
+  s->combine_applied = false;
   if (apply_combine && s->conf.combiner == 1) {
     /* partitions of the sorted records (needed for parts2) */
     if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
@@ -1716,20 +1722,20 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     uint64_t bytes2 = 0;
     if (scan_u64((uint64_t*)lens2.p, (uint64_t*)off2.p, M, &bytes2)) return -12;
     HIP_CHECK(hipMemcpy((uint64_t*)off2.p + M, &bytes2, 8, hipMemcpyHostToDevice));
-    if (combined.data.alloc(bytes2 ? bytes2 : 1)) return -12;
-    if (combined.klen.alloc(4ull * M)) return -12;
+    if (s->combine_data.alloc(bytes2 ? bytes2 : 1)) return -12;
+    if (s->combine_klen.alloc(4ull * M)) return -12;
     if (parts2.alloc(4ull * M) || idx2.alloc(4ull * M)) return -12;
     hipLaunchKernelGGL(k_combine_fold, dim3(grid_waves(M)), dim3(BLOCK), 0, 0, rt,
                        d_idx, (const uint32_t*)pos2.p, (const uint64_t*)off2.p,
-                       (const uint32_t*)s->parts_sorted.p, (uint8_t*)combined.data.p,
-                       (uint32_t*)combined.klen.p, (uint32_t*)parts2.p, M, n);
+                       (const uint32_t*)s->parts_sorted.p, (uint8_t*)s->combine_data.p,
+                       (uint32_t*)s->combine_klen.p, (uint32_t*)parts2.p, M, n);
     /* swap the view: single-"spill" table over the folded records */
-    std::swap(combined.off, off2);
+    std::swap(s->combine_off, off2);
     RecTable rt2 = {};
     rt2.nspills = 1;
-    rt2.data[0] = (const uint8_t*)combined.data.p;
-    rt2.off[0] = (const uint64_t*)combined.off.p;
-    rt2.klen[0] = (const uint32_t*)combined.klen.p;
+    rt2.data[0] = (const uint8_t*)s->combine_data.p;
+    rt2.off[0] = (const uint64_t*)s->combine_off.p;
+    rt2.klen[0] = (const uint32_t*)s->combine_klen.p;
     rt2.base[0] = 0; rt2.base[1] = M;
     rt2.key_type = rt.key_type;
     rt = rt2;
@@ -1744,6 +1750,8 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     std::swap(s->parts_sorted, parts2);
     writer_rle = 0; /* folded keys are unique; rle never triggers */
     s->combined_parts_valid = true;
+    s->combine_applied = true;
+    s->combine_n = M;
   } else {
     s->combined_parts_valid = false;
   }
@@ -2043,6 +2051,17 @@ extern "C" int tzs_sorter_spill(tzs_sorter* s) {
   s->cur_n = 0;
   s->cur_bytes = 0;
   if (rc) { delete sp; return rc; }
+  if (s->combine_applied) {
+    /* later merges must read the combined spill (reference semantics):
+       swap in the folded columnar set */
+    std::swap(sp->data, s->combine_data);
+    std::swap(sp->off, s->combine_off);
+    std::swap(sp->klen, s->combine_klen);
+    sp->n = s->combine_n;
+    sp->rec_u = 0;
+    sp->klen_u = 0;
+    s->combine_applied = false;
+  }
   s->spills.push_back(sp);
   s->ctr.spilled_records += sp->n;
   s->ctr.num_spills = (int64_t)s->spills.size();
