@@ -39,15 +39,20 @@ def vint_write(v):
 
 
 def read_stream(stream, with_header=True, verify_crc=True):
-    """Parse one IFile stream; returns list of (key_ser, val_ser, same_key)."""
+    """Parse one IFile stream; returns list of (key_ser, val_ser, same_key).
+    Handles both TIF\0 (plain) and TIF\1 (DefaultCodec/zlib-compressed
+    payload, CRC over the COMPRESSED bytes — IFile.java:352-368, pinned by
+    the reference golden fixture)."""
     if with_header:
         if stream[:3] != b"TIF":
             raise ValueError("bad IFile magic")
-        if stream[3] != 0:
-            raise ValueError("compressed IFile segments are out of scope (SURVEY §8f)")
         body = stream[4:-4]
-        if verify_crc and zlib.crc32(stream[4:-4]) != int.from_bytes(stream[-4:], "big"):
+        if verify_crc and zlib.crc32(body) != int.from_bytes(stream[-4:], "big"):
             raise ValueError("IFile CRC mismatch")
+        if stream[3] == 1:
+            body = zlib.decompress(bytes(body))
+        elif stream[3] != 0:
+            raise ValueError("unknown IFile compression flag")
     else:
         body = stream
     pos = 0
@@ -101,3 +106,13 @@ def serialize_int_writable(v: int) -> bytes:
 
 def deserialize_int_writable(ser: bytes) -> int:
     return int.from_bytes(ser[:4], "big", signed=True)
+
+
+def compress_segment(plain_stream: bytes) -> bytes:
+    """Re-frame a plain TIF\0 segment as TIF\1 with DefaultCodec(zlib)
+    payload (checksum below the codec,
+    IFile.java:352-368).  rawLength of the segment is unchanged; the new
+    partLength is the returned length."""
+    assert plain_stream[:4] == b"TIF\x00"
+    payload = zlib.compress(plain_stream[4:-4], 9)
+    return b"TIF\x01" + payload + zlib.crc32(payload).to_bytes(4, "big")
