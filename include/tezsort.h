@@ -30,7 +30,7 @@ const char* tzs_last_error(void);
  * Python layer (tez_amd/conf.py) parses the tez.runtime.* strings into this.
  */
 typedef enum {
-  TZS_KEY_BYTES = 0,   /* BytesWritable: 4B BE len + content (fixed content len only in r1) */
+  TZS_KEY_BYTES = 0,   /* BytesWritable: 4B BE len + content (variable lengths supported; order = proxy-then-serialized) */
   TZS_KEY_TEXT  = 1    /* Text: vint len + UTF-8 content */
 } tzs_key_type;
 

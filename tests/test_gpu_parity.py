@@ -234,3 +234,30 @@ def test_write_files_reference_layout(engine, tmp_path):
     assert fo.read_bytes() == data
     parsed = o.index_decode(fi.read_bytes(), 4)
     assert parsed == idx
+
+
+def test_parity_variable_length_bytes_keys(engine):
+    """Variable-length BytesWritable keys: the reference order is 3-byte
+    content proxy FIRST, then the serialized compare (4B length prefix =>
+    length-then-content) — PipelinedSorter.java:451-457 +
+    TezBytesComparator.  Includes the divergence case where content order
+    and serialized order disagree ('aaaAx' vs 'aaaB')."""
+    rng = random.Random(41)
+    contents = [b"aaaAx", b"aaaB", b"\x01", b"\x00\x00", b"", b"a", b"a\x00",
+                b"aaa", b"aaaZ" * 3, b"zz"]
+    pairs = []
+    for i in range(4000):
+        c = contents[rng.randrange(len(contents))] + \
+            (bytes([rng.randrange(256)]) * rng.randrange(0, 3))
+        pairs.append((o.serialize_bytes_writable(c),
+                      o.serialize_bytes_writable(b"v%05d" % i)))
+    # plus random unique variable-length keys
+    for i in range(4000):
+        c = bytes(rng.randrange(256) for _ in range(rng.randrange(0, 24)))
+        pairs.append((o.serialize_bytes_writable(c),
+                      o.serialize_bytes_writable(b"w%05d" % i)))
+    got, gidx, _ = _run_engine_host_path(engine, pairs, 8, engine.KEY_BYTES,
+                                         engine.CMP_TEZBYTES)
+    want = _oracle_single_spill(pairs, 8, o.KEY_BYTES, o.CMP_TEZBYTES)
+    assert gidx == o.index_decode(want["index"], 8)
+    assert got == want["data"]

@@ -234,7 +234,8 @@ __global__ void k_shift_offsets(const uint64_t* src, uint64_t* dst, uint64_t shi
 __global__ void k_max_u32(const uint32_t* a, uint32_t n, uint32_t* out);
 __global__ void k_part_hist(const uint32_t* parts, uint32_t n, uint32_t* counts, int P);
 __global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* sidx,
-                              int lb0, int use_len, uint64_t* lkey, uint32_t m);
+                              int lb0, int use_len, int ser_mode, uint64_t* lkey,
+                              uint32_t m);
 __global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
                                    const uint8_t* same, uint64_t total_body,
                                    uint32_t n, int P, uint64_t* out);
@@ -255,10 +256,15 @@ __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32
  * (adaptive pass count — DESIGN.md §4); rarer-than-1% ties are resolved by
  * the refinement levels, whose equality test must see the same mask. */
 __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
-                                  int pbits, int sort_bytes, uint64_t* d_key,
-                                  uint32_t* d_idx, uint32_t n) {
+                                  int pbits, int sort_bytes, int ser_mode,
+                                  uint64_t* d_key, uint32_t* d_idx, uint32_t n) {
   /* d_part == nullptr: compute the HashPartitioner placement here (fused —
-     a separate hash kernel re-reads all content bytes, ~10 GB at n=1e8). */
+     a separate hash kernel re-reads all content bytes, ~10 GB at n=1e8).
+     ser_mode (TezBytesComparator): order = partition, 3-byte content proxy,
+     then the SERIALIZED key bytes (4B BE length first => length-then-content
+     ties) — the reference prefix-int + raw-compare order
+     (PipelinedSorter.java:451-463, TezBytesComparator.java:38-62).
+     !ser_mode (Text): partition then content bytes, shorter-first ties. */
   uint64_t mask = (sort_bytes >= 8) ? ~0ull : ~0ull << (8 * (8 - sort_bytes));
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
@@ -266,10 +272,22 @@ __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
     uint32_t part;
     if (d_part) part = (uint32_t)d_part[i];
     else part = (uint32_t)((d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P);
-    uint64_t c = 0;
-    uint32_t m = v.clen < 8 ? v.clen : 8;
-    for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
-    uint64_t key = pbits ? (((uint64_t)part << (64 - pbits)) | (c >> pbits)) : c;
+    uint64_t key;
+    if (ser_mode) {
+      uint32_t proxy = ((v.clen > 0 ? (uint32_t)v.content[0] : 0u) << 16) |
+                       ((v.clen > 1 ? (uint32_t)v.content[1] : 0u) << 8) |
+                       (v.clen > 2 ? (uint32_t)v.content[2] : 0u);
+      uint64_t ser = 0;
+      uint32_t m = v.klen < 8 ? v.klen : 8;
+      for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)v.key[b] << (56 - 8 * b);
+      key = ((uint64_t)part << (64 - pbits)) | ((uint64_t)proxy << (64 - pbits - 24))
+            | (ser >> (pbits + 24));
+    } else {
+      uint64_t c = 0;
+      uint32_t m = v.clen < 8 ? v.clen : 8;
+      for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
+      key = pbits ? (((uint64_t)part << (64 - pbits)) | (c >> pbits)) : c;
+    }
     d_key[i] = key & mask;
     d_idx[i] = i;
   }
@@ -507,6 +525,7 @@ __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
                                  const uint8_t* eq,
                                  const uint64_t* inrun_scan, const uint64_t* runstart_scan,
                                  uint32_t n, int level_byte0, int use_len_level,
+                                 int ser_mode,
                                  uint64_t* lkey, uint32_t* seg, uint32_t* pos) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
@@ -517,13 +536,15 @@ __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
     uint32_t is_start = (eq[i] == 0) ? 1u : 0u;
     uint32_t sg = (uint32_t)runstart_scan[i] + is_start - 1;
     RecView v = rt_view(rt, sidx[i]);
+    const uint8_t* src = ser_mode ? v.key : v.content;
+    uint32_t slen = ser_mode ? v.klen : v.clen;
     uint64_t k = 0;
     if (use_len_level) {
-      k = v.clen;
+      k = slen;
     } else {
       for (int b = 0; b < 8; b++) {
         uint32_t cb = (uint32_t)(level_byte0 + b);
-        uint8_t byte = (cb < v.clen) ? v.content[cb] : 0;
+        uint8_t byte = (cb < slen) ? src[cb] : 0;
         k |= (uint64_t)byte << (56 - 8 * b);
       }
     }
@@ -1569,10 +1590,16 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   /* adaptive radix width: enough composite bits that expected tie-involved
      records stay below ~1% of n (ties go through refinement anyway):
      bits = pbits + log2(n) + 6.  Tests at small n exercise refinement hard. */
+  int ser_mode = (s->conf.comparator == TZS_CMP_TEZBYTES) ? 1 : 0;
   int needed_bits = pbits + 6;
   for (uint64_t v = n; v; v >>= 1) needed_bits++;
   int SB = (needed_bits + 7) / 8;
   if (SB < 2) SB = 2;
+  if (ser_mode) {
+    /* partition + full 24-bit proxy must lie inside the sorted bytes */
+    int minsb = (pbits + 24 + 7) / 8;
+    if (SB < minsb) SB = minsb;
+  }
   if (SB > 8) SB = 8;
   /* 1. composites */
   if (s->skey.alloc(sizeof(uint64_t) * n)) return -12;
@@ -1580,7 +1607,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   uint64_t* d_key = (uint64_t*)s->skey.p;
   uint32_t* d_idx = (uint32_t*)s->sidx.p;
   hipLaunchKernelGGL(k_build_composite, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                     d_part_unsorted, P, pbits, SB, d_key, d_idx, n);
+                     d_part_unsorted, P, pbits, SB, ser_mode, d_key, d_idx, n);
   (void)hipEventRecord(ev[1]);
 
   /* 2. base radix sort over the top SB bytes of the composite */
@@ -1596,7 +1623,11 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   uint8_t* d_eq = (uint8_t*)s->eq.p;
   hipLaunchKernelGGL(k_eq_init, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, d_eq, n);
   /* max content length: for refinement level count */
-  int c0 = (8 * SB - pbits) / 8;
+  /* refinement start byte within the comparator's byte source: sorted-covered
+     bits are 8*SB - pbits (content mode) or 8*SB - pbits - 24 past the proxy
+     (serialized mode; the proxy is rechecked implicitly because serialized
+     bytes repeat the content after the 4B length) */
+  int c0 = ser_mode ? (8 * SB - pbits - 24) / 8 : (8 * SB - pbits) / 8;
   if (c0 < 0) c0 = 0;
   /* determine max clen lazily: use a safe cap by scanning klen on host?  We
      compute it from the conf: key_type BYTES => clen = klen-4 (max over
@@ -1652,7 +1683,8 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     int lb0 = c0 + 8 * li;
     hipLaunchKernelGGL(k_compact_refine, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
                        d_eq, (uint64_t*)inrun_scan.p, (uint64_t*)runstart_scan.p, n,
-                       lb0, use_len, (uint64_t*)lkey.p, (uint32_t*)seg.p, (uint32_t*)pos.p);
+                       lb0, use_len, ser_mode, (uint64_t*)lkey.p, (uint32_t*)seg.p,
+                       (uint32_t*)pos.p);
     /* slotpos = copy of pos (ascending) before sort */
     HIP_CHECK(hipMemcpyAsync(slotpos.p, pos.p, sizeof(uint32_t) * m,
                              hipMemcpyDeviceToDevice));
@@ -1670,7 +1702,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
        as two u32s?  Simpler: re-gather lkey by pos after the final sort. */
     {
       hipLaunchKernelGGL(k_gather_lkey, dim3(grid1d(m)), dim3(BLOCK), 0, 0, rt,
-                         (const uint32_t*)pos.p, d_idx, lb0, use_len,
+                         (const uint32_t*)pos.p, d_idx, lb0, use_len, ser_mode,
                          (uint64_t*)lkey.p, m);
     }
     /* apply permutation to d_idx and update eq */
@@ -2000,16 +2032,19 @@ __global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
 }
 
 __global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* sidx,
-                              int lb0, int use_len, uint64_t* lkey, uint32_t m) {
+                              int lb0, int use_len, int ser_mode, uint64_t* lkey,
+                              uint32_t m) {
   for (uint32_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m;
        j += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, sidx[pos[j]]);
+    const uint8_t* src = ser_mode ? v.key : v.content;
+    uint32_t slen = ser_mode ? v.klen : v.clen;
     uint64_t k = 0;
-    if (use_len) k = v.clen;
+    if (use_len) k = slen;
     else
       for (int b = 0; b < 8; b++) {
         uint32_t cb = (uint32_t)(lb0 + b);
-        uint8_t byte = (cb < v.clen) ? v.content[cb] : 0;
+        uint8_t byte = (cb < slen) ? src[cb] : 0;
         k |= (uint64_t)byte << (56 - 8 * b);
       }
     lkey[j] = k;
