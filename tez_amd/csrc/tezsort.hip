@@ -256,7 +256,7 @@ __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32
  * (adaptive pass count — DESIGN.md §4); rarer-than-1% ties are resolved by
  * the refinement levels, whose equality test must see the same mask. */
 __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
-                                  int pbits, int sort_bytes, int ser_mode,
+                                  int pbits, int ref_pb, int sort_bytes, int ser_mode,
                                   uint64_t* d_key, uint32_t* d_idx, uint32_t n) {
   /* d_part == nullptr: compute the HashPartitioner placement here (fused —
      a separate hash kernel re-reads all content bytes, ~10 GB at n=1e8).
@@ -274,14 +274,21 @@ __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
     else part = (uint32_t)((d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P);
     uint64_t key;
     if (ser_mode) {
+      /* the reference prefix keeps only proxy >>> (bitcount(P)+1) bits
+         (PipelinedSorter.java:457): ties on the TRUNCATED proxy fall through
+         to the serialized compare, so the composite must truncate too */
       uint32_t proxy = ((v.clen > 0 ? (uint32_t)v.content[0] : 0u) << 16) |
                        ((v.clen > 1 ? (uint32_t)v.content[1] : 0u) << 8) |
                        (v.clen > 2 ? (uint32_t)v.content[2] : 0u);
+      int pw = 24 - ref_pb;            /* surviving proxy bits */
+      if (pw < 0) pw = 0;
+      uint32_t proxy_t = pw ? (proxy >> (24 - pw)) : 0;
       uint64_t ser = 0;
       uint32_t m = v.klen < 8 ? v.klen : 8;
       for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)v.key[b] << (56 - 8 * b);
-      key = ((uint64_t)part << (64 - pbits)) | ((uint64_t)proxy << (64 - pbits - 24))
-            | (ser >> (pbits + 24));
+      key = ((uint64_t)part << (64 - pbits))
+            | ((uint64_t)proxy_t << (64 - pbits - pw))
+            | (ser >> (pbits + pw));
     } else {
       uint64_t c = 0;
       uint32_t m = v.clen < 8 ? v.clen : 8;
@@ -1591,13 +1598,17 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
      records stay below ~1% of n (ties go through refinement anyway):
      bits = pbits + log2(n) + 6.  Tests at small n exercise refinement hard. */
   int ser_mode = (s->conf.comparator == TZS_CMP_TEZBYTES) ? 1 : 0;
+  int ref_pb = 1; /* bitcount(P)+1, PipelinedSorter.java:165 */
+  for (int v2 = P; v2; v2 >>= 1) ref_pb++;
+  int proxy_w = 24 - ref_pb;
+  if (proxy_w < 0) proxy_w = 0;
   int needed_bits = pbits + 6;
   for (uint64_t v = n; v; v >>= 1) needed_bits++;
   int SB = (needed_bits + 7) / 8;
   if (SB < 2) SB = 2;
   if (ser_mode) {
-    /* partition + full 24-bit proxy must lie inside the sorted bytes */
-    int minsb = (pbits + 24 + 7) / 8;
+    /* partition + surviving proxy bits must lie inside the sorted bytes */
+    int minsb = (pbits + proxy_w + 7) / 8;
     if (SB < minsb) SB = minsb;
   }
   if (SB > 8) SB = 8;
@@ -1607,7 +1618,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   uint64_t* d_key = (uint64_t*)s->skey.p;
   uint32_t* d_idx = (uint32_t*)s->sidx.p;
   hipLaunchKernelGGL(k_build_composite, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                     d_part_unsorted, P, pbits, SB, ser_mode, d_key, d_idx, n);
+                     d_part_unsorted, P, pbits, ref_pb, SB, ser_mode, d_key, d_idx, n);
   (void)hipEventRecord(ev[1]);
 
   /* 2. base radix sort over the top SB bytes of the composite */
@@ -1627,7 +1638,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
      bits are 8*SB - pbits (content mode) or 8*SB - pbits - 24 past the proxy
      (serialized mode; the proxy is rechecked implicitly because serialized
      bytes repeat the content after the 4B length) */
-  int c0 = ser_mode ? (8 * SB - pbits - 24) / 8 : (8 * SB - pbits) / 8;
+  int c0 = ser_mode ? (8 * SB - pbits - proxy_w) / 8 : (8 * SB - pbits) / 8;
   if (c0 < 0) c0 = 0;
   /* determine max clen lazily: use a safe cap by scanning klen on host?  We
      compute it from the conf: key_type BYTES => clen = klen-4 (max over
