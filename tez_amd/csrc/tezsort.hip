@@ -1597,7 +1597,20 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   /* adaptive radix width: enough composite bits that expected tie-involved
      records stay below ~1% of n (ties go through refinement anyway):
      bits = pbits + log2(n) + 6.  Tests at small n exercise refinement hard. */
+  /* TezBytes order is (partition, truncated proxy, serialized bytes).  When
+     every record's serialized klen is equal (uniform spills), the length
+     word is constant and the truncated proxy is a prefix function of the
+     content, so the order reduces to (partition, content) — use the cheap
+     content composite (full 58-bit discrimination) in that case; the
+     faithful serialized composite only for variable-length keys. */
   int ser_mode = (s->conf.comparator == TZS_CMP_TEZBYTES) ? 1 : 0;
+  if (ser_mode) {
+    bool uniform_klen = true;
+    uint32_t k0 = rt.klen_u[0];
+    for (int sp2 = 0; sp2 < rt.nspills; sp2++)
+      if (rt.klen_u[sp2] == 0 || rt.klen_u[sp2] != k0) { uniform_klen = false; break; }
+    if (uniform_klen && k0 != 0) ser_mode = 0;
+  }
   int ref_pb = 1; /* bitcount(P)+1, PipelinedSorter.java:165 */
   for (int v2 = P; v2; v2 >>= 1) ref_pb++;
   int proxy_w = 24 - ref_pb;
