@@ -728,119 +728,13 @@ __global__ void k_sorted_parts(const uint64_t* skeys, int pbits, uint32_t* parts
        i += gridDim.x * blockDim.x)
     parts[i] = pbits ? (uint32_t)(skeys[i] >> (64 - pbits)) : 0;
 }
-/* per-record emit: stream_off[i] = seg_payload_start[part] + (scan[i] - part_scan_base[part]).
- * One WAVE per record: lane-parallel payload copy (the serialized key and
- * value are contiguous in the record, so the payload is one span); lane 0
- * writes the marker/vint header.  Coalesced on both sides. */
+/* per-record IFile emit (replaces the IFile.Writer.append loop,
+ * IFile.java:444-615): batch of 64 records per wave iteration — each lane
+ * fetches ONE record's descriptor in parallel, then two records are kept in
+ * flight per wave (independent 32-lane halves + a 2-deep rotate pipeline)
+ * so gather latency overlaps stores.  Records longer than 128 B take the
+ * simple per-record loop. */
 __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
-                               const uint64_t* scan, const uint32_t* parts,
-                               const uint64_t* seg_payload_start,
-                               const uint64_t* part_scan_base,
-                               uint8_t* out, uint32_t n) {
-  /* Batch of 64 records per wave iteration: each lane fetches ONE record's
-     descriptor in parallel (a single memory-latency exposure covers 64
-     records), then the wave copies the 64 payloads back-to-back via
-     broadcast — streaming, not latency-bound. */
-  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
-  uint32_t lane = threadIdx.x & (WAVE - 1);
-  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
-  for (uint64_t base = (uint64_t)wave * WAVE; base < n;
-       base += (uint64_t)nwaves * WAVE) {
-    uint32_t i = (uint32_t)base + lane;
-    uint64_t my_src = 0, my_dst = 0, my_h0 = 0, my_h1 = 0;
-    uint32_t my_len = 0, my_hdr = 0;
-    if (i < n) {
-      RecDesc v = desc[i];
-      uint32_t p = parts[i];
-      my_dst = seg_payload_start[p] + (scan[i] - part_scan_base[p]);
-      uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
-      uint8_t hdrbuf[16] = {0};
-      uint32_t hdr = 0;
-      if (same[i]) {
-        if (!prev_same) hdrbuf[hdr++] = 0xFE; /* RLE_MARKER -2 */
-        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-        my_src = v.src + v.klen;
-        my_len = v.vlen;
-      } else {
-        if (prev_same) hdrbuf[hdr++] = 0xFD; /* V_END_MARKER -3 */
-        hdr += d_vint_write(hdrbuf + hdr, v.klen);
-        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
-        my_src = v.src;               /* key ‖ val contiguous */
-        my_len = v.klen + v.vlen;
-      }
-      my_hdr = hdr;
-      for (int b = 0; b < 8; b++) my_h0 |= (uint64_t)hdrbuf[b] << (8 * b);
-      for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
-    }
-    uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
-    /* payload-length cap for the software-pipelined path: each lane buffers
-       PIPE_B bytes of a record (covers records up to 64*PIPE_B bytes) */
-    constexpr int PIPE_B = 2;
-    uint32_t maxlen = my_len;
-    for (int sh = 32; sh >= 1; sh >>= 1) {
-      uint32_t o = __shfl_xor(maxlen, sh);
-      if (o > maxlen) maxlen = o;
-    }
-    if (nvalid == WAVE && maxlen <= WAVE * PIPE_B) {
-      /* 2-deep pipeline: issue record r+1's gather while storing record r —
-         the plain loop serializes on one load->store chain per record and
-         runs at ~10% duty cycle */
-      uint64_t src0 = __shfl(my_src, 0), dst0 = __shfl(my_dst, 0);
-      uint64_t h00 = __shfl(my_h0, 0), h10 = __shfl(my_h1, 0);
-      uint32_t len0 = __shfl(my_len, 0), hdr0 = __shfl(my_hdr, 0);
-      uint8_t b0[PIPE_B];
-      const uint8_t* sp0 = (const uint8_t*)(uintptr_t)src0;
-#pragma unroll
-      for (int k = 0; k < PIPE_B; k++)
-        b0[k] = (lane + k * WAVE < len0) ? sp0[lane + k * WAVE] : 0;
-      for (uint32_t r = 0; r < WAVE; r++) {
-        uint64_t src1 = 0, dst1 = 0, h01 = 0, h11 = 0;
-        uint32_t len1 = 0, hdr1 = 0;
-        uint8_t b1[PIPE_B] = {0};
-        if (r + 1 < WAVE) {
-          src1 = __shfl(my_src, r + 1); dst1 = __shfl(my_dst, r + 1);
-          h01 = __shfl(my_h0, r + 1); h11 = __shfl(my_h1, r + 1);
-          len1 = __shfl(my_len, r + 1); hdr1 = __shfl(my_hdr, r + 1);
-          const uint8_t* sp1 = (const uint8_t*)(uintptr_t)src1;
-#pragma unroll
-          for (int k = 0; k < PIPE_B; k++)
-            b1[k] = (lane + k * WAVE < len1) ? sp1[lane + k * WAVE] : 0;
-        }
-        uint8_t* w = out + dst0;
-        if (lane < hdr0)
-          w[lane] = (lane < 8) ? (uint8_t)(h00 >> (8 * lane))
-                               : (uint8_t)(h10 >> (8 * (lane - 8)));
-        w += hdr0;
-#pragma unroll
-        for (int k = 0; k < PIPE_B; k++)
-          if (lane + k * WAVE < len0) w[lane + k * WAVE] = b0[k];
-        src0 = src1; dst0 = dst1; h00 = h01; h10 = h11; len0 = len1; hdr0 = hdr1;
-#pragma unroll
-        for (int k = 0; k < PIPE_B; k++) b0[k] = b1[k];
-      }
-    } else {
-      for (uint32_t r = 0; r < nvalid; r++) {
-        uint64_t src = __shfl(my_src, r);
-        uint64_t dsto = __shfl(my_dst, r);
-        uint64_t h0 = __shfl(my_h0, r);
-        uint64_t h1 = __shfl(my_h1, r);
-        uint32_t len = __shfl(my_len, r);
-        uint32_t hdr = __shfl(my_hdr, r);
-        uint8_t* w = out + dsto;
-        if (lane < hdr)
-          w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
-                               : (uint8_t)(h1 >> (8 * (lane - 8)));
-        w += hdr;
-        const uint8_t* sp = (const uint8_t*)(uintptr_t)src;
-        for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp[b];
-      }
-    }
-  }
-}
-
-/* emit v3 (TZS_EMIT_V=3): two records in flight per wave — independent
- * 32-lane halves halve the number of serial load->store chains per wave. */
-__global__ void k_emit_records_v3(const RecDesc* desc, const uint8_t* same,
                                   const uint64_t* scan, const uint32_t* parts,
                                   const uint64_t* seg_payload_start,
                                   const uint64_t* part_scan_base,
@@ -2034,18 +1928,6 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   if (up(d_rlen, h_range_len.data(), 8 * P)) return -12;
 
   /* 6. emit records */
-  static int emit_v = -1;
-  if (emit_v < 0) {
-    const char* e = getenv("TZS_EMIT_V");
-    emit_v = (e && e[0] == '3') ? 3 : 1;
-  }
-  if (emit_v == 3)
-    hipLaunchKernelGGL(k_emit_records_v3, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
-                       (const RecDesc*)descbuf.p,
-                       (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
-                       (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
-                       (const uint64_t*)d_scanbase.p, d_out, n);
-  else
   hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
                      (const RecDesc*)descbuf.p,
                      (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
