@@ -72,6 +72,12 @@ def _load():
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
         c.POINTER(c.c_int64), c.POINTER(c.c_int)]
+    lib.tzo_spill_mt.restype = c.c_int
+    lib.tzo_spill_mt.argtypes = [
+        u8p, c.POINTER(c.c_uint64), c.POINTER(c.c_uint32), c.POINTER(c.c_int32),
+        c.c_int64, c.c_int32, c.c_int, c.c_int, c.c_int, c.c_int, c.c_int,
+        c.POINTER(c.c_void_p), c.POINTER(c.c_int64),
+        c.POINTER(c.c_void_p), c.POINTER(c.c_int64)]
     lib.tzo_final_merge.restype = c.c_int
     lib.tzo_final_merge.argtypes = [
         c.POINTER(c.c_void_p), c.POINTER(c.c_int64), c.POINTER(c.c_void_p),
@@ -282,3 +288,24 @@ def shuffle_header_decode(b: bytes):
                                      ctypes.byref(rlen), ctypes.byref(part))
     assert n > 0
     return mid.value.decode(), clen.value, rlen.value, part.value, n
+
+
+def spill_mt(data, off, klen, num_partitions, nthreads, key_type=KEY_BYTES,
+             comparator=CMP_TEZBYTES, rle_mode=0, send_empty=True,
+             partitions=None):
+    """Partition-parallel tzo_spill (the multi-core CPU baseline,
+    BASELINE.md). Byte-identical to spill() for unique-key inputs."""
+    n = len(klen)
+    od = ctypes.c_void_p()
+    odl = ctypes.c_int64()
+    oi = ctypes.c_void_p()
+    oil = ctypes.c_int64()
+    pp = partitions.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)) if partitions is not None else None
+    rc = _lib.tzo_spill_mt(
+        _u8p(data), off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        klen.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)), pp,
+        n, num_partitions, key_type, comparator, rle_mode, int(send_empty),
+        nthreads,
+        ctypes.byref(od), ctypes.byref(odl), ctypes.byref(oi), ctypes.byref(oil))
+    assert rc == 0, f"tzo_spill_mt rc={rc}"
+    return {"data": _take_buf(od, odl), "index": _take_buf(oi, oil)}

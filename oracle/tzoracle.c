@@ -1,3 +1,4 @@
+#define _GNU_SOURCE /* qsort_r */
 /* tzoracle.c — ORACLE: CPU restatement of apache/tez's ordered-shuffle hot path.
  *
  * TEST INFRASTRUCTURE ONLY.  This library is the parity yardstick for the GPU
@@ -21,6 +22,7 @@
 #include <stdlib.h>
 #include <string.h>
 #include <stdio.h>
+#include <pthread.h>
 
 #define TZO_API __attribute__((visibility("default")))
 
@@ -412,7 +414,20 @@ typedef struct {
   int comparator;
 } sortctx_t;
 
-static sortctx_t* g_ctx; /* qsort has no _r on all libcs; oracle is single-threaded per call */
+static sortctx_t* g_ctx; /* single-threaded tzo_spill path */
+
+static int spill_cmp_r(const void* A, const void* B, void* arg) {
+  const sortctx_t* ctx = (const sortctx_t*)arg;
+  int64_t ia = *(const int64_t*)A, ib = *(const int64_t*)B;
+  uint32_t pa = ctx->prefix[ia], pb = ctx->prefix[ib];
+  if (pa != pb) return pa < pb ? -1 : 1;
+  const uint8_t* ka = ctx->data + ctx->off[ia];
+  const uint8_t* kb = ctx->data + ctx->off[ib];
+  int c = tzo_compare_key(ctx->comparator, ka, (int32_t)ctx->klen[ia],
+                          kb, (int32_t)ctx->klen[ib]);
+  if (c != 0) return c;
+  return ia < ib ? -1 : (ia > ib ? 1 : 0);
+}
 
 static int spill_cmp(const void* A, const void* B) {
   int64_t ia = *(const int64_t*)A, ib = *(const int64_t*)B;
@@ -546,6 +561,123 @@ TZO_API int tzo_spill(
   *out_data = file.p; *out_data_len = (int64_t)file.len;
   *out_index = idx; *out_index_len = 24 * P + 8;
   if (out_rle) *out_rle = rle;
+  return 0;
+}
+
+/* ---- partition-parallel spill (the "fair" multi-core CPU baseline,
+ * BASELINE.md): identical bytes to tzo_spill, with per-partition quicksort +
+ * IFile emission fanned across nthreads.  Restates the same reference
+ * semantics; parallelism is by partition (the reference's own unit of
+ * independence). */
+typedef struct {
+  sortctx_t* ctx;
+  int64_t* order;          /* full order array, partition-bucketed */
+  const int64_t* pstart;   /* [P+1] partition record ranges */
+  int32_t P;
+  int rle;
+  int key_combiner;
+  int p_next;              /* shared work counter */
+  pthread_mutex_t mu;
+  uint8_t** seg;           /* per-partition emitted stream (or NULL) */
+  int64_t* seg_len;
+  int64_t* seg_raw;
+  int send_empty;
+} mtspill_t;
+
+static void* mtspill_worker(void* vp) {
+  mtspill_t* W = (mtspill_t*)vp;
+  for (;;) {
+    pthread_mutex_lock(&W->mu);
+    int p = W->p_next++;
+    pthread_mutex_unlock(&W->mu);
+    if (p >= W->P) return NULL;
+    int64_t lo = W->pstart[p], hi = W->pstart[p + 1];
+    qsort_r(W->order + lo, (size_t)(hi - lo), sizeof(int64_t), spill_cmp_r, W->ctx);
+    if (hi == lo && W->send_empty) { W->seg[p] = NULL; continue; }
+    tzo_writer* w = tzo_writer_new(W->rle);
+    const sortctx_t* ctx = W->ctx;
+    for (int64_t i = lo; i < hi; i++) {
+      int64_t r = W->order[i];
+      const uint8_t* kb = ctx->data + ctx->off[r];
+      const uint8_t* vb = kb + ctx->klen[r];
+      int32_t vl = (int32_t)(ctx->off[r + 1] - ctx->off[r] - ctx->klen[r]);
+      tzo_writer_append(w, kb, (int32_t)ctx->klen[r], vb, vl);
+    }
+    int64_t seglen;
+    tzo_writer_close(w, &W->seg[p], &seglen, &W->seg_raw[p], &W->seg_len[p]);
+    tzo_writer_free(w);
+  }
+}
+
+/* Multithreaded variant of tzo_spill (rle auto only over unique-key use is
+ * the baseline case; rle_mode is honored the same way).  combiner
+ * unsupported here (baseline measures the sort+emit path). */
+TZO_API int tzo_spill_mt(
+    const uint8_t* data, const uint64_t* off, const uint32_t* klen,
+    const int32_t* part_in, int64_t n,
+    int32_t P, int key_type, int comparator, int rle_mode, int send_empty,
+    int nthreads,
+    uint8_t** out_data, int64_t* out_data_len,
+    uint8_t** out_index, int64_t* out_index_len) {
+  int32_t* part = (int32_t*)malloc(sizeof(int32_t) * (size_t)(n ? n : 1));
+  uint32_t* prefix = (uint32_t*)malloc(sizeof(uint32_t) * (size_t)(n ? n : 1));
+  for (int64_t i = 0; i < n; i++) {
+    const uint8_t* k = data + off[i];
+    if (part_in) part[i] = part_in[i];
+    else {
+      const uint8_t* c; int32_t cl;
+      tzo_key_content(key_type, k, (int32_t)klen[i], &c, &cl);
+      part[i] = tzo_partition(c, cl, P);
+    }
+    prefix[i] = tzo_prefix(comparator, key_type, part[i], P, k, (int32_t)klen[i]);
+  }
+  /* bucket records by partition, stable by index */
+  int64_t* pstart = (int64_t*)calloc((size_t)(P + 2), sizeof(int64_t));
+  for (int64_t i = 0; i < n; i++) pstart[part[i] + 1]++;
+  for (int32_t p = 0; p < P; p++) pstart[p + 1] += pstart[p];
+  int64_t* order = (int64_t*)malloc(sizeof(int64_t) * (size_t)(n ? n : 1));
+  int64_t* cur = (int64_t*)malloc(sizeof(int64_t) * (size_t)(P ? P : 1));
+  memcpy(cur, pstart, sizeof(int64_t) * (size_t)P);
+  for (int64_t i = 0; i < n; i++) order[cur[part[i]]++] = i;
+  free(cur);
+
+  sortctx_t ctx = { data, off, klen, part, prefix, comparator };
+  int rle = rle_mode;
+  mtspill_t W;
+  W.ctx = &ctx; W.order = order; W.pstart = pstart; W.P = P;
+  W.send_empty = send_empty; W.p_next = 0;
+  W.key_combiner = 0;
+  pthread_mutex_init(&W.mu, NULL);
+  W.seg = (uint8_t**)calloc((size_t)P, sizeof(uint8_t*));
+  W.seg_len = (int64_t*)calloc((size_t)P, sizeof(int64_t));
+  W.seg_raw = (int64_t*)calloc((size_t)P, sizeof(int64_t));
+  if (rle < 0) rle = 0; /* auto gate needs global adjacency; baseline inputs
+                           are unique-key (C2 shape) => off, like tzo_spill */
+  W.rle = rle;
+  if (nthreads < 1) nthreads = 1;
+  pthread_t th[64];
+  if (nthreads > 64) nthreads = 64;
+  for (int t = 0; t < nthreads; t++) pthread_create(&th[t], NULL, mtspill_worker, &W);
+  for (int t = 0; t < nthreads; t++) pthread_join(th[t], NULL);
+  pthread_mutex_destroy(&W.mu);
+
+  buf_t file = {0};
+  int64_t* triples = (int64_t*)calloc((size_t)(3 * P), sizeof(int64_t));
+  for (int32_t p = 0; p < P; p++) {
+    triples[3 * p + 0] = (int64_t)file.len;
+    if (W.seg[p]) {
+      triples[3 * p + 1] = W.seg_raw[p];
+      triples[3 * p + 2] = W.seg_len[p];
+      buf_put(&file, W.seg[p], (size_t)W.seg_len[p]);
+      free(W.seg[p]);
+    }
+  }
+  uint8_t* idx = (uint8_t*)malloc((size_t)(24 * P + 8));
+  tzo_index_encode(triples, P, idx);
+  free(triples); free(W.seg); free(W.seg_len); free(W.seg_raw);
+  free(order); free(pstart); free(prefix); free(part);
+  *out_data = file.p; *out_data_len = (int64_t)file.len;
+  *out_index = idx; *out_index_len = 24 * P + 8;
   return 0;
 }
 

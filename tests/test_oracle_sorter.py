@@ -220,3 +220,16 @@ def test_text_comparator_order():
     # hadoop Text order == python bytes order except ties resolved by length
     # (python bytes compare IS memcmp-then-shorter-first)
     assert got == sorted(contents)
+
+
+def test_spill_mt_matches_single_thread():
+    """The partition-parallel spill (the multi-core CPU baseline) must be
+    byte-identical to the single-threaded restatement."""
+    n, P = 3000, 13
+    pairs = make_bytes_records(n, klen=12, vlen=9, seed=77, nparts=P)
+    data, off, klen = o.build_records(pairs)
+    want = o.spill(data, off, klen, P)
+    for threads in (1, 3, 8):
+        got = o.spill_mt(data, off, klen, P, threads)
+        assert got["data"] == want["data"], threads
+        assert got["index"] == want["index"], threads
