@@ -32,6 +32,30 @@ REC_SER = 4 + KLEN + 4 + VLEN  # serialized record bytes (BytesWritable k+v)
 SEED = 0x7E2C2
 
 
+def run_step_c3(tez_amd, gen_batches):
+    """C3: forced multi-spill + k-way merge (BASELINE configs[2]).  Each
+    batch is absorbed and spilled (sorted + IFile-emitted), then flush runs
+    the 32-way merge into the final output."""
+    t0 = time.perf_counter()
+    conf = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
+                             comparator=tez_amd.CMP_TEXT)
+    s = tez_amd.Sorter(conf)
+    for d, off, kl, n in gen_batches:
+        s.write_batch_device(d, off, kl, None, n)
+        s.spill()
+    t1 = time.perf_counter()
+    s.flush()
+    t2 = time.perf_counter()
+    ctr = s.counters()
+    tms = s.times()
+    s.close()
+    tms["host_absorb_ns"] = int((t1 - t0) * 1e9)
+    tms["host_flush_ns"] = int((t2 - t1) * 1e9)
+    tms["host_create_ns"] = 0
+    tms["host_close_ns"] = 0
+    return ctr, tms
+
+
 def run_step_single(tez_amd, conf, d, off, kl, n):
     t0 = time.perf_counter()
     s = tez_amd.Sorter(conf)
@@ -113,6 +137,11 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--records", type=int, default=100_000_000,
                     help="total records across all ranks (C2 default 1e8)")
+    ap.add_argument("--workload", choices=["c2", "c3"], default="c2",
+                    help="c3: 1 GPU, Text/Zipf keys, 256 partitions, forced "
+                         "spills merged at flush (BASELINE configs[2]; "
+                         "--records total, --spills segments)")
+    ap.add_argument("--spills", type=int, default=32)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--force-exchange", action="store_true",
                     help="run the all-to-all-v exchange + reduce-merge path even "
@@ -150,9 +179,21 @@ def main():
 
     n_local = args.records // n_gpus
     conf = tez_amd.make_conf(PARTS)
-    d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local, kind=0,
-                                        klen=KLEN, vlen=VLEN, conf=conf)
-    tez_amd.free_device(part)
+    if args.workload == "c3":
+        assert n_gpus == 1, "c3 is the single-GPU merge config"
+        c3conf = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
+                                   comparator=tez_amd.CMP_TEXT)
+        per = args.records // args.spills
+        gen_batches = []
+        for k in range(args.spills):
+            d, off, kl, part = tez_amd.generate(seed=SEED + 7 * k, n=per, kind=1,
+                                                klen=0, vlen=64, conf=c3conf)
+            tez_amd.free_device(part)
+            gen_batches.append((d, off, kl, per))
+    else:
+        d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local, kind=0,
+                                            klen=KLEN, vlen=VLEN, conf=conf)
+        tez_amd.free_device(part)
 
     def barrier_sync():
         if dist:
@@ -162,6 +203,8 @@ def main():
             torch.cuda.synchronize()
 
     def one_step():
+        if args.workload == "c3":
+            return run_step_c3(tez_amd, gen_batches)
         if use_exchange:
             return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
         return run_step_single(tez_amd, conf, d, off, kl, n_local)
@@ -181,7 +224,11 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    tez_amd.free_device(d, off, kl)
+    if args.workload == "c3":
+        for d, off, kl, _n in gen_batches:
+            tez_amd.free_device(d, off, kl)
+    else:
+        tez_amd.free_device(d, off, kl)
 
     total_bytes_per_step = last_ctr["output_bytes"] * n_gpus  # whole-job Σ
     ms_per_step = elapsed / args.steps * 1e3
@@ -204,7 +251,7 @@ def main():
                         and world == 1 else None),
         }
         cpu = None
-        if not args.skip_cpu_baseline and world == 1:
+        if not args.skip_cpu_baseline and world == 1 and args.workload == "c2":
             cpu = cpu_baseline_line()
         out = {
             "metric": "shuffled+sorted KV bytes/sec",
@@ -220,8 +267,11 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": "C2: 1e8 rec x (16B unique key + 64B val), 64 partitions,"
-                            " BytesWritable/TezBytesComparator, ordered shuffle",
+                "workload": ("C3: Text/Zipf keys 4-32B + 64B vals, 256 partitions, "
+                             f"{args.spills} spills k-way merged"
+                             if args.workload == "c3" else
+                             "C2: 1e8 rec x (16B unique key + 64B val), 64 partitions,"
+                             " BytesWritable/TezBytesComparator, ordered shuffle"),
                 "records": args.records,
                 "key_bytes": KLEN,
                 "value_bytes": VLEN,

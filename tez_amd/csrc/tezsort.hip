@@ -1203,6 +1203,10 @@ static std::unordered_map<size_t, std::vector<void*>>& pool_map() {
   return m;
 }
 static size_t pool_class(size_t n) {
+  /* pow2 classes up to 1 GiB; 256 MiB steps beyond (a 105 GB stream must
+     not round to 128 GB on a 288 GB device) */
+  const size_t GB = 1ull << 30;
+  if (n > GB) return (n + (256ull << 20) - 1) & ~((256ull << 20) - 1);
   size_t c = 1 << 16;
   while (c < n) c <<= 1;
   return c;
@@ -2184,6 +2188,14 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
    * DESIGN.md §4/§5.) */
   int nsp = (int)s->spills.size();
   if (nsp > MAX_SPILLS) FAIL(-22, "too many spills (%d > %d)", nsp, MAX_SPILLS);
+  /* the merge reads the columnar record sets only; the per-spill IFile
+     streams are dead weight now (the reference deletes spill files after the
+     final merge, PipelinedSorter.java:844-849) — free them up front so the
+     1e9-record C3 shape fits in HBM */
+  for (auto* sp2 : s->spills) {
+    sp2->ifile.release();
+    sp2->ifile_len = 0;
+  }
   RecTable rt = {};
   rt.nspills = nsp;
   rt.key_type = s->conf.key_type;
