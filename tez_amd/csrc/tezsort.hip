@@ -738,7 +738,7 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
                                   const uint64_t* scan, const uint32_t* parts,
                                   const uint64_t* seg_payload_start,
                                   const uint64_t* part_scan_base,
-                                  uint8_t* out, uint32_t n) {
+                                  uint8_t* out, uint32_t n, int force_simple) {
   constexpr int HB = 4; /* bytes per lane per record (records to 128 B) */
   uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   uint32_t lane = threadIdx.x & (WAVE - 1);
@@ -779,7 +779,7 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       uint32_t v2 = __shfl_xor(maxlen, sh);
       if (v2 > maxlen) maxlen = v2;
     }
-    if (nvalid == WAVE && maxlen <= 32 * HB) {
+    if (nvalid == WAVE && maxlen <= 32 * HB && !force_simple) {
       /* half h handles records 2r+h; 2-deep rotate pipeline per half */
       uint32_t r0 = half;           /* first record index for this half */
       uint64_t src0 = __shfl(my_src, r0), dst0 = __shfl(my_dst, r0);
@@ -1253,6 +1253,11 @@ struct DBuf {
 /* exclusive scan of u64 array (device), returns total via last+add trick */
 static int scan_u64(const uint64_t* d_in, uint64_t* d_out, uint32_t n, uint64_t* h_total) {
   if (n == 0) { if (h_total) *h_total = 0; return 0; }
+  /* read the last input BEFORE the scan: for in-place scans (d_in == d_out)
+     it is overwritten with the exclusive prefix */
+  uint64_t last_val = 0;
+  if (h_total)
+    HIP_CHECK(hipMemcpy(&last_val, d_in + (n - 1), 8, hipMemcpyDeviceToHost));
   uint32_t nb = nblocks_for(n, SCAN_TILE);
   static thread_local DBuf sums1, sums2, sums3;
   if (sums1.alloc(sizeof(uint64_t) * (nb + 1))) return -12;
@@ -1276,9 +1281,8 @@ static int scan_u64(const uint64_t* d_in, uint64_t* d_out, uint32_t n, uint64_t*
                        (uint64_t*)sums1.p, n);
   }
   if (h_total) {
-    uint64_t last_off = 0, last_val = 0;
+    uint64_t last_off = 0;
     HIP_CHECK(hipMemcpy(&last_off, d_out + (n - 1), 8, hipMemcpyDeviceToHost));
-    HIP_CHECK(hipMemcpy(&last_val, d_in + (n - 1), 8, hipMemcpyDeviceToHost));
     *h_total = last_off + last_val;
   }
   return 0;
@@ -1560,6 +1564,7 @@ extern "C" int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
     uint64_t res[4];
     HIP_CHECK(hipMemcpy(res, mm.p, 32, hipMemcpyDeviceToHost));
     bool uni = (res[0] == res[1]) && (res[2] == res[3]) && res[0] <= 0xFFFFFFFFull;
+    if (getenv("TZS_NO_UNIFORM")) uni = false;
     if (s->cur_first_batch) {
       s->cur_rec_u = uni ? (uint32_t)res[0] : 0;
       s->cur_klen_u = uni ? (uint32_t)res[2] : 0;
@@ -1932,11 +1937,13 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   if (up(d_rlen, h_range_len.data(), 8 * P)) return -12;
 
   /* 6. emit records */
+  static int force_simple = -1;
+  if (force_simple < 0) force_simple = getenv("TZS_EMIT_SIMPLE") ? 1 : 0;
   hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
                      (const RecDesc*)descbuf.p,
                      (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
                      (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
-                     (const uint64_t*)d_scanbase.p, d_out, n);
+                     (const uint64_t*)d_scanbase.p, d_out, n, force_simple);
   (void)hipEventRecord(ev[5]);
 
   /* 7. CRC: chunk bases per partition */
