@@ -59,7 +59,11 @@ typedef struct tzs_conf {
                                           PipelinedSorter.java:602-609,816-821) */
   int32_t min_spills_for_combine;      /* tez.runtime.combine.min.spills, default 3
                                           (PipelinedSorter.java:244) */
-  int32_t reserved0;
+  int32_t discard_spill_streams;       /* 1: free each spill's emitted IFile bytes
+                                          right after the spill completes (valid only
+                                          with final merge enabled; the merge reads
+                                          the columnar set — for workloads near the
+                                          288 GB HBM capacity) */
 } tzs_conf;
 
 void tzs_conf_default(tzs_conf* c, int32_t num_partitions);

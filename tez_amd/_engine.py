@@ -32,7 +32,7 @@ class TzsConf(ctypes.Structure):
         ("rank", ctypes.c_int32),
         ("combiner", ctypes.c_int32),
         ("min_spills_for_combine", ctypes.c_int32),
-        ("reserved0", ctypes.c_int32),
+        ("discard_spill_streams", ctypes.c_int32),
     ]
 
 

@@ -1455,6 +1455,7 @@ extern "C" void tzs_conf_default(tzs_conf* c, int32_t num_partitions) {
   c->rank = 0;
   c->combiner = 0;
   c->min_spills_for_combine = 3; /* TEZ_RUNTIME_COMBINE_MIN_SPILLS, PipelinedSorter.java:244 */
+  c->discard_spill_streams = 0;
 }
 
 extern "C" int tzs_sorter_create(const tzs_conf* conf, tzs_sorter** out) {
@@ -2131,6 +2132,10 @@ extern "C" int tzs_sorter_spill(tzs_sorter* s) {
     sp->rec_u = 0;
     sp->klen_u = 0;
     s->combine_applied = false;
+  }
+  if (s->conf.discard_spill_streams && s->conf.final_merge_enabled) {
+    sp->ifile.release();
+    sp->ifile_len = 0;
   }
   s->spills.push_back(sp);
   s->ctr.spilled_records += sp->n;
