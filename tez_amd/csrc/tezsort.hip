@@ -1292,6 +1292,13 @@ static thread_local int64_t g_scatter_ns = 0;
 static thread_local int64_t g_scatter_launches = 0;
 static thread_local int64_t g_scatter_elems = 0;
 
+static thread_local DBuf g_rs_counts, g_rs_offsets, g_rs_totals, g_rs_bases,
+    g_rs_tk64, g_rs_tk32, g_rs_ta0, g_rs_ta1;
+static void radix_release_temps() {
+  g_rs_counts.release(); g_rs_offsets.release();
+  g_rs_tk64.release(); g_rs_tk32.release(); g_rs_ta0.release(); g_rs_ta1.release();
+}
+
 /* stable LSD radix over KeyT with payload arrays. Sorts in place (ping-pong,
  * result left in the primary arrays). */
 /* dbuf_key/dbuf_a0/dbuf_a1 may be passed so an odd pass count SWAPS the
@@ -1303,7 +1310,10 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                       DBuf* dbuf_a1 = nullptr) {
   if (n <= 1) return 0;
   uint32_t nb = nblocks_for(n, TILE);
-  static thread_local DBuf counts, offsets, totals, bases, tk64, tk32, ta0, ta1;
+  DBuf& counts = g_rs_counts; DBuf& offsets = g_rs_offsets;
+  DBuf& totals = g_rs_totals; DBuf& bases = g_rs_bases;
+  DBuf& tk64 = g_rs_tk64; DBuf& tk32 = g_rs_tk32;
+  DBuf& ta0 = g_rs_ta0; DBuf& ta1 = g_rs_ta1;
   if (counts.alloc(sizeof(uint32_t) * nb * RADIX)) return -12;
   if (offsets.alloc(sizeof(uint32_t) * nb * RADIX)) return -12;
   if (totals.alloc(sizeof(uint32_t) * RADIX)) return -12;
@@ -1748,6 +1758,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                        (const uint32_t*)pos.p, (const uint32_t*)slotpos.p, d_idx,
                        (uint32_t*)idx_new.p, m);
     std::swap(s->sidx, idx_new);
+    idx_new.release(); /* old sidx buffer — back to the pool */
     d_idx = (uint32_t*)s->sidx.p;
     hipLaunchKernelGGL(k_eq_update, dim3(grid1d(m)), dim3(BLOCK), 0, 0,
                        (const uint64_t*)lkey.p, (const uint32_t*)seg.p,
@@ -1755,6 +1766,12 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     level++;
   }
   (void)hipEventRecord(ev[3]);
+  /* large-n headroom: refinement scratch and radix ping-pong temps are dead
+     from here; return them to the pool before the output-stream allocation
+     (at C3's 1e9 records these hold ~60 GB) */
+  inrun.release(); runstart.release(); inrun_scan.release(); runstart_scan.release();
+  lkey.release(); seg.release(); pos.release(); slotpos.release();
+  radix_release_temps();
 
   /* 4. writer-rle decision + same flags */
   uint32_t neq_final = 0;
@@ -1847,6 +1864,8 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                      (uint64_t*)s->sizes.p, n);
   uint64_t total_body = 0;
   if (scan_u64((uint64_t*)s->sizes.p, (uint64_t*)s->scan.p, n, &total_body)) return -12;
+  s->sizes.release();
+  s->skey.release();  /* composites are dead once parts_sorted exists */
   /* partition record ranges: host-side from a partition histogram */
   std::vector<uint32_t> h_pcount(P, 0);
   {
