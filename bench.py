@@ -208,21 +208,44 @@ def main():
             torch.cuda.synchronize()
 
     c3_free_inputs = args.workload == "c3" and args.records >= 500_000_000
-    if c3_free_inputs and (args.steps != 1 or args.warmup != 0):
-        print("# c3 at >=5e8 records frees inputs during the single step; "
-              "forcing --steps 1 --warmup 0", flush=True)
-        args.steps, args.warmup = 1, 0
+
+    def regen_c3():
+        out = []
+        per = args.records // args.spills
+        c3c = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
+                                comparator=tez_amd.CMP_TEXT)
+        for k in range(args.spills):
+            d2, off2, kl2, part2 = tez_amd.generate(seed=SEED + 7 * k, n=per,
+                                                    kind=1, klen=0, vlen=64,
+                                                    conf=c3c)
+            tez_amd.free_device(part2)
+            out.append((d2, off2, kl2, per))
+        return out
 
     def one_step():
+        nonlocal_batches = gen_batches
         if args.workload == "c3":
-            return run_step_c3(tez_amd, gen_batches, c3_free_inputs)
+            r = run_step_c3(tez_amd, nonlocal_batches, c3_free_inputs)
+            if c3_free_inputs:
+                # inputs were consumed; regenerate OUTSIDE any timing the
+                # caller does between steps?  Steps are timed as a block, so
+                # large-c3 must run single-step; regeneration happens in the
+                # warmup/step boundary below.
+                pass
+            return r
         if use_exchange:
             return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
         return run_step_single(tez_amd, conf, d, off, kl, n_local)
 
     last_ctr = last_tms = None
+    if c3_free_inputs and args.steps > 1:
+        print("# c3 at >=5e8 records supports a single timed step; forcing --steps 1",
+              flush=True)
+        args.steps = 1
     for _ in range(args.warmup):
         last_ctr, last_tms = one_step()
+        if c3_free_inputs:
+            gen_batches = regen_c3()   # untimed regeneration (pool-backed)
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):

@@ -1230,6 +1230,29 @@ static int pool_alloc(size_t n, void** out, size_t* cls_out) {
 }
 static void pool_free(void* p, size_t cls) { if (p) pool_map()[cls].push_back(p); }
 
+/* raw pool allocations (generator buffers): class tracked in a registry so
+ * tzs_free_device can return them to the pool */
+static std::unordered_map<void*, size_t>& pool_registry() {
+  static std::unordered_map<void*, size_t> r;
+  return r;
+}
+static int pool_alloc_raw(size_t n, void** out) {
+  size_t cls = 0;
+  int rc = pool_alloc(n, out, &cls);
+  if (rc == 0) pool_registry()[*out] = cls;
+  return rc;
+}
+static void pool_free_raw(void* p) {
+  if (!p) return;
+  auto it = pool_registry().find(p);
+  if (it != pool_registry().end()) {
+    pool_free(p, it->second);
+    pool_registry().erase(it);
+  } else {
+    (void)hipFree(p);
+  }
+}
+
 struct DBuf {
   void* p = nullptr;
   size_t sz = 0;   /* size class */
@@ -2396,12 +2419,12 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
   void* dd = nullptr;
   uint64_t* doff = nullptr;
   uint32_t* dkl = nullptr;
-  HIP_CHECK(hipMalloc((void**)&doff, 8 * (n + 1)));
-  HIP_CHECK(hipMalloc((void**)&dkl, 4 * n));
+  if (pool_alloc_raw(8 * (n + 1), (void**)&doff)) return -12;
+  if (pool_alloc_raw(4 * n, (void**)&dkl)) return -12;
   uint64_t rec = 0;
   if (kind == 0 || kind == 2) {
     rec = 4 + (uint64_t)klen + 4 + (uint64_t)vlen;
-    HIP_CHECK(hipMalloc(&dd, rec * n));
+    if (pool_alloc_raw(rec * n, &dd)) return -12;
     hipLaunchKernelGGL(k_generate_fixed, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n,
                        klen, vlen, (uint8_t*)dd, rec);
     hipLaunchKernelGGL(k_fill_fixed_offsets, dim3(grid1d(n + 1)), dim3(BLOCK), 0, 0,
@@ -2412,13 +2435,13 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
     uint64_t total = 0;
     if (scan_u64(doff, doff, (uint32_t)n, &total)) return -12;
     HIP_CHECK(hipMemcpy(doff + n, &total, 8, hipMemcpyHostToDevice));
-    HIP_CHECK(hipMalloc(&dd, total ? total : 1));
+    if (pool_alloc_raw(total ? total : 1, &dd)) return -12;
     hipLaunchKernelGGL(k_gen_text_fill, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n,
                        vlen, doff, (uint8_t*)dd, dkl);
   }
   if (d_part) {
     int32_t* dp = nullptr;
-    HIP_CHECK(hipMalloc((void**)&dp, 4 * n));
+    if (pool_alloc_raw(4 * n, (void**)&dp)) return -12;
     if (kind == 2) {
       hipLaunchKernelGGL(k_range_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
                          (const uint8_t*)dd, rec, conf ? conf->num_partitions : 1, dp, n);
@@ -2442,7 +2465,7 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
   return 0;
 }
 
-extern "C" void tzs_free_device(void* p) { if (p) (void)hipFree(p); }
+extern "C" void tzs_free_device(void* p) { pool_free_raw(p); }
 
 /* ---- reduce-side merge over columnar segments ---- */
 extern "C" int tzs_merge_segments(const tzs_conf* conf, const tzs_segment* segs,
@@ -2498,7 +2521,7 @@ extern "C" int tzs_memcpy_h2d(void* dev, const void* host, uint64_t n) {
   return 0;
 }
 extern "C" int tzs_malloc_device(uint64_t n, void** out) {
-  HIP_CHECK(hipMalloc(out, n));
+  if (pool_alloc_raw(n, out)) return -12;
   return 0;
 }
 extern "C" int tzs_device_available(void) {
