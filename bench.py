@@ -223,16 +223,8 @@ def main():
         return out
 
     def one_step():
-        nonlocal_batches = gen_batches
         if args.workload == "c3":
-            r = run_step_c3(tez_amd, nonlocal_batches, c3_free_inputs)
-            if c3_free_inputs:
-                # inputs were consumed; regenerate OUTSIDE any timing the
-                # caller does between steps?  Steps are timed as a block, so
-                # large-c3 must run single-step; regeneration happens in the
-                # warmup/step boundary below.
-                pass
-            return r
+            return run_step_c3(tez_amd, gen_batches, c3_free_inputs)
         if use_exchange:
             return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
         return run_step_single(tez_amd, conf, d, off, kl, n_local)
