@@ -120,7 +120,7 @@ def cpu_baseline_line(sample_records=1_000_000):
     t0 = time.perf_counter()
     o.spill(data, offs, klens, PARTS, key_type=o.KEY_BYTES, comparator=o.CMP_TEZBYTES)
     dt1 = time.perf_counter() - t0
-    cores = os.cpu_count() or 1
+    cores = min(os.cpu_count() or 1, PARTS)  # partition-parallel: at most P workers
     t0 = time.perf_counter()
     o.spill_mt(data, offs, klens, PARTS, cores)
     dtm = time.perf_counter() - t0
