@@ -233,9 +233,6 @@ __global__ void k_shift_offsets(const uint64_t* src, uint64_t* dst, uint64_t shi
                                 int64_t n);
 __global__ void k_max_u32(const uint32_t* a, uint32_t n, uint32_t* out);
 __global__ void k_part_hist(const uint32_t* parts, uint32_t n, uint32_t* counts, int P);
-__global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* sidx,
-                              int lb0, int use_len, int ser_mode, uint64_t* lkey,
-                              uint32_t m);
 __global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
                                    const uint8_t* same, uint64_t total_body,
                                    uint32_t n, int P, uint64_t* out);
@@ -369,11 +366,12 @@ __global__ void k_radix_scan_digits(uint32_t* totals, uint32_t* bases) {
  * global writes — a direct per-element scatter write-allocates a 64B line
  * per 12B element (PMC: 22 GB written for 6 GB of payload); digit runs of
  * TILE/256 elements restore coalescing. */
-template <typename KeyT, bool HAS_A1>
+template <typename KeyT, bool HAS_A1, bool HAS_B64 = false>
 __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
     const KeyT* keys_in, KeyT* keys_out,
     const uint32_t* a0_in, uint32_t* a0_out,
     const uint32_t* a1_in, uint32_t* a1_out,
+    const uint64_t* b64_in, uint64_t* b64_out,
     uint32_t n, int byte_idx,
     const uint32_t* offsets, const uint32_t* bases) {
   __shared__ uint32_t tilecnt[RADIX];       /* per-digit running count in tile */
@@ -382,6 +380,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
   __shared__ KeyT ls_key[TILE];
   __shared__ uint32_t ls_a0[TILE];
   __shared__ uint32_t ls_a1[HAS_A1 ? TILE : 1];
+  __shared__ uint64_t ls_b64[HAS_B64 ? TILE : 1];
   __shared__ uint8_t ls_dig[TILE];
   for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
   uint32_t start = blockIdx.x * TILE;
@@ -395,6 +394,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
   KeyT my_key[TILE_ROUNDS];
   uint32_t my_a0[TILE_ROUNDS];
   uint32_t my_a1v[TILE_ROUNDS];
+  uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
   int round = 0;
   for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
     uint32_t i = r0 + threadIdx.x;
@@ -419,6 +419,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
     my_dig[round] = d;
     my_a0[round] = active ? a0_in[i] : 0;
     if (HAS_A1) my_a1v[round] = active ? a1_in[i] : 0;
+    if (HAS_B64) my_b64[round] = active ? b64_in[i] : 0;
     if (active) {
       uint32_t prior = 0;
       for (int w = 0; w < wv; w++) prior += wavehist[w][d];
@@ -460,6 +461,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
       ls_key[slot] = my_key[round];
       ls_a0[slot] = my_a0[round];
       if (HAS_A1) ls_a1[slot] = my_a1v[round];
+      if (HAS_B64) ls_b64[slot] = my_b64[round];
       ls_dig[slot] = (uint8_t)my_dig[round];
     }
   }
@@ -471,6 +473,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
     keys_out[pos] = ls_key[j];
     a0_out[pos] = ls_a0[j];
     if (HAS_A1) a1_out[pos] = ls_a1[j];
+    if (HAS_B64) b64_out[pos] = ls_b64[j];
   }
 }
 
@@ -1317,9 +1320,11 @@ static thread_local int64_t g_scatter_elems = 0;
 
 static thread_local DBuf g_rs_counts, g_rs_offsets, g_rs_totals, g_rs_bases,
     g_rs_tk64, g_rs_tk32, g_rs_ta0, g_rs_ta1;
+static thread_local DBuf g_rs_tb64;
 static void radix_release_temps() {
   g_rs_counts.release(); g_rs_offsets.release();
   g_rs_tk64.release(); g_rs_tk32.release(); g_rs_ta0.release(); g_rs_ta1.release();
+  g_rs_tb64.release();
 }
 
 /* stable LSD radix over KeyT with payload arrays. Sorts in place (ping-pong,
@@ -1330,7 +1335,8 @@ template <typename KeyT>
 static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                       int nbytes_key, int first_byte = 0,
                       DBuf* dbuf_key = nullptr, DBuf* dbuf_a0 = nullptr,
-                      DBuf* dbuf_a1 = nullptr) {
+                      DBuf* dbuf_a1 = nullptr, uint64_t* d_b64 = nullptr,
+                      DBuf* dbuf_b64 = nullptr) {
   if (n <= 1) return 0;
   uint32_t nb = nblocks_for(n, TILE);
   DBuf& counts = g_rs_counts; DBuf& offsets = g_rs_offsets;
@@ -1346,6 +1352,9 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
   if (ta0.alloc(sizeof(uint32_t) * n)) return -12;
   bool has_a1 = d_a1 != nullptr;
   if (has_a1 && ta1.alloc(sizeof(uint32_t) * n)) return -12;
+  bool has_b64 = d_b64 != nullptr;
+  DBuf& tb64 = g_rs_tb64;
+  if (has_b64 && tb64.alloc(sizeof(uint64_t) * n)) return -12;
 
   hipEvent_t evs[16], eve[16];
   int nev = 0;
@@ -1355,6 +1364,8 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
   uint32_t* a0out = (uint32_t*)ta0.p;
   uint32_t* a1in = d_a1;
   uint32_t* a1out = (uint32_t*)ta1.p;
+  uint64_t* b64in = d_b64;
+  uint64_t* b64out = (uint64_t*)tb64.p;
   int passes = 0;
   for (int b = first_byte; b < nbytes_key; b++) {
     hipLaunchKernelGGL((k_radix_hist<KeyT>), dim3(nb), dim3(BLOCK), 0, 0, kin, n, b,
@@ -1365,18 +1376,23 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                        (uint32_t*)totals.p, (uint32_t*)bases.p);
     if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                     (void)hipEventRecord(evs[nev]); }
-    if (has_a1)
+    if (has_b64)
+      hipLaunchKernelGGL((k_radix_scatter<KeyT, false, true>), dim3(nb), dim3(BLOCK),
+                         0, 0, kin, kout, a0in, a0out, nullptr, nullptr, b64in,
+                         b64out, n, b, (uint32_t*)offsets.p, (uint32_t*)bases.p);
+    else if (has_a1)
       hipLaunchKernelGGL((k_radix_scatter<KeyT, true>), dim3(nb), dim3(BLOCK), 0, 0,
-                         kin, kout, a0in, a0out, a1in, a1out, n, b,
+                         kin, kout, a0in, a0out, a1in, a1out, nullptr, nullptr, n, b,
                          (uint32_t*)offsets.p, (uint32_t*)bases.p);
     else
       hipLaunchKernelGGL((k_radix_scatter<KeyT, false>), dim3(nb), dim3(BLOCK), 0, 0,
-                         kin, kout, a0in, a0out, nullptr, nullptr, n, b,
-                         (uint32_t*)offsets.p, (uint32_t*)bases.p);
+                         kin, kout, a0in, a0out, nullptr, nullptr, nullptr, nullptr,
+                         n, b, (uint32_t*)offsets.p, (uint32_t*)bases.p);
     if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
     std::swap(kin, kout);
     std::swap(a0in, a0out);
     if (has_a1) std::swap(a1in, a1out);
+    if (has_b64) std::swap(b64in, b64out);
     passes++;
   }
   (void)hipDeviceSynchronize();
@@ -1389,16 +1405,20 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     (void)hipEventDestroy(evs[e]); (void)hipEventDestroy(eve[e]);
   }
   if (passes & 1) {
-    if (dbuf_key && dbuf_a0 && (!has_a1 || dbuf_a1)) {
+    if (dbuf_key && dbuf_a0 && (!has_a1 || dbuf_a1) && (!has_b64 || dbuf_b64)) {
       /* hand the temp buffers to the caller; keep the old primaries as temps */
       std::swap(*dbuf_key, tk);
       std::swap(*dbuf_a0, ta0);
       if (has_a1) std::swap(*dbuf_a1, ta1);
+      if (has_b64) std::swap(*dbuf_b64, tb64);
     } else {
       HIP_CHECK(hipMemcpyAsync(d_key, kin, sizeof(KeyT) * n, hipMemcpyDeviceToDevice));
       HIP_CHECK(hipMemcpyAsync(d_a0, a0in, sizeof(uint32_t) * n, hipMemcpyDeviceToDevice));
       if (has_a1)
         HIP_CHECK(hipMemcpyAsync(d_a1, a1in, sizeof(uint32_t) * n,
+                                 hipMemcpyDeviceToDevice));
+      if (has_b64)
+        HIP_CHECK(hipMemcpyAsync(d_b64, b64in, sizeof(uint64_t) * n,
                                  hipMemcpyDeviceToDevice));
     }
   }
@@ -1761,23 +1781,16 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     /* slotpos = copy of pos (ascending) before sort */
     HIP_CHECK(hipMemcpyAsync(slotpos.p, pos.p, sizeof(uint32_t) * m,
                              hipMemcpyDeviceToDevice));
-    /* sort compact by (seg, lkey): LSD lkey bytes then seg bytes */
+    /* sort compact by (seg, lkey): LSD lkey bytes then seg bytes; the seg
+       passes carry lkey as a 64-bit payload so it stays aligned for the
+       eq update (re-gathering it measured ~210 ms/step at C3 1e9) */
     int segbytes = 1;
     while ((nruns >> (8 * segbytes)) != 0 && segbytes < 4) segbytes++;
     rc = radix_sort<uint64_t>((uint64_t*)lkey.p, (uint32_t*)seg.p, (uint32_t*)pos.p, m, 8);
     if (rc) return rc;
-    rc = radix_sort<uint32_t>((uint32_t*)seg.p, (uint32_t*)pos.p, nullptr, m, segbytes);
+    rc = radix_sort<uint32_t>((uint32_t*)seg.p, (uint32_t*)pos.p, nullptr, m, segbytes,
+                              0, nullptr, nullptr, nullptr, (uint64_t*)lkey.p, &lkey);
     if (rc) return rc;
-
-    /* wait: after the seg sort, lkey is NOT permuted alongside (we passed pos
-       as aux of seg sort, losing lkey alignment).  Redo: sort with lkey as
-       aux too.  radix_sort<uint32_t> carries only u32 payloads — carry lkey
-       as two u32s?  Simpler: re-gather lkey by pos after the final sort. */
-    {
-      hipLaunchKernelGGL(k_gather_lkey, dim3(grid1d(m)), dim3(BLOCK), 0, 0, rt,
-                         (const uint32_t*)pos.p, d_idx, lb0, use_len, ser_mode,
-                         (uint64_t*)lkey.p, m);
-    }
     /* apply permutation to d_idx and update eq */
     static thread_local DBuf idx_new;
     if (idx_new.alloc(sizeof(uint32_t) * n)) return -12;
@@ -2115,25 +2128,6 @@ __global__ void k_gather_part_meta(const uint64_t* pstart, const uint64_t* scan,
   }
 }
 
-__global__ void k_gather_lkey(RecTable rt, const uint32_t* pos, const uint32_t* sidx,
-                              int lb0, int use_len, int ser_mode, uint64_t* lkey,
-                              uint32_t m) {
-  for (uint32_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m;
-       j += gridDim.x * blockDim.x) {
-    RecView v = rt_view(rt, sidx[pos[j]]);
-    const uint8_t* src = ser_mode ? v.key : v.content;
-    uint32_t slen = ser_mode ? v.klen : v.clen;
-    uint64_t k = 0;
-    if (use_len) k = slen;
-    else
-      for (int b = 0; b < 8; b++) {
-        uint32_t cb = (uint32_t)(lb0 + b);
-        uint8_t byte = (cb < slen) ? src[cb] : 0;
-        k |= (uint64_t)byte << (56 - 8 * b);
-      }
-    lkey[j] = k;
-  }
-}
 
 extern "C" int tzs_sorter_spill(tzs_sorter* s) {
   int rc = absorb_host_staging(s);
