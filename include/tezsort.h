@@ -151,8 +151,6 @@ void tzs_sorter_close(tzs_sorter* s);
  * Segments are device-resident columnar record sets (the xGMI exchange wire,
  * DESIGN.md §4) or IFile-framed bytes.
  */
-typedef struct tzs_merge tzs_merge;
-
 typedef struct tzs_segment {
   const void*     d_data;   /* columnar records (key‖val serialized) */
   const uint64_t* d_off;    /* [n+1] */

@@ -713,14 +713,6 @@ static int seg_key_cmp(int comparator, mseg_t* a, mseg_t* b) {
   if (c != 0) return c;
   return a->seq - b->seq;
 }
-static void heap_up(mheap_t* h, int i) {
-  while (i > 0) {
-    int p = (i - 1) / 2;
-    if (seg_key_cmp(h->comparator, h->heap[i], h->heap[p]) < 0) {
-      mseg_t* t = h->heap[i]; h->heap[i] = h->heap[p]; h->heap[p] = t; i = p;
-    } else break;
-  }
-}
 static void heap_down(mheap_t* h, int i) {
   for (;;) {
     int l = 2 * i + 1, r = l + 1, m = i;

@@ -1564,9 +1564,11 @@ static int absorb_host_staging(tzs_sorter* s) {
   return rc;
 }
 
+#include <chrono>
 extern "C" int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
                                              const uint64_t* d_off, const uint32_t* d_klen,
                                              const int32_t* d_part, int64_t n) {
+  auto t0 = std::chrono::steady_clock::now();
   if (n == 0) return 0;
   if (n > 0xFFFFFFFll) FAIL(-22, "batch too large");
   /* append into current buffer (device-side copy) */
@@ -1634,6 +1636,8 @@ extern "C" int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
   s->ctr.output_records += n;
   s->ctr.output_bytes += (int64_t)nbytes; /* serialized k+v bytes (framing incl.) */
   HIP_CHECK(hipDeviceSynchronize());
+  s->times.absorb_ns += std::chrono::duration_cast<std::chrono::nanoseconds>(
+      std::chrono::steady_clock::now() - t0).count();
   return 0;
 }
 
