@@ -160,6 +160,23 @@ __device__ __forceinline__ int d_vint_decoded_size(int8_t first) {
 __device__ __forceinline__ int32_t d_hash_bytes(const uint8_t* p, int32_t n) {
   /* h = 31*h + signed(byte) — fetch 8 bytes at a time (byte loads cost an
      instruction each and serialize the chain on memory latency) */
+  if (n == 16) {
+    /* split the 16-step chain into two independent 8-chains:
+       h = A*31^8 + B with A seeded 1, B seeded 0 (exact same mod-2^32 value) */
+    uint64_t w0, w1;
+    __builtin_memcpy(&w0, p, 8);
+    __builtin_memcpy(&w1, p + 8, 8);
+    int32_t A = 1, B = 0;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      A = (int32_t)((uint32_t)A * 31u) + (int8_t)(uint8_t)(w0 >> (8 * j));
+      B = (int32_t)((uint32_t)B * 31u) + (int8_t)(uint8_t)(w1 >> (8 * j));
+    }
+    uint32_t p8 = 1;
+#pragma unroll
+    for (int j = 0; j < 8; j++) p8 *= 31u;
+    return (int32_t)((uint32_t)A * p8 + (uint32_t)B);
+  }
   int32_t h = 1;
   int32_t i = 0;
   for (; i + 8 <= n; i += 8) {
@@ -873,7 +890,7 @@ __global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
  * reduces its chunk list sequentially with the 256-byte shift matrix. */
 #define CRC_CHUNK 256
 __constant__ uint32_t c_crc_table[256];
-__constant__ uint32_t c_crc_table4[4][256]; /* slice-by-4: T0 = standard */
+__constant__ uint32_t c_crc_table8[8][256]; /* slice-by-8: T0 = standard */
 __constant__ uint32_t c_crc_mats[CRC_MATS][32];
 
 __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
@@ -900,10 +917,10 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     const uint8_t* stream, const uint64_t* range_start, const uint64_t* range_len,
     const uint64_t* chunk_base, const uint64_t* sc_base /* [P+1] */,
     uint32_t nparts, uint32_t total_sc, uint32_t* chunk_crc) {
-  __shared__ uint32_t tab4[4][256];
+  __shared__ uint32_t tab8[8][256];
   __shared__ uint32_t stage32[CRC_SC_CHUNKS * CRC_ROW_W];
-  for (int i = threadIdx.x; i < 1024; i += blockDim.x)
-    ((uint32_t*)tab4)[i] = ((const uint32_t*)c_crc_table4)[i];
+  for (int i = threadIdx.x; i < 2048; i += blockDim.x)
+    ((uint32_t*)tab8)[i] = ((const uint32_t*)c_crc_table8)[i];
   for (uint32_t sc = blockIdx.x; sc < total_sc; sc += gridDim.x) {
     uint32_t lo = 0, hi = nparts;
     while (lo + 1 < hi) {
@@ -933,15 +950,18 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
       if (len > CRC_CHUNK) len = CRC_CHUNK;
       uint32_t crc = 0xFFFFFFFFu;
       const uint32_t* row = stage32 + t * CRC_ROW_W;
-      uint32_t full = (uint32_t)(len / 4);
-      for (uint32_t w = 0; w < full; w++) {
-        uint32_t c = crc ^ row[w];
-        crc = tab4[3][c & 0xFF] ^ tab4[2][(c >> 8) & 0xFF]
-            ^ tab4[1][(c >> 16) & 0xFF] ^ tab4[0][c >> 24];
+      uint32_t full8 = (uint32_t)(len / 8);
+      for (uint32_t w = 0; w < full8; w++) {
+        uint32_t lo = crc ^ row[2 * w];
+        uint32_t hi = row[2 * w + 1];
+        crc = tab8[7][lo & 0xFF] ^ tab8[6][(lo >> 8) & 0xFF]
+            ^ tab8[5][(lo >> 16) & 0xFF] ^ tab8[4][lo >> 24]
+            ^ tab8[3][hi & 0xFF] ^ tab8[2][(hi >> 8) & 0xFF]
+            ^ tab8[1][(hi >> 16) & 0xFF] ^ tab8[0][hi >> 24];
       }
-      for (uint32_t b = full * 4; b < len; b++) {
+      for (uint32_t b = full8 * 8; b < len; b++) {
         uint8_t byte = (uint8_t)(row[b / 4] >> (8 * (b & 3)));
-        crc = (crc >> 8) ^ tab4[0][(crc ^ byte) & 0xFF];
+        crc = (crc >> 8) ^ tab8[0][(crc ^ byte) & 0xFF];
       }
       chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + t] = crc ^ 0xFFFFFFFFu;
     }
@@ -1164,12 +1184,12 @@ static int ensure_device_constants() {
   h_build_crc_mats();
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table), h_crc_table, sizeof(h_crc_table)));
   {
-    static uint32_t t4[4][256];
-    for (int i = 0; i < 256; i++) t4[0][i] = h_crc_table[i];
-    for (int k = 1; k < 4; k++)
+    static uint32_t t8[8][256];
+    for (int i = 0; i < 256; i++) t8[0][i] = h_crc_table[i];
+    for (int k = 1; k < 8; k++)
       for (int i = 0; i < 256; i++)
-        t4[k][i] = (t4[k - 1][i] >> 8) ^ h_crc_table[t4[k - 1][i] & 0xFF];
-    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table4), t4, sizeof(t4)));
+        t8[k][i] = (t8[k - 1][i] >> 8) ^ h_crc_table[t8[k - 1][i] & 0xFF];
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table8), t8, sizeof(t8)));
   }
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_mats), h_crc_shift_mat, sizeof(h_crc_shift_mat)));
   done = true;
