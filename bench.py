@@ -61,6 +61,26 @@ def run_step_c3(tez_amd, gen_batches, free_inputs=False):
     return ctr, tms
 
 
+def run_step_c5(tez_amd, d, off, kl, part, n):
+    """C5 slice: range-partitioned (TotalOrderPartitioner-style) explicit
+    placements through the C-ABI, one spill, flush = rename."""
+    t0 = time.perf_counter()
+    conf = tez_amd.make_conf(128)
+    s = tez_amd.Sorter(conf)
+    s.write_batch_device(d, off, kl, part, n)
+    t1 = time.perf_counter()
+    s.flush()
+    t2 = time.perf_counter()
+    ctr = s.counters()
+    tms = s.times()
+    s.close()
+    tms["host_create_ns"] = 0
+    tms["host_absorb_ns"] = int((t1 - t0) * 1e9)
+    tms["host_flush_ns"] = int((t2 - t1) * 1e9)
+    tms["host_close_ns"] = 0
+    return ctr, tms
+
+
 def run_step_single(tez_amd, conf, d, off, kl, n):
     t0 = time.perf_counter()
     s = tez_amd.Sorter(conf)
@@ -142,10 +162,12 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--records", type=int, default=100_000_000,
                     help="total records across all ranks (C2 default 1e8)")
-    ap.add_argument("--workload", choices=["c2", "c3"], default="c2",
+    ap.add_argument("--workload", choices=["c2", "c3", "c5"], default="c2",
                     help="c3: 1 GPU, Text/Zipf keys, 256 partitions, forced "
                          "spills merged at flush (BASELINE configs[2]; "
-                         "--records total, --spills segments)")
+                         "--records total, --spills segments). "
+                         "c5: TeraSort-shaped 10B key + 90B value, range "
+                         "partitions (single-GPU slice of configs[4])")
     ap.add_argument("--spills", type=int, default=32)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--force-exchange", action="store_true",
@@ -184,7 +206,16 @@ def main():
 
     n_local = args.records // n_gpus
     conf = tez_amd.make_conf(PARTS)
-    if args.workload == "c3":
+    c5_part = None
+    if args.workload == "c5":
+        assert n_gpus == 1, "c5 bench line is the single-GPU slice"
+        c5conf = tez_amd.make_conf(128)
+        if args.records == 100_000_000:
+            args.records = 500_000_000  # default C5 slice: 5e8 x 100B = 50 GB
+        d, off, kl, c5_part = tez_amd.generate(seed=SEED + 5, n=args.records,
+                                               kind=2, klen=10, vlen=90,
+                                               conf=c5conf)
+    elif args.workload == "c3":
         assert n_gpus == 1, "c3 is the single-GPU merge config"
         c3conf = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
                                    comparator=tez_amd.CMP_TEXT)
@@ -223,6 +254,8 @@ def main():
         return out
 
     def one_step():
+        if args.workload == "c5":
+            return run_step_c5(tez_amd, d, off, kl, c5_part, args.records)
         if args.workload == "c3":
             return run_step_c3(tez_amd, gen_batches, c3_free_inputs)
         if use_exchange:
@@ -254,6 +287,8 @@ def main():
         if not c3_free_inputs:
             for d, off, kl, _n in gen_batches:
                 tez_amd.free_device(d, off, kl)
+    elif args.workload == "c5":
+        tez_amd.free_device(d, off, kl, c5_part)
     else:
         tez_amd.free_device(d, off, kl)
 
@@ -294,7 +329,10 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": ("C3: Text/Zipf keys 4-32B + 64B vals, 256 partitions, "
+                "workload": ("C5 slice: TeraSort 10B key + 90B value, 128 range "
+                             "partitions (TotalOrderPartitioner-style)"
+                             if args.workload == "c5" else
+                             "C3: Text/Zipf keys 4-32B + 64B vals, 256 partitions, "
                              f"{args.spills} spills k-way merged"
                              if args.workload == "c3" else
                              "C2: 1e8 rec x (16B unique key + 64B val), 64 partitions,"

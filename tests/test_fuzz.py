@@ -128,7 +128,7 @@ def test_gpu_randomized_config_sweep():
     if not tez_amd.device_available():
         pytest.skip("no GPU")
     rng = random.Random(0xF1122)
-    for trial in range(14):
+    for trial in range(20):
         P = rng.choice([1, 2, 3, 7, 16, 63, 200])
         text = rng.random() < 0.5
         dup = rng.random() < 0.4
