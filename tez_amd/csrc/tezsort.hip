@@ -1590,7 +1590,7 @@ extern "C" int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
                                              const int32_t* d_part, int64_t n) {
   auto t0 = std::chrono::steady_clock::now();
   if (n == 0) return 0;
-  if (n > 0xFFFFFFFll) FAIL(-22, "batch too large");
+  if (n > 4000000000ll) FAIL(-22, "batch too large (u32 record ids)");
   /* append into current buffer (device-side copy) */
   uint64_t nbytes = 0, first = 0;
   HIP_CHECK(hipMemcpy(&first, d_off, 8, hipMemcpyDeviceToHost));
