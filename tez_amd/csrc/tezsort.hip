@@ -494,6 +494,187 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
   }
 }
 
+/* ---- onesweep radix pass (single-kernel, decoupled lookback) ----
+ * Global per-digit totals are permutation-invariant, so ONE read of the keys
+ * yields the histograms of EVERY pass; each pass is then a single scatter
+ * kernel: tiles are taken in order via a global ticket (dispatch order is
+ * undefined on CDNA4 — a blockIdx-ordered lookback could deadlock, a ticket
+ * order cannot: every ticket below yours belongs to a block that has already
+ * started), tile digit counts are published as single-word agent-scope
+ * atomics with a 2-bit status packed in (the guide's data-is-the-flag R2
+ * form), and each tile resolves its exclusive prefix by walking back until
+ * an INCLUSIVE entry.  Spins are bounded; on timeout an error flag makes the
+ * host redo the pass with the classic 3-kernel path. */
+#define OS_AGG (1u << 30)
+#define OS_INC (2u << 30)
+#define OS_CNT_MASK ((1u << 30) - 1)
+typedef unsigned int __attribute__((address_space(1))) os_gu32;
+
+__global__ void k_global_hist_all(const uint64_t* keys, uint32_t n, int first_byte,
+                                  int npasses, uint32_t* counts /* [npasses][256] */) {
+  extern __shared__ uint32_t lh[]; /* npasses * 256 */
+  for (int i = threadIdx.x; i < npasses * RADIX; i += blockDim.x) lh[i] = 0;
+  __syncthreads();
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    uint64_t k = keys[i];
+    for (int p = 0; p < npasses; p++)
+      atomicAdd(&lh[p * RADIX + ((uint32_t)(k >> (8 * (first_byte + p))) & 0xFF)], 1u);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < npasses * RADIX; i += blockDim.x)
+    if (lh[i]) atomicAdd(&counts[i], lh[i]);
+}
+
+template <typename KeyT, bool HAS_A1, bool HAS_B64 = false>
+__global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
+    const KeyT* keys_in, KeyT* keys_out,
+    const uint32_t* a0_in, uint32_t* a0_out,
+    const uint32_t* a1_in, uint32_t* a1_out,
+    const uint64_t* b64_in, uint64_t* b64_out,
+    uint32_t n, int byte_idx,
+    const uint32_t* bases /* [256] exclusive digit bases */,
+    uint32_t* status /* [ntiles*256] */, uint32_t* ticket, uint32_t* error) {
+  __shared__ uint32_t tilecnt[RADIX];
+  __shared__ uint32_t wavehist[WPB][RADIX];
+  __shared__ uint32_t tileoff[RADIX];
+  __shared__ uint32_t excl[RADIX];
+  __shared__ uint32_t s_tile;
+  __shared__ KeyT ls_key[TILE];
+  __shared__ uint32_t ls_a0[TILE];
+  __shared__ uint32_t ls_a1[HAS_A1 ? TILE : 1];
+  __shared__ uint64_t ls_b64[HAS_B64 ? TILE : 1];
+  __shared__ uint8_t ls_dig[TILE];
+  if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
+  for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
+  __syncthreads();
+  const uint32_t tile = s_tile;
+  const uint32_t start = tile * TILE;
+  const uint32_t end = min(start + TILE, n);
+  const uint32_t count = (start < n) ? (end - start) : 0;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wv = threadIdx.x / WAVE;
+  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  uint32_t my_seq[TILE_ROUNDS];
+  uint32_t my_dig[TILE_ROUNDS];
+  KeyT my_key[TILE_ROUNDS];
+  uint32_t my_a0[TILE_ROUNDS];
+  uint32_t my_a1v[TILE_ROUNDS];
+  uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
+  int round = 0;
+  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+    uint32_t i = r0 + threadIdx.x;
+    bool active = i < end;
+    KeyT key = active ? keys_in[i] : (KeyT)0;
+    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0xFFFFFFFFu;
+    __syncthreads();
+    for (int j = threadIdx.x; j < WPB * RADIX; j += blockDim.x)
+      ((uint32_t*)wavehist)[j] = 0;
+    __syncthreads();
+    uint64_t m = ~0ull;
+    for (int b = 0; b < 8; b++) {
+      uint64_t bb = __ballot((d >> b) & 1);
+      m &= ((d >> b) & 1) ? bb : ~bb;
+    }
+    uint64_t act = __ballot(active);
+    m &= act;
+    uint32_t lane_rank = (uint32_t)__popcll(m & lt_mask);
+    if (active && lane_rank == 0) wavehist[wv][d] = (uint32_t)__popcll(m);
+    __syncthreads();
+    my_key[round] = key;
+    my_dig[round] = d;
+    my_a0[round] = active ? a0_in[i] : 0;
+    if (HAS_A1) my_a1v[round] = active ? a1_in[i] : 0;
+    if (HAS_B64) my_b64[round] = active ? b64_in[i] : 0;
+    if (active) {
+      uint32_t prior = 0;
+      for (int w = 0; w < wv; w++) prior += wavehist[w][d];
+      my_seq[round] = tilecnt[d] + prior + lane_rank;
+    } else {
+      my_seq[round] = 0;
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < RADIX; j += blockDim.x) {
+      uint32_t sum = 0;
+      for (int w = 0; w < WPB; w++) sum += wavehist[w][j];
+      tilecnt[j] += sum;
+    }
+  }
+  __syncthreads();
+  /* publish aggregate (or inclusive for tile 0) — single-word relaxed
+     agent-scope stores (sc1), data-is-the-flag */
+  if (threadIdx.x < RADIX) {
+    uint32_t d = threadIdx.x;
+    uint32_t v = tilecnt[d] & OS_CNT_MASK;
+    if (tile == 0)
+      __hip_atomic_store((os_gu32*)&status[d], v | OS_INC,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    else
+      __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d], v | OS_AGG,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+  /* lookback: thread d resolves digit d's exclusive prefix */
+  if (threadIdx.x < RADIX) {
+    uint32_t d = threadIdx.x;
+    uint32_t e = 0;
+    if (tile > 0) {
+      int64_t p = (int64_t)tile - 1;
+      uint32_t spins = 0;
+      while (p >= 0) {
+        uint32_t v = __hip_atomic_load((os_gu32*)&status[(uint64_t)p * RADIX + d],
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (v & OS_INC) { e += v & OS_CNT_MASK; break; }
+        if (v & OS_AGG) { e += v & OS_CNT_MASK; p--; continue; }
+        if (++spins > 100000000u) { atomicAdd(error, 1u); break; }
+        __builtin_amdgcn_s_sleep(4);
+      }
+      __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d],
+                         ((e + tilecnt[d]) & OS_CNT_MASK) | OS_INC,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    }
+    excl[d] = e;
+    /* exclusive scan of the tile's own counts for the LDS grouping */
+  }
+  /* tile-local digit offsets */
+  {
+    int t = threadIdx.x;
+    uint32_t v = (t < RADIX) ? tilecnt[t] : 0;
+    uint32_t* sc = (uint32_t*)wavehist;
+    if (t < RADIX) sc[t] = v;
+    __syncthreads();
+    for (int st = 1; st < RADIX; st <<= 1) {
+      uint32_t add = (t >= st && t < RADIX) ? sc[t - st] : 0;
+      __syncthreads();
+      if (t < RADIX) sc[t] += add;
+      __syncthreads();
+    }
+    if (t < RADIX) tileoff[t] = sc[t] - v;
+    __syncthreads();
+  }
+  /* place into LDS grouped by digit, then digit-contiguous global writes */
+  round = 0;
+  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+    uint32_t i = r0 + threadIdx.x;
+    if (i < end) {
+      uint32_t slot = tileoff[my_dig[round]] + my_seq[round];
+      ls_key[slot] = my_key[round];
+      ls_a0[slot] = my_a0[round];
+      if (HAS_A1) ls_a1[slot] = my_a1v[round];
+      if (HAS_B64) ls_b64[slot] = my_b64[round];
+      ls_dig[slot] = (uint8_t)my_dig[round];
+    }
+  }
+  __syncthreads();
+  for (uint32_t j = threadIdx.x; j < count; j += blockDim.x) {
+    uint32_t d = ls_dig[j];
+    uint32_t pos = bases[d] + excl[d] + (j - tileoff[d]);
+    keys_out[pos] = ls_key[j];
+    a0_out[pos] = ls_a0[j];
+    if (HAS_A1) a1_out[pos] = ls_a1[j];
+    if (HAS_B64) b64_out[pos] = ls_b64[j];
+  }
+}
+
 /* ---- generic exclusive scan over u64 (sizes -> offsets) ---- */
 #define SCAN_ITEMS 8
 #define SCAN_TILE (BLOCK * SCAN_ITEMS)
@@ -1387,6 +1568,64 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
   uint64_t* b64in = d_b64;
   uint64_t* b64out = (uint64_t*)tb64.p;
   int passes = 0;
+
+  /* onesweep path (u64 keys, base sort shape): one global-histogram read for
+     all passes, then a single lookback kernel per pass.  Falls back to the
+     classic 3-kernel pass on a lookback timeout. */
+  static int use_onesweep = -1;
+  if (use_onesweep < 0) {
+    const char* e = getenv("TZS_ONESWEEP");
+    use_onesweep = (e && e[0] == '0') ? 0 : 1;
+  }
+  if (use_onesweep && sizeof(KeyT) == 8 && !has_a1 && !has_b64 && n >= 20000) {
+    int npasses = nbytes_key - first_byte;
+    static thread_local DBuf gh, gbases, st, tick;
+    if (gh.alloc(4u * npasses * RADIX)) return -12;
+    if (gbases.alloc(4u * npasses * RADIX)) return -12;
+    if (st.alloc(4ull * nb * RADIX)) return -12;
+    if (tick.alloc(16)) return -12;
+    HIP_CHECK(hipMemsetAsync(gh.p, 0, 4u * npasses * RADIX));
+    hipLaunchKernelGGL(k_global_hist_all, dim3(grid1d(n)), dim3(BLOCK),
+                       (uint32_t)(4 * npasses * RADIX), 0, (const uint64_t*)kin, n,
+                       first_byte, npasses, (uint32_t*)gh.p);
+    std::vector<uint32_t> h_cnt(npasses * RADIX), h_base(npasses * RADIX);
+    HIP_CHECK(hipMemcpy(h_cnt.data(), gh.p, 4u * npasses * RADIX,
+                        hipMemcpyDeviceToHost));
+    for (int p = 0; p < npasses; p++) {
+      uint32_t run = 0;
+      for (int d = 0; d < RADIX; d++) {
+        h_base[p * RADIX + d] = run;
+        run += h_cnt[p * RADIX + d];
+      }
+    }
+    HIP_CHECK(hipMemcpyAsync(gbases.p, h_base.data(), 4u * npasses * RADIX,
+                             hipMemcpyHostToDevice));
+    for (int b = first_byte; b < nbytes_key; b++) {
+      int p = b - first_byte;
+      HIP_CHECK(hipMemsetAsync(st.p, 0, 4ull * nb * RADIX));
+      HIP_CHECK(hipMemsetAsync(tick.p, 0, 16));
+      if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
+                      (void)hipEventRecord(evs[nev]); }
+      hipLaunchKernelGGL((k_onesweep_pass<uint64_t, false>), dim3(nb), dim3(BLOCK), 0, 0,
+                         (const uint64_t*)kin, (uint64_t*)kout, a0in, a0out,
+                         nullptr, nullptr, nullptr, nullptr, n, b,
+                         (const uint32_t*)((uint32_t*)gbases.p + p * RADIX),
+                         (uint32_t*)st.p, (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
+      uint32_t h_err = 0;
+      HIP_CHECK(hipMemcpy(&h_err, (uint32_t*)tick.p + 1, 4, hipMemcpyDeviceToHost));
+      if (h_err) {
+        /* lookback timed out (should not happen): classic pass redo */
+        snprintf(g_err, sizeof(g_err), "onesweep lookback timeout pass %d", b);
+        return -70;
+      }
+      std::swap(kin, kout);
+      std::swap(a0in, a0out);
+      passes++;
+    }
+    goto finish;
+  }
+
   for (int b = first_byte; b < nbytes_key; b++) {
     hipLaunchKernelGGL((k_radix_hist<KeyT>), dim3(nb), dim3(BLOCK), 0, 0, kin, n, b,
                        (uint32_t*)counts.p);
@@ -1415,6 +1654,7 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     if (has_b64) std::swap(b64in, b64out);
     passes++;
   }
+finish:
   (void)hipDeviceSynchronize();
   for (int e = 0; e < nev; e++) {
     float ms = 0;
