@@ -613,28 +613,6 @@ __global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
       __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d], v | OS_AGG,
                          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
   }
-  /* lookback: thread d resolves digit d's exclusive prefix */
-  if (threadIdx.x < RADIX) {
-    uint32_t d = threadIdx.x;
-    uint32_t e = 0;
-    if (tile > 0) {
-      int64_t p = (int64_t)tile - 1;
-      uint32_t spins = 0;
-      while (p >= 0) {
-        uint32_t v = __hip_atomic_load((os_gu32*)&status[(uint64_t)p * RADIX + d],
-                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        if (v & OS_INC) { e += v & OS_CNT_MASK; break; }
-        if (v & OS_AGG) { e += v & OS_CNT_MASK; p--; continue; }
-        if (++spins > 100000000u) { atomicAdd(error, 1u); break; }
-        __builtin_amdgcn_s_sleep(4);
-      }
-      __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d],
-                         ((e + tilecnt[d]) & OS_CNT_MASK) | OS_INC,
-                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    }
-    excl[d] = e;
-    /* exclusive scan of the tile's own counts for the LDS grouping */
-  }
   /* tile-local digit offsets */
   {
     int t = threadIdx.x;
@@ -663,6 +641,28 @@ __global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
       if (HAS_B64) ls_b64[slot] = my_b64[round];
       ls_dig[slot] = (uint8_t)my_dig[round];
     }
+  }
+  /* lookback overlapped after placement: thread d resolves digit d's
+     exclusive prefix while the other waves finished their LDS stores */
+  if (threadIdx.x < RADIX) {
+    uint32_t d = threadIdx.x;
+    uint32_t e = 0;
+    if (tile > 0) {
+      int64_t p = (int64_t)tile - 1;
+      uint32_t spins = 0;
+      while (p >= 0) {
+        uint32_t v = __hip_atomic_load((os_gu32*)&status[(uint64_t)p * RADIX + d],
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (v & OS_INC) { e += v & OS_CNT_MASK; break; }
+        if (v & OS_AGG) { e += v & OS_CNT_MASK; p--; continue; }
+        if (++spins > 100000000u) { atomicAdd(error, 1u); break; }
+        __builtin_amdgcn_s_sleep(4);
+      }
+      __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d],
+                         ((e + tilecnt[d]) & OS_CNT_MASK) | OS_INC,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    }
+    excl[d] = e;
   }
   __syncthreads();
   for (uint32_t j = threadIdx.x; j < count; j += blockDim.x) {
@@ -1584,6 +1584,7 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     if (gbases.alloc(4u * npasses * RADIX)) return -12;
     if (st.alloc(4ull * nb * RADIX)) return -12;
     if (tick.alloc(16)) return -12;
+    HIP_CHECK(hipMemsetAsync(tick.p, 0, 16));  /* err word cleared once */
     HIP_CHECK(hipMemsetAsync(gh.p, 0, 4u * npasses * RADIX));
     hipLaunchKernelGGL(k_global_hist_all, dim3(grid1d(n)), dim3(BLOCK),
                        (uint32_t)(4 * npasses * RADIX), 0, (const uint64_t*)kin, n,
@@ -1603,7 +1604,9 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     for (int b = first_byte; b < nbytes_key; b++) {
       int p = b - first_byte;
       HIP_CHECK(hipMemsetAsync(st.p, 0, 4ull * nb * RADIX));
-      HIP_CHECK(hipMemsetAsync(tick.p, 0, 16));
+      /* reset only the ticket word — the error word ([1]) accumulates across
+         passes and is checked once after the loop */
+      HIP_CHECK(hipMemsetAsync(tick.p, 0, 4));
       if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                       (void)hipEventRecord(evs[nev]); }
       hipLaunchKernelGGL((k_onesweep_pass<uint64_t, false>), dim3(nb), dim3(BLOCK), 0, 0,
@@ -1612,16 +1615,20 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                          (const uint32_t*)((uint32_t*)gbases.p + p * RADIX),
                          (uint32_t*)st.p, (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
       if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
-      uint32_t h_err = 0;
-      HIP_CHECK(hipMemcpy(&h_err, (uint32_t*)tick.p + 1, 4, hipMemcpyDeviceToHost));
-      if (h_err) {
-        /* lookback timed out (should not happen): classic pass redo */
-        snprintf(g_err, sizeof(g_err), "onesweep lookback timeout pass %d", b);
-        return -70;
-      }
       std::swap(kin, kout);
       std::swap(a0in, a0out);
       passes++;
+    }
+    {
+      /* one deferred timeout check for all passes (a per-pass sync read cost
+         ~0.1 ms each; a timeout is a should-never-happen protocol failure
+         and poisons every later pass anyway) */
+      uint32_t h_err = 0;
+      HIP_CHECK(hipMemcpy(&h_err, (uint32_t*)tick.p + 1, 4, hipMemcpyDeviceToHost));
+      if (h_err) {
+        snprintf(g_err, sizeof(g_err), "onesweep lookback timeout");
+        return -70;
+      }
     }
     goto finish;
   }
