@@ -36,18 +36,21 @@ def run_step_c3(tez_amd, gen_batches, free_inputs=False):
     """C3: forced multi-spill + k-way merge (BASELINE configs[2]).  Each
     batch is absorbed and spilled (sorted + IFile-emitted), then flush runs
     the 32-way merge into the final output.  At the full 1e9-record size
-    (~92 GB payload) spill streams are transient and inputs are freed after
-    the absorb copy so everything fits the 288 GB HBM."""
+    (~92 GB payload) the sorter ADOPTS each batch (zero-copy absorb: one
+    materialization of the records total, like the reference's collect
+    serializing into the sort buffer) and spill streams are transient so
+    everything fits the 288 GB HBM."""
     t0 = time.perf_counter()
     conf = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
                              comparator=tez_amd.CMP_TEXT,
                              discard_spill_streams=1 if free_inputs else 0)
     s = tez_amd.Sorter(conf)
     for d, off, kl, n in gen_batches:
-        s.write_batch_device(d, off, kl, None, n)
-        s.spill()
         if free_inputs:
-            tez_amd.free_device(d, off, kl)
+            s.write_batch_device_adopt(d, off, kl, None, n)  # consumes inputs
+        else:
+            s.write_batch_device(d, off, kl, None, n)
+        s.spill()
     t1 = time.perf_counter()
     s.flush()
     t2 = time.perf_counter()
@@ -81,11 +84,14 @@ def run_step_c5(tez_amd, d, off, kl, part, n):
     return ctr, tms
 
 
-def run_step_single(tez_amd, conf, d, off, kl, n):
+def run_step_single(tez_amd, conf, d, off, kl, n, adopt=False):
     t0 = time.perf_counter()
     s = tez_amd.Sorter(conf)
     t1 = time.perf_counter()
-    s.write_batch_device(d, off, kl, None, n)
+    if adopt:
+        s.write_batch_device_adopt(d, off, kl, None, n)
+    else:
+        s.write_batch_device(d, off, kl, None, n)
     t2 = time.perf_counter()
     s.flush()
     t3 = time.perf_counter()
@@ -100,11 +106,14 @@ def run_step_single(tez_amd, conf, d, off, kl, n):
     return ctr, tms
 
 
-def run_step_multi(tez_amd, rank, world, device, d, off, kl, n):
+def run_step_multi(tez_amd, rank, world, device, d, off, kl, n, adopt=False):
     from tez_amd import exchange as ex
     conf = tez_amd.make_conf(PARTS, world_size=world, rank=rank)
     m = tez_amd.Sorter(conf)
-    m.write_batch_device(d, off, kl, None, n)
+    if adopt:
+        m.write_batch_device_adopt(d, off, kl, None, n)
+    else:
+        m.write_batch_device(d, off, kl, None, n)
     m.flush()
     ctr = m.counters()
     d_data, d_off2, d_klen2, rec_ranges, byte_ranges = m.sorted_columnar()
@@ -170,6 +179,10 @@ def main():
                          "partitions (single-GPU slice of configs[4])")
     ap.add_argument("--spills", type=int, default=32)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--no-adopt", action="store_true",
+                    help="force the copying absorb path (default: the sorter "
+                         "adopts per-step pre-generated inputs zero-copy when "
+                         "(warmup+steps) x payload fits comfortably in HBM)")
     ap.add_argument("--force-exchange", action="store_true",
                     help="run the all-to-all-v exchange + reduce-merge path even "
                          "at world_size=1 (bench-code validation)")
@@ -207,6 +220,7 @@ def main():
     n_local = args.records // n_gpus
     conf = tez_amd.make_conf(PARTS)
     c5_part = None
+    adopt = False
     if args.workload == "c5":
         assert n_gpus == 1, "c5 bench line is the single-GPU slice"
         c5conf = tez_amd.make_conf(128)
@@ -227,9 +241,26 @@ def main():
             tez_amd.free_device(part)
             gen_batches.append((d, off, kl, per))
     else:
-        d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local, kind=0,
-                                            klen=KLEN, vlen=VLEN, conf=conf)
-        tez_amd.free_device(part)
+        # C2: one input set per step so the sorter can ADOPT it zero-copy
+        # (one materialization of the records, like the reference's collect
+        # serializing into its sort buffer).  Falls back to one shared input
+        # + copying absorb when the pre-generated sets would not fit HBM.
+        total_sets = args.warmup + args.steps
+        adopt = (not args.no_adopt
+                 and total_sets * n_local * REC_SER <= 120e9)
+        if adopt:
+            input_sets = []
+            for sidx in range(total_sets):
+                d, off, kl, part = tez_amd.generate(
+                    seed=SEED + rank + 7919 * sidx, n=n_local, kind=0,
+                    klen=KLEN, vlen=VLEN, conf=conf)
+                tez_amd.free_device(part)
+                input_sets.append((d, off, kl))
+            set_cursor = [0]
+        else:
+            d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local, kind=0,
+                                                klen=KLEN, vlen=VLEN, conf=conf)
+            tez_amd.free_device(part)
 
     def barrier_sync():
         if dist:
@@ -258,6 +289,13 @@ def main():
             return run_step_c5(tez_amd, d, off, kl, c5_part, args.records)
         if args.workload == "c3":
             return run_step_c3(tez_amd, gen_batches, c3_free_inputs)
+        if adopt:
+            sd, so, sk = input_sets[set_cursor[0]]
+            set_cursor[0] += 1
+            if use_exchange:
+                return run_step_multi(tez_amd, rank, world, device, sd, so, sk,
+                                      n_local, adopt=True)
+            return run_step_single(tez_amd, conf, sd, so, sk, n_local, adopt=True)
         if use_exchange:
             return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
         return run_step_single(tez_amd, conf, d, off, kl, n_local)
@@ -289,8 +327,9 @@ def main():
                 tez_amd.free_device(d, off, kl)
     elif args.workload == "c5":
         tez_amd.free_device(d, off, kl, c5_part)
-    else:
+    elif not adopt:
         tez_amd.free_device(d, off, kl)
+    # adopt mode: every input set was consumed (ownership moved to the sorter)
 
     total_bytes_per_step = last_ctr["output_bytes"] * n_gpus  # whole-job Σ
     ms_per_step = elapsed / args.steps * 1e3

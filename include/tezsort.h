@@ -98,6 +98,16 @@ int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
                                   const uint64_t* d_off, const uint32_t* d_klen,
                                   const int32_t* d_part, int64_t n);
 
+/* Zero-copy variant: the sorter takes OWNERSHIP of buffers previously
+ * allocated by tzs_malloc_device / tzs_generate (the caller must not use or
+ * free them afterwards).  Valid only as the first batch of a spill.  The
+ * reference's collect() copies because input arrives record-at-a-time
+ * (PipelinedSorter.java:399-467); device-resident producers hand whole
+ * buffers over instead. */
+int tzs_sorter_write_batch_device_adopt(tzs_sorter* s, void* d_data,
+                                        uint64_t* d_off, uint32_t* d_klen,
+                                        int32_t* d_part, int64_t n);
+
 /* Force a spill of everything absorbed since the last spill
  * (PipelinedSorter.spill, PipelinedSorter.java:559-648). Returns spill id >= 0. */
 int tzs_sorter_spill(tzs_sorter* s);

@@ -218,6 +218,47 @@ def test_device_generator_matches_oracle_pipeline(engine):
     assert got == want["data"]
 
 
+def test_parity_adopted_batch(engine):
+    """Zero-copy absorb (tzs_sorter_write_batch_device_adopt): the sorter
+    takes ownership of tzs-allocated device buffers; output must byte-equal
+    the copying path / oracle.  Also checks the guard: adopting into a
+    non-empty sorter must fail."""
+    import numpy as np
+    pairs = _mk_fixed(4000, 16, 64, seed=59)
+    data, offs, klens = o.build_records(pairs)
+    d, doff, dkl, _ = engine.upload_records(bytes(data), offs, klens)
+    conf = engine.make_conf(64)
+    s = engine.Sorter(conf)
+    s.write_batch_device_adopt(d, doff, dkl, None, len(pairs))
+    s.flush()
+    got, gidx = s.output()
+    s.close()
+    want = _oracle_single_spill(pairs, 64, o.KEY_BYTES, o.CMP_TEZBYTES)
+    assert gidx == o.index_decode(want["index"], 64)
+    assert got == want["data"]
+
+    # guard: second adopt into a non-empty sorter fails loudly
+    d2, doff2, dkl2, _ = engine.upload_records(bytes(data), offs, klens)
+    s2 = engine.Sorter(engine.make_conf(8))
+    s2.write_batch_device(d2, doff2, dkl2, None, len(pairs))
+    try:
+        s2.write_batch_device_adopt(d2, doff2, dkl2, None, len(pairs))
+        assert False, "adopt into non-empty sorter should fail"
+    except RuntimeError:
+        pass
+    s2.close()
+    engine.free_device(d2, doff2, dkl2)
+
+    # guard: foreign (non-registry) pointers are rejected
+    s3 = engine.Sorter(engine.make_conf(8))
+    try:
+        s3.write_batch_device_adopt(12345678, 2345678, 345678, None, 10)
+        assert False, "foreign pointers should be rejected"
+    except RuntimeError:
+        pass
+    s3.close()
+
+
 def test_write_files_reference_layout(engine, tmp_path):
     pairs = _mk_fixed(100, 16, 16, seed=23)
     conf = engine.make_conf(4)

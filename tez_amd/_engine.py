@@ -90,6 +90,9 @@ def lib():
                                    c.c_int32, c.c_int32]
     L.tzs_sorter_write_batch_device.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p,
                                                 c.c_void_p, c.c_void_p, c.c_int64]
+    L.tzs_sorter_write_batch_device_adopt.argtypes = [c.c_void_p, c.c_void_p,
+                                                      c.c_void_p, c.c_void_p,
+                                                      c.c_void_p, c.c_int64]
     L.tzs_sorter_spill.argtypes = [c.c_void_p]
     L.tzs_sorter_flush.argtypes = [c.c_void_p]
     L.tzs_sorter_num_spills.argtypes = [c.c_void_p]
@@ -158,6 +161,12 @@ class Sorter:
     def write_batch_device(self, d_data, d_off, d_klen, d_part, n):
         _ck(lib().tzs_sorter_write_batch_device(
             self.h, d_data, d_off, d_klen, d_part, n), "write_batch")
+
+    def write_batch_device_adopt(self, d_data, d_off, d_klen, d_part, n):
+        """Zero-copy absorb: sorter takes ownership of tzs-allocated device
+        buffers (first batch of a spill only); caller must not free them."""
+        _ck(lib().tzs_sorter_write_batch_device_adopt(
+            self.h, d_data, d_off, d_klen, d_part, n), "write_batch_adopt")
 
     def spill(self):
         return _ck(lib().tzs_sorter_spill(self.h), "spill")

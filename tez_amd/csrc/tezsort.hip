@@ -1908,6 +1908,69 @@ extern "C" int tzs_sorter_write_batch_device(tzs_sorter* s, const void* d_data,
   return 0;
 }
 
+extern "C" int tzs_sorter_write_batch_device_adopt(tzs_sorter* s, void* d_data,
+                                                   uint64_t* d_off, uint32_t* d_klen,
+                                                   int32_t* d_part, int64_t n) {
+  /* Zero-copy absorb: the sorter takes OWNERSHIP of device buffers that were
+   * allocated through this library (tzs_malloc_device / tzs_generate).  The
+   * reference's collect() copies because its input arrives record-at-a-time
+   * from the processor (PipelinedSorter.java:399-467); a device-resident
+   * producer handing over whole buffers is the MI355X-native equivalent and
+   * skips the 2x-payload D2D copy.  Only valid as the FIRST batch of a spill
+   * and only for registry-backed pointers; the caller must not touch or free
+   * the buffers afterwards. */
+  auto t0 = std::chrono::steady_clock::now();
+  if (n == 0) return 0;
+  if (n > 4000000000ll) FAIL(-22, "batch too large (u32 record ids)");
+  if (s->cur_n != 0 || !s->host_klen.empty())
+    FAIL(-22, "adopt requires an empty current buffer (first batch of spill)");
+  auto& reg = pool_registry();
+  auto id = reg.find(d_data), io = reg.find(d_off), ik = reg.find(d_klen);
+  auto ip = d_part ? reg.find(d_part) : reg.end();
+  if (id == reg.end() || io == reg.end() || ik == reg.end() ||
+      (d_part && ip == reg.end()))
+    FAIL(-22, "adopt requires buffers allocated by tzs_malloc_device/tzs_generate");
+  uint64_t nbytes = 0, first = 0;
+  HIP_CHECK(hipMemcpy(&first, d_off, 8, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipMemcpy(&nbytes, d_off + n, 8, hipMemcpyDeviceToHost));
+  if (first != 0) FAIL(-22, "d_off must start at 0");
+  /* uniform-record detection (same rule as the copy path) */
+  {
+    static thread_local DBuf mm;
+    if (mm.alloc(32)) return -12;
+    uint64_t init[4] = {~0ull, 0, ~0ull, 0};
+    HIP_CHECK(hipMemcpyAsync(mm.p, init, 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_check_uniform, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                       d_off, d_klen, n, (uint64_t*)mm.p);
+    uint64_t res[4];
+    HIP_CHECK(hipMemcpy(res, mm.p, 32, hipMemcpyDeviceToHost));
+    bool uni = (res[0] == res[1]) && (res[2] == res[3]) && res[0] <= 0xFFFFFFFFull;
+    if (getenv("TZS_NO_UNIFORM")) uni = false;
+    s->cur_rec_u = uni ? (uint32_t)res[0] : 0;
+    s->cur_klen_u = uni ? (uint32_t)res[2] : 0;
+    s->cur_first_batch = false;
+  }
+  /* take ownership: registry entry -> DBuf (released to the pool later) */
+  s->cur_data.release(); s->cur_data.p = d_data; s->cur_data.sz = id->second; reg.erase(id);
+  s->cur_off.release();  s->cur_off.p = d_off;   s->cur_off.sz = io->second;  reg.erase(io);
+  s->cur_klen.release(); s->cur_klen.p = d_klen; s->cur_klen.sz = ik->second; reg.erase(ik);
+  if (d_part) {
+    s->cur_part.release(); s->cur_part.p = d_part; s->cur_part.sz = ip->second; reg.erase(ip);
+    s->have_explicit_parts = true;
+  } else {
+    if (s->cur_part.alloc(sizeof(int32_t) * n)) return -12;
+    s->have_explicit_parts = false;
+  }
+  s->cur_n = n;
+  s->cur_bytes = nbytes;
+  s->ctr.output_records += n;
+  s->ctr.output_bytes += (int64_t)nbytes;
+  HIP_CHECK(hipDeviceSynchronize());
+  s->times.absorb_ns += std::chrono::duration_cast<std::chrono::nanoseconds>(
+      std::chrono::steady_clock::now() - t0).count();
+  return 0;
+}
+
 __global__ void k_shift_offsets(const uint64_t* src, uint64_t* dst, uint64_t shift,
                                 int64_t n) {
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
