@@ -1073,6 +1073,11 @@ __global__ void k_permute_records(RecTable rt, const uint32_t* sidx,
 __constant__ uint32_t c_crc_table[256];
 __constant__ uint32_t c_crc_table8[8][256]; /* slice-by-8: T0 = standard */
 __constant__ uint32_t c_crc_mats[CRC_MATS][32];
+/* byte-sliced SHIFT operators (multiply by x^(8*64) / x^(8*256)):
+ * shiftN(crc) = T[0][crc&FF] ^ T[1][crc>>8 &FF] ^ T[2][..] ^ T[3][crc>>24]
+ * — 4 table lookups instead of a 32-step serial GF(2) matrix apply. */
+__constant__ uint32_t c_crc_t64[4][256];
+__constant__ uint32_t c_crc_t256[4][256];
 
 __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
   for (int k = 0; nbytes; k++, nbytes >>= 1)
@@ -1099,9 +1104,13 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     const uint64_t* chunk_base, const uint64_t* sc_base /* [P+1] */,
     uint32_t nparts, uint32_t total_sc, uint32_t* chunk_crc) {
   __shared__ uint32_t tab8[8][256];
+  __shared__ uint32_t t64[4][256];
   __shared__ uint32_t stage32[CRC_SC_CHUNKS * CRC_ROW_W];
+  __shared__ uint32_t qcrc[BLOCK];
   for (int i = threadIdx.x; i < 2048; i += blockDim.x)
     ((uint32_t*)tab8)[i] = ((const uint32_t*)c_crc_table8)[i];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x)
+    ((uint32_t*)t64)[i] = ((const uint32_t*)c_crc_t64)[i];
   for (uint32_t sc = blockIdx.x; sc < total_sc; sc += gridDim.x) {
     uint32_t lo = 0, hi = nparts;
     while (lo + 1 < hi) {
@@ -1124,13 +1133,32 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
       stage32[(wi >> 6) * CRC_ROW_W + (wi & 63)] = w;
     }
     __syncthreads();
+    /* 4 threads per 256B chunk, 64B quarter each (all 256 threads compute;
+     * the dependent slice-by-8 chain drops from 32 to 8 steps); quarter
+     * results merge with the byte-sliced shift-by-64 operator. */
     uint32_t t = threadIdx.x;
-    uint64_t c0 = (uint64_t)t * CRC_CHUNK;
-    if (t < CRC_SC_CHUNKS && c0 < avail) {
-      uint64_t len = avail - c0;
-      if (len > CRC_CHUNK) len = CRC_CHUNK;
+    uint32_t chunk = t >> 2, quarter = t & 3;
+    uint64_t c0 = (uint64_t)chunk * CRC_CHUNK;
+    bool live = c0 < avail;
+    uint64_t len = live ? min(avail - c0, (uint64_t)CRC_CHUNK) : 0;
+    const uint32_t* row = stage32 + chunk * CRC_ROW_W;
+    if (live && len == CRC_CHUNK) {
+      uint32_t crc = (quarter == 0) ? 0xFFFFFFFFu : 0u;
+      const uint32_t* qrow = row + quarter * 16;
+      #pragma unroll
+      for (uint32_t w = 0; w < 8; w++) {
+        uint32_t lo = crc ^ qrow[2 * w];
+        uint32_t hi = qrow[2 * w + 1];
+        crc = tab8[7][lo & 0xFF] ^ tab8[6][(lo >> 8) & 0xFF]
+            ^ tab8[5][(lo >> 16) & 0xFF] ^ tab8[4][lo >> 24]
+            ^ tab8[3][hi & 0xFF] ^ tab8[2][(hi >> 8) & 0xFF]
+            ^ tab8[1][(hi >> 16) & 0xFF] ^ tab8[0][hi >> 24];
+      }
+      qcrc[t] = crc;
+    } else if (live && quarter == 0) {
+      /* ragged tail chunk (only the last chunk of a partition): classic
+       * slice-by-8 + bytewise tail on one thread */
       uint32_t crc = 0xFFFFFFFFu;
-      const uint32_t* row = stage32 + t * CRC_ROW_W;
       uint32_t full8 = (uint32_t)(len / 8);
       for (uint32_t w = 0; w < full8; w++) {
         uint32_t lo = crc ^ row[2 * w];
@@ -1144,7 +1172,20 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
         uint8_t byte = (uint8_t)(row[b / 4] >> (8 * (b & 3)));
         crc = (crc >> 8) ^ tab8[0][(crc ^ byte) & 0xFF];
       }
-      chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + t] = crc ^ 0xFFFFFFFFu;
+      qcrc[t] = crc;
+    }
+    __syncthreads();
+    if (live && quarter == 0) {
+      uint32_t crc = qcrc[t];
+      if (len == CRC_CHUNK) {
+        #pragma unroll
+        for (uint32_t q = 1; q < 4; q++) {
+          crc = t64[0][crc & 0xFF] ^ t64[1][(crc >> 8) & 0xFF]
+              ^ t64[2][(crc >> 16) & 0xFF] ^ t64[3][crc >> 24];
+          crc ^= qcrc[t + q];
+        }
+      }
+      chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + chunk] = crc ^ 0xFFFFFFFFu;
     }
   }
 }
@@ -1161,6 +1202,10 @@ __global__ void k_crc_combine_groups(const uint64_t* range_len, const uint64_t* 
                                      uint32_t* group_crc, uint64_t* group_len) {
   __shared__ uint32_t s_crc[BLOCK];
   __shared__ uint64_t s_len[BLOCK];
+  __shared__ uint32_t t256[4][256];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x)
+    ((uint32_t*)t256)[i] = ((const uint32_t*)c_crc_t256)[i];
+  __syncthreads();
   for (uint32_t g = blockIdx.x; g < total_groups; g += gridDim.x) {
     /* find partition via binary search over group_base */
     uint32_t lo = 0, hi = nparts;
@@ -1181,7 +1226,12 @@ __global__ void k_crc_combine_groups(const uint64_t* range_len, const uint64_t* 
     for (uint64_t c = t0; c < min(t0 + 8, cend); c++) {
       uint64_t clen = len - c * CRC_CHUNK;
       if (clen > CRC_CHUNK) clen = CRC_CHUNK;
-      crc = d_crc_shift(crc, clen) ^ chunk_crc[chunk_base[p] + c];
+      if (clen == CRC_CHUNK)
+        crc = t256[0][crc & 0xFF] ^ t256[1][(crc >> 8) & 0xFF]
+            ^ t256[2][(crc >> 16) & 0xFF] ^ t256[3][crc >> 24]
+            ^ chunk_crc[chunk_base[p] + c];
+      else
+        crc = d_crc_shift(crc, clen) ^ chunk_crc[chunk_base[p] + c];
       mylen += clen;
     }
     s_crc[threadIdx.x] = crc;
@@ -1207,14 +1257,29 @@ __global__ void k_crc_combine_groups(const uint64_t* range_len, const uint64_t* 
 __global__ void k_crc_combine_final(const uint64_t* group_base, const uint32_t* group_crc,
                                     const uint64_t* group_len, uint32_t nparts,
                                     uint32_t* part_crc) {
-  uint32_t p = blockIdx.x * blockDim.x + threadIdx.x;
-  if (p >= nparts) return;
+  /* one WAVE per partition; each GF(2) matrix application is lane-parallel:
+   * lane i contributes mats[k][i] when bit i of the crc is set, then a
+   * butterfly XOR-reduces across the wave (6 steps instead of a 32-step
+   * serial fold per shift). */
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  if (wave >= nparts) return;
+  uint32_t p = wave;
   uint32_t crc = 0;
   for (uint64_t g = group_base[p]; g < group_base[p + 1]; g++) {
-    if (group_len[g])
-      crc = d_crc_shift(crc, group_len[g]) ^ group_crc[g];
+    uint64_t nb = group_len[g];
+    if (!nb) continue;
+    for (int k = 0; nb; k++, nb >>= 1) {
+      if (nb & 1) {
+        uint32_t contrib = (lane < 32 && ((crc >> lane) & 1)) ? c_crc_mats[k][lane] : 0;
+        #pragma unroll
+        for (int s = 32; s >= 1; s >>= 1) contrib ^= __shfl_xor(contrib, s, WAVE);
+        crc = contrib;
+      }
+    }
+    crc ^= group_crc[g];
   }
-  part_crc[p] = crc;
+  if (lane == 0) part_crc[p] = crc;
 }
 
 /* write headers, EOF trailers and CRC for each partition segment */
@@ -1373,6 +1438,18 @@ static int ensure_device_constants() {
     HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_table8), t8, sizeof(t8)));
   }
   HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_mats), h_crc_shift_mat, sizeof(h_crc_shift_mat)));
+  {
+    /* byte-sliced shift-by-64 / shift-by-256 operators:
+     * T[j][b] = (x^(8*N)) * (b << 8j) mod P — so shiftN(crc) is 4 lookups */
+    static uint32_t t64[4][256], t256[4][256];
+    for (int j = 0; j < 4; j++)
+      for (int b = 0; b < 256; b++) {
+        t64[j][b] = gf2_times(h_crc_shift_mat[6], (uint32_t)b << (8 * j));
+        t256[j][b] = gf2_times(h_crc_shift_mat[8], (uint32_t)b << (8 * j));
+      }
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_t64), t64, sizeof(t64)));
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_crc_t256), t256, sizeof(t256)));
+  }
   done = true;
   return 0;
 }
@@ -2387,7 +2464,8 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                        (const uint64_t*)d_groupbase.p, (const uint32_t*)d_chunkcrc.p, P,
                        (uint32_t)total_groups, (uint32_t*)d_groupcrc.p,
                        (uint64_t*)d_grouplen.p);
-  hipLaunchKernelGGL(k_crc_combine_final, dim3((P + BLOCK - 1) / BLOCK), dim3(BLOCK), 0, 0,
+  hipLaunchKernelGGL(k_crc_combine_final, dim3((P * WAVE + BLOCK - 1) / BLOCK),
+                     dim3(BLOCK), 0, 0,
                      (const uint64_t*)d_groupbase.p, (const uint32_t*)d_groupcrc.p,
                      (const uint64_t*)d_grouplen.p, P, (uint32_t*)d_partcrc.p);
   /* re-patch to write CRC trailers (header/EOF rewrite is idempotent) */
