@@ -1039,6 +1039,54 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
   }
 }
 
+/* Uniform-record emit fast path: when every record serializes to the same
+ * length and RLE is off, the IFile body is a constant-stride stream
+ * (hdr vints ‖ key ‖ val per record).  One LANE per record: the 88-byte
+ * C2 record moves as eleven u64 loads and ~12 u64 stores (head/tail bytes
+ * for output alignment) instead of the generic path's byte-granular
+ * half-wave gather — ~20x fewer issue slots, dense aggregate writes. */
+template <typename WordT, int MAXW>
+__global__ __launch_bounds__(BLOCK) void k_emit_uniform(
+    const RecDesc* desc, const uint64_t* scan, const uint32_t* parts,
+    const uint64_t* seg_payload_start, const uint64_t* part_scan_base,
+    uint8_t* out, uint32_t n, uint32_t hdrlen /* <= 8 */, uint64_t hdrword,
+    uint32_t reclen /* = klen+vlen; % sizeof(WordT) == 0; <= MAXW words */) {
+  constexpr uint32_t WB = (uint32_t)sizeof(WordT);
+  const uint32_t total = hdrlen + reclen;
+  const uint32_t nwords = reclen / WB;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    uint32_t p = parts[i];
+    uint64_t dst = seg_payload_start[p] + (scan[i] - part_scan_base[p]);
+    const WordT* srcw = (const WordT*)(uintptr_t)desc[i].src;
+    WordT s[MAXW + 1];
+    #pragma unroll
+    for (uint32_t k = 0; k < MAXW + 1; k++) s[k] = (k < nwords) ? srcw[k] : 0;
+    /* q0: first logical offset >= hdrlen that is WB-aligned in OUTPUT */
+    uint32_t a = (uint32_t)(dst % WB);
+    uint32_t q0 = hdrlen + ((WB - ((a + hdrlen) % WB)) % WB);
+    for (uint32_t q = 0; q < q0 && q < total; q++) {
+      uint8_t b = (q < hdrlen) ? (uint8_t)(hdrword >> (8 * q))
+                               : (uint8_t)(s[0] >> (8 * (q - hdrlen)));
+      out[dst + q] = b;
+    }
+    /* aligned word body: word k covers logical [q0 + WB*k, +WB) */
+    uint32_t c = (q0 - hdrlen) % WB;
+    uint32_t kmax = (total >= q0 + WB) ? (total - q0) / WB : 0;
+    WordT* ww = (WordT*)(out + dst + q0);
+    if (c == 0) {
+      for (uint32_t k = 0; k < kmax; k++) ww[k] = s[k];
+    } else {
+      for (uint32_t k = 0; k < kmax; k++)
+        ww[k] = (WordT)((s[k] >> (8 * c)) | (s[k + 1] << (8 * (WB - c))));
+    }
+    for (uint32_t q = q0 + WB * kmax; q < total; q++) {
+      uint32_t o = q - hdrlen;
+      out[dst + q] = (uint8_t)(s[o / WB] >> (8 * (o % WB)));
+    }
+  }
+}
+
 /* permuted-columnar materialization (exchange wire — DESIGN.md §4):
  * records gathered into sorted order as (data, off, klen) so partition
  * ranges are contiguous per destination rank. */
@@ -1097,20 +1145,28 @@ __device__ __forceinline__ uint32_t d_crc_shift(uint32_t crc, uint64_t nbytes) {
  * dependent-chain length of bytewise).  Rows padded one word so lane t's
  * reads land on distinct banks. */
 #define CRC_SC_CHUNKS 64
-#define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 32 KiB */
-#define CRC_ROW_W (CRC_CHUNK / 4 + 1)            /* 65 words per chunk row */
+#define CRC_SC_BYTES (CRC_SC_CHUNKS * CRC_CHUNK) /* 16 KiB per super-chunk */
+/* Register-direct chunk CRC: 4 threads per 256B chunk, one 64B quarter each.
+ * Each lane loads its quarter straight from the stream into 17 registers
+ * (4-aligned u32 + funnel for the range's byte phase; the word after a range
+ * always lands inside that range's 4B CRC trailer, so the +1 read is in
+ * bounds), runs an 8-step slice-by-8 chain from registers, and the three
+ * shuffle-gathered quarter CRCs merge with the byte-sliced shift-by-64
+ * operator.  No data staging, no per-iteration barriers — the old LDS-staged
+ * version was 84% wave-parked (PMC, profiles/) on the load->sync->compute
+ * serialization. */
 __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     const uint8_t* stream, const uint64_t* range_start, const uint64_t* range_len,
     const uint64_t* chunk_base, const uint64_t* sc_base /* [P+1] */,
     uint32_t nparts, uint32_t total_sc, uint32_t* chunk_crc) {
   __shared__ uint32_t tab8[8][256];
   __shared__ uint32_t t64[4][256];
-  __shared__ uint32_t stage32[CRC_SC_CHUNKS * CRC_ROW_W];
-  __shared__ uint32_t qcrc[BLOCK];
   for (int i = threadIdx.x; i < 2048; i += blockDim.x)
     ((uint32_t*)tab8)[i] = ((const uint32_t*)c_crc_table8)[i];
   for (int i = threadIdx.x; i < 1024; i += blockDim.x)
     ((uint32_t*)t64)[i] = ((const uint32_t*)c_crc_t64)[i];
+  __syncthreads();
+  const uint32_t lane = threadIdx.x & (WAVE - 1);
   for (uint32_t sc = blockIdx.x; sc < total_sc; sc += gridDim.x) {
     uint32_t lo = 0, hi = nparts;
     while (lo + 1 < hi) {
@@ -1125,64 +1181,59 @@ __global__ __launch_bounds__(BLOCK) void k_crc_chunks(
     uint64_t gbase = range_start[p] + byte0;
     uint32_t a = (uint32_t)(gbase & 3);
     const uint32_t* wsrc = (const uint32_t*)(stream + gbase - a);
-    uint32_t nwords = (uint32_t)((avail + 3) / 4);
-    __syncthreads();
-    for (uint32_t wi = threadIdx.x; wi < nwords; wi += blockDim.x) {
-      uint32_t w = wsrc[wi];
-      if (a) w = (w >> (8 * a)) | (wsrc[wi + 1] << (32 - 8 * a));
-      stage32[(wi >> 6) * CRC_ROW_W + (wi & 63)] = w;
-    }
-    __syncthreads();
-    /* 4 threads per 256B chunk, 64B quarter each (all 256 threads compute;
-     * the dependent slice-by-8 chain drops from 32 to 8 steps); quarter
-     * results merge with the byte-sliced shift-by-64 operator. */
-    uint32_t t = threadIdx.x;
-    uint32_t chunk = t >> 2, quarter = t & 3;
+    uint32_t chunk = threadIdx.x >> 2, quarter = threadIdx.x & 3;
     uint64_t c0 = (uint64_t)chunk * CRC_CHUNK;
     bool live = c0 < avail;
     uint64_t len = live ? min(avail - c0, (uint64_t)CRC_CHUNK) : 0;
-    const uint32_t* row = stage32 + chunk * CRC_ROW_W;
+    uint32_t crc = 0;
     if (live && len == CRC_CHUNK) {
-      uint32_t crc = (quarter == 0) ? 0xFFFFFFFFu : 0u;
-      const uint32_t* qrow = row + quarter * 16;
+      uint32_t j0 = ((uint32_t)c0 >> 2) + quarter * 16;
+      uint32_t r[17];
       #pragma unroll
-      for (uint32_t w = 0; w < 8; w++) {
-        uint32_t lo = crc ^ qrow[2 * w];
-        uint32_t hi = qrow[2 * w + 1];
-        crc = tab8[7][lo & 0xFF] ^ tab8[6][(lo >> 8) & 0xFF]
-            ^ tab8[5][(lo >> 16) & 0xFF] ^ tab8[4][lo >> 24]
-            ^ tab8[3][hi & 0xFF] ^ tab8[2][(hi >> 8) & 0xFF]
-            ^ tab8[1][(hi >> 16) & 0xFF] ^ tab8[0][hi >> 24];
-      }
-      qcrc[t] = crc;
-    } else if (live && quarter == 0) {
-      /* ragged tail chunk (only the last chunk of a partition): classic
-       * slice-by-8 + bytewise tail on one thread */
-      uint32_t crc = 0xFFFFFFFFu;
-      uint32_t full8 = (uint32_t)(len / 8);
-      for (uint32_t w = 0; w < full8; w++) {
-        uint32_t lo = crc ^ row[2 * w];
-        uint32_t hi = row[2 * w + 1];
-        crc = tab8[7][lo & 0xFF] ^ tab8[6][(lo >> 8) & 0xFF]
-            ^ tab8[5][(lo >> 16) & 0xFF] ^ tab8[4][lo >> 24]
-            ^ tab8[3][hi & 0xFF] ^ tab8[2][(hi >> 8) & 0xFF]
-            ^ tab8[1][(hi >> 16) & 0xFF] ^ tab8[0][hi >> 24];
-      }
-      for (uint32_t b = full8 * 8; b < len; b++) {
-        uint8_t byte = (uint8_t)(row[b / 4] >> (8 * (b & 3)));
-        crc = (crc >> 8) ^ tab8[0][(crc ^ byte) & 0xFF];
-      }
-      qcrc[t] = crc;
-    }
-    __syncthreads();
-    if (live && quarter == 0) {
-      uint32_t crc = qcrc[t];
-      if (len == CRC_CHUNK) {
+      for (int k = 0; k < 17; k++) r[k] = wsrc[j0 + k];
+      crc = (quarter == 0) ? 0xFFFFFFFFu : 0u;
+      if (a == 0) {
         #pragma unroll
-        for (uint32_t q = 1; q < 4; q++) {
+        for (uint32_t w = 0; w < 8; w++) {
+          uint32_t lo32 = crc ^ r[2 * w];
+          uint32_t hi32 = r[2 * w + 1];
+          crc = tab8[7][lo32 & 0xFF] ^ tab8[6][(lo32 >> 8) & 0xFF]
+              ^ tab8[5][(lo32 >> 16) & 0xFF] ^ tab8[4][lo32 >> 24]
+              ^ tab8[3][hi32 & 0xFF] ^ tab8[2][(hi32 >> 8) & 0xFF]
+              ^ tab8[1][(hi32 >> 16) & 0xFF] ^ tab8[0][hi32 >> 24];
+        }
+      } else {
+        uint32_t sh = 8 * a, ish = 32 - sh;
+        #pragma unroll
+        for (uint32_t w = 0; w < 8; w++) {
+          uint32_t lo32 = crc ^ ((r[2 * w] >> sh) | (r[2 * w + 1] << ish));
+          uint32_t hi32 = (r[2 * w + 1] >> sh) | (r[2 * w + 2] << ish);
+          crc = tab8[7][lo32 & 0xFF] ^ tab8[6][(lo32 >> 8) & 0xFF]
+              ^ tab8[5][(lo32 >> 16) & 0xFF] ^ tab8[4][lo32 >> 24]
+              ^ tab8[3][hi32 & 0xFF] ^ tab8[2][(hi32 >> 8) & 0xFF]
+              ^ tab8[1][(hi32 >> 16) & 0xFF] ^ tab8[0][hi32 >> 24];
+        }
+      }
+    } else if (live && quarter == 0) {
+      /* ragged tail chunk (only the last chunk of a partition): bytewise */
+      const uint8_t* sp = stream + gbase + c0;
+      crc = 0xFFFFFFFFu;
+      for (uint64_t b = 0; b < len; b++)
+        crc = (crc >> 8) ^ tab8[0][(crc ^ sp[b]) & 0xFF];
+    }
+    /* shuffle-combine the four quarter CRCs (quarters are adjacent lanes) */
+    uint32_t qbase = lane & ~3u;
+    uint32_t v1 = __shfl(crc, qbase + 1);
+    uint32_t v2 = __shfl(crc, qbase + 2);
+    uint32_t v3 = __shfl(crc, qbase + 3);
+    if (live && quarter == 0) {
+      if (len == CRC_CHUNK) {
+        uint32_t vs[3] = {v1, v2, v3};
+        #pragma unroll
+        for (int q = 0; q < 3; q++) {
           crc = t64[0][crc & 0xFF] ^ t64[1][(crc >> 8) & 0xFF]
               ^ t64[2][(crc >> 16) & 0xFF] ^ t64[3][crc >> 24];
-          crc ^= qcrc[t + q];
+          crc ^= vs[q];
         }
       }
       chunk_crc[chunk_base[p] + sc_local * CRC_SC_CHUNKS + chunk] = crc ^ 0xFFFFFFFFu;
@@ -2412,11 +2463,67 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   /* 6. emit records */
   static int force_simple = -1;
   if (force_simple < 0) force_simple = getenv("TZS_EMIT_SIMPLE") ? 1 : 0;
-  hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
-                     (const RecDesc*)descbuf.p,
-                     (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
-                     (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
-                     (const uint64_t*)d_scanbase.p, d_out, n, force_simple);
+  /* k_emit_uniform (lane-per-record word moves) measured SLOWER than the
+   * half-wave path: each store instruction scatters 64 partial-line writes
+   * across 64 records and write coalescing collapses (37 ms vs 10 ms at C2).
+   * Kept behind TZS_EMIT_UNIFORM=1 for experiments; default off. */
+  static int no_uniform_emit = -1;
+  if (no_uniform_emit < 0) {
+    const char* e = getenv("TZS_EMIT_UNIFORM");
+    no_uniform_emit = (e && e[0] == '1') ? 0 : 1;
+  }
+  /* uniform fast path: constant record serialization + RLE provably off
+     (writer rle disabled AND either zero adjacent-equal keys or a single
+     non-RLE source spill) => constant-stride body, lane-per-record word
+     moves (k_emit_uniform) */
+  uint32_t uni_rec = 0, uni_klen = 0;
+  {
+    bool uni = rt.nspills >= 1 && !s->combine_applied;
+    for (int sp2 = 0; sp2 < rt.nspills && uni; sp2++) {
+      if (!rt.rec_u[sp2] || !rt.klen_u[sp2]) uni = false;
+      else if (sp2 == 0) { uni_rec = rt.rec_u[sp2]; uni_klen = rt.klen_u[sp2]; }
+      else if (uni_rec != rt.rec_u[sp2] || uni_klen != rt.klen_u[sp2]) uni = false;
+    }
+    if (!uni) uni_rec = 0;
+  }
+  bool no_same = (writer_rle == 0) &&
+      (neq_final == 0 || (nspills_rle == 1 && h_spill_rle && !h_spill_rle[0]));
+  uint32_t uh_len = 0;
+  uint64_t uh_word = 0;
+  if (uni_rec && no_same) {
+    uint8_t hb[12] = {0};
+    auto hvint = [](uint8_t* b, uint32_t v) -> uint32_t {
+      if (v <= 127) { b[0] = (uint8_t)v; return 1; }
+      uint32_t nb2 = 0, t = v;
+      while (t) { t >>= 8; nb2++; }
+      b[0] = (uint8_t)(-112 - (int)nb2);
+      for (uint32_t k = 0; k < nb2; k++) b[1 + k] = (uint8_t)(v >> (8 * (nb2 - 1 - k)));
+      return 1 + nb2;
+    };
+    uh_len = hvint(hb, uni_klen);
+    uh_len += hvint(hb + uh_len, uni_rec - uni_klen);
+    for (int b = 0; b < 8; b++) uh_word |= (uint64_t)hb[b] << (8 * b);
+    if (uh_len > 8) uni_rec = 0; /* header too long for the fast path */
+  }
+  if (!no_uniform_emit && !force_simple && uni_rec && no_same &&
+      uni_rec % 8 == 0 && uni_rec <= 16 * 8) {
+    hipLaunchKernelGGL((k_emit_uniform<uint64_t, 16>), dim3(grid1d(n)), dim3(BLOCK),
+                       0, 0, (const RecDesc*)descbuf.p, (const uint64_t*)s->scan.p,
+                       (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
+                       (const uint64_t*)d_scanbase.p, d_out, n, uh_len, uh_word, uni_rec);
+  } else if (!no_uniform_emit && !force_simple && uni_rec && no_same &&
+             uni_rec % 4 == 0 && uni_rec <= 32 * 4) {
+    hipLaunchKernelGGL((k_emit_uniform<uint32_t, 32>), dim3(grid1d(n)), dim3(BLOCK),
+                       0, 0, (const RecDesc*)descbuf.p, (const uint64_t*)s->scan.p,
+                       (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
+                       (const uint64_t*)d_scanbase.p, d_out, n, uh_len, uh_word, uni_rec);
+  } else {
+    hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
+                       (const RecDesc*)descbuf.p,
+                       (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
+                       (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
+                       (const uint64_t*)d_scanbase.p, d_out, n, force_simple);
+  }
   (void)hipEventRecord(ev[5]);
 
   /* 7. CRC: chunk bases per partition */
