@@ -27,7 +27,10 @@
 #define WPB (BLOCK / WAVE)
 #define RADIX 256
 #define TILE_ROUNDS 8
-#define TILE (TILE_ROUNDS * BLOCK) /* elements per scatter block */
+#define TILE (TILE_ROUNDS * BLOCK) /* elements per scatter block
+  (16 rounds = 4096/tile measured: sort 12.5 -> 13.7 ms at C2 — the
+   doubled LDS halves resident blocks and the occupancy loss outweighs
+   the longer digit runs) */
 
 /* ------------------------------------------------------------------ */
 /* error plumbing                                                      */
@@ -283,30 +286,67 @@ __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, i);
+    /* word-windowed access to the serialized key region: byte j is sliced
+       out of 4-aligned u32 loads (the <=3-byte overread stays inside the
+       +64 B pool pad; the compiler CSEs repeated window words).  The old
+       bytewise form issued ~27 one-byte loads per record. */
+    uintptr_t kA = (uintptr_t)v.key;
+    const uint32_t* kw = (const uint32_t*)(kA & ~(uintptr_t)3);
+    uint32_t ksh = (uint32_t)(kA & 3);
+    auto kbyte = [&](uint32_t j) -> uint32_t {
+      uint32_t q = ksh + j;
+      return (kw[q >> 2] >> (8 * (q & 3))) & 0xFFu;
+    };
+    uint32_t coff = (uint32_t)(v.content - v.key);
     uint32_t part;
-    if (d_part) part = (uint32_t)d_part[i];
-    else part = (uint32_t)((d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P);
+    if (d_part) {
+      part = (uint32_t)d_part[i];
+    } else {
+      /* HashPartitioner.hashBytes word-at-a-time: 4 Horner steps fold into
+         h = h*31^4 + b0*31^3 + b1*31^2 + b2*31 + b3 (mod 2^32, signed
+         bytes sign-extended — WritableComparator.hashBytes semantics) */
+      int32_t h = 0;
+      uint32_t j = 0;
+      while (j < v.clen && ((ksh + coff + j) & 3)) {
+        h = h * 31 + (int32_t)(int8_t)kbyte(coff + j);
+        j++;
+      }
+      const int32_t M2 = 961, M3 = 29791, M4 = 923521;
+      while (j + 4 <= v.clen) {
+        uint32_t w = kw[(ksh + coff + j) >> 2];
+        h = h * M4 + (int32_t)(int8_t)(w & 0xFF) * M3
+            + (int32_t)(int8_t)((w >> 8) & 0xFF) * M2
+            + (int32_t)(int8_t)((w >> 16) & 0xFF) * 31
+            + (int32_t)(int8_t)(w >> 24);
+        j += 4;
+      }
+      while (j < v.clen) {
+        h = h * 31 + (int32_t)(int8_t)kbyte(coff + j);
+        j++;
+      }
+      part = ((uint32_t)h & 0x7fffffff) % P;
+    }
     uint64_t key;
     if (ser_mode) {
       /* the reference prefix keeps only proxy >>> (bitcount(P)+1) bits
          (PipelinedSorter.java:457): ties on the TRUNCATED proxy fall through
          to the serialized compare, so the composite must truncate too */
-      uint32_t proxy = ((v.clen > 0 ? (uint32_t)v.content[0] : 0u) << 16) |
-                       ((v.clen > 1 ? (uint32_t)v.content[1] : 0u) << 8) |
-                       (v.clen > 2 ? (uint32_t)v.content[2] : 0u);
+      uint32_t proxy = ((v.clen > 0 ? kbyte(coff) : 0u) << 16) |
+                       ((v.clen > 1 ? kbyte(coff + 1) : 0u) << 8) |
+                       (v.clen > 2 ? kbyte(coff + 2) : 0u);
       int pw = 24 - ref_pb;            /* surviving proxy bits */
       if (pw < 0) pw = 0;
       uint32_t proxy_t = pw ? (proxy >> (24 - pw)) : 0;
       uint64_t ser = 0;
       uint32_t m = v.klen < 8 ? v.klen : 8;
-      for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)v.key[b] << (56 - 8 * b);
+      for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)kbyte(b) << (56 - 8 * b);
       key = ((uint64_t)part << (64 - pbits))
             | ((uint64_t)proxy_t << (64 - pbits - pw))
             | (ser >> (pbits + pw));
     } else {
       uint64_t c = 0;
       uint32_t m = v.clen < 8 ? v.clen : 8;
-      for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
+      for (uint32_t b = 0; b < m; b++) c |= (uint64_t)kbyte(coff + b) << (56 - 8 * b);
       key = pbits ? (((uint64_t)part << (64 - pbits)) | (c >> pbits)) : c;
     }
     d_key[i] = key & mask;
@@ -709,6 +749,77 @@ __global__ void k_scan_add(uint64_t* out, const uint64_t* block_offsets, uint32_
   uint64_t add = block_offsets[blockIdx.x];
   for (uint32_t i = base + threadIdx.x; i < min(base + SCAN_TILE, n); i += blockDim.x)
     out[i] += add;
+}
+
+/* single-pass u64 exclusive scan with decoupled lookback (ticketed tiles,
+ * same protocol as k_onesweep_pass): one read + one write of the data
+ * instead of the 3-kernel partials/scan/add chain.  Status word: u64 with
+ * flags in the top 2 bits (values here are byte totals < 2^40). */
+#define OSS_AGG (1ull << 62)
+#define OSS_INC (2ull << 62)
+#define OSS_VAL (OSS_AGG - 1)
+typedef unsigned long long __attribute__((address_space(1))) os_gu64;
+__global__ __launch_bounds__(BLOCK) void k_scan_lookback(
+    const uint64_t* in, uint64_t* out, uint32_t n,
+    uint64_t* status /* [ntiles] */, uint32_t* ticket, uint32_t* error,
+    uint64_t* total_out) {
+  __shared__ uint64_t lds[BLOCK];
+  __shared__ uint32_t s_tile;
+  __shared__ uint64_t s_excl;
+  if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
+  __syncthreads();
+  const uint32_t tile = s_tile;
+  const uint32_t base = tile * SCAN_TILE;
+  uint64_t vals[SCAN_ITEMS];
+  uint64_t sum = 0;
+  #pragma unroll
+  for (int j = 0; j < SCAN_ITEMS; j++) {
+    uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
+    vals[j] = (i < n) ? in[i] : 0;
+    sum += vals[j];
+  }
+  lds[threadIdx.x] = sum;
+  __syncthreads();
+  for (int s = 1; s < BLOCK; s <<= 1) {
+    uint64_t t = (threadIdx.x >= (uint32_t)s) ? lds[threadIdx.x - s] : 0;
+    __syncthreads();
+    lds[threadIdx.x] += t;
+    __syncthreads();
+  }
+  uint64_t block_total = lds[BLOCK - 1];
+  if (threadIdx.x == 0) {
+    if (tile == 0) {
+      __hip_atomic_store((os_gu64*)&status[0], OSS_INC | block_total,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      s_excl = 0;
+    } else {
+      __hip_atomic_store((os_gu64*)&status[tile], OSS_AGG | block_total,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      uint64_t e = 0;
+      int64_t p = (int64_t)tile - 1;
+      uint32_t spins = 0;
+      while (p >= 0) {
+        uint64_t v = __hip_atomic_load((os_gu64*)&status[p],
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (v & OSS_INC) { e += v & OSS_VAL; break; }
+        if (v & OSS_AGG) { e += v & OSS_VAL; p--; continue; }
+        if (++spins > 100000000u) { atomicAdd(error, 1u); break; }
+        __builtin_amdgcn_s_sleep(4);
+      }
+      __hip_atomic_store((os_gu64*)&status[tile], OSS_INC | (e + block_total),
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      s_excl = e;
+    }
+  }
+  __syncthreads();
+  uint64_t excl = s_excl + lds[threadIdx.x] - sum;
+  #pragma unroll
+  for (int j = 0; j < SCAN_ITEMS; j++) {
+    uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
+    if (i < n) { out[i] = excl; excl += vals[j]; }
+  }
+  if (total_out && threadIdx.x == 0 && (uint64_t)base + SCAN_TILE >= n)
+    *total_out = s_excl + block_total;
 }
 
 /* ---- refinement ---- */
@@ -1615,6 +1726,36 @@ struct DBuf {
 /* exclusive scan of u64 array (device), returns total via last+add trick */
 static int scan_u64(const uint64_t* d_in, uint64_t* d_out, uint32_t n, uint64_t* h_total) {
   if (n == 0) { if (h_total) *h_total = 0; return 0; }
+  {
+    /* single-pass decoupled-lookback path (one read + one write) */
+    static int no_oss = -1;
+    if (no_oss < 0) no_oss = getenv("TZS_NO_SCAN_LOOKBACK") ? 1 : 0;
+    uint32_t nb_l = nblocks_for(n, SCAN_TILE);
+    if (!no_oss && nb_l > 8) {
+      static thread_local DBuf st, tick;
+      if (st.alloc(8ull * nb_l) == 0 && tick.alloc(32) == 0) {
+        HIP_CHECK(hipMemsetAsync(st.p, 0, 8ull * nb_l));
+        HIP_CHECK(hipMemsetAsync(tick.p, 0, 32));
+        hipLaunchKernelGGL(k_scan_lookback, dim3(nb_l), dim3(BLOCK), 0, 0,
+                           d_in, d_out, n, (uint64_t*)st.p, (uint32_t*)tick.p,
+                           (uint32_t*)tick.p + 1, (uint64_t*)tick.p + 2);
+        uint64_t hh[2] = {0, 0}; /* {ticket+error, total} */
+        HIP_CHECK(hipMemcpy(hh, tick.p, 16, hipMemcpyDeviceToHost));
+        uint32_t err = (uint32_t)(hh[0] >> 32);
+        if (!err) {
+          if (h_total) {
+            HIP_CHECK(hipMemcpy(h_total, (uint64_t*)tick.p + 2, 8,
+                                hipMemcpyDeviceToHost));
+          }
+          return 0;
+        }
+        /* lookback timeout: in-place input is already destroyed — that
+           should never happen; fail loudly rather than silently misscan */
+        if (d_in == d_out) FAIL(-70, "scan lookback timeout (in-place)");
+        /* out-of-place: fall through to the classic path */
+      }
+    }
+  }
   /* read the last input BEFORE the scan: for in-place scans (d_in == d_out)
      it is overwritten with the exclusive prefix */
   uint64_t last_val = 0;
