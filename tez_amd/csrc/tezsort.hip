@@ -286,67 +286,30 @@ __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, i);
-    /* word-windowed access to the serialized key region: byte j is sliced
-       out of 4-aligned u32 loads (the <=3-byte overread stays inside the
-       +64 B pool pad; the compiler CSEs repeated window words).  The old
-       bytewise form issued ~27 one-byte loads per record. */
-    uintptr_t kA = (uintptr_t)v.key;
-    const uint32_t* kw = (const uint32_t*)(kA & ~(uintptr_t)3);
-    uint32_t ksh = (uint32_t)(kA & 3);
-    auto kbyte = [&](uint32_t j) -> uint32_t {
-      uint32_t q = ksh + j;
-      return (kw[q >> 2] >> (8 * (q & 3))) & 0xFFu;
-    };
-    uint32_t coff = (uint32_t)(v.content - v.key);
     uint32_t part;
-    if (d_part) {
-      part = (uint32_t)d_part[i];
-    } else {
-      /* HashPartitioner.hashBytes word-at-a-time: 4 Horner steps fold into
-         h = h*31^4 + b0*31^3 + b1*31^2 + b2*31 + b3 (mod 2^32, signed
-         bytes sign-extended — WritableComparator.hashBytes semantics) */
-      int32_t h = 0;
-      uint32_t j = 0;
-      while (j < v.clen && ((ksh + coff + j) & 3)) {
-        h = h * 31 + (int32_t)(int8_t)kbyte(coff + j);
-        j++;
-      }
-      const int32_t M2 = 961, M3 = 29791, M4 = 923521;
-      while (j + 4 <= v.clen) {
-        uint32_t w = kw[(ksh + coff + j) >> 2];
-        h = h * M4 + (int32_t)(int8_t)(w & 0xFF) * M3
-            + (int32_t)(int8_t)((w >> 8) & 0xFF) * M2
-            + (int32_t)(int8_t)((w >> 16) & 0xFF) * 31
-            + (int32_t)(int8_t)(w >> 24);
-        j += 4;
-      }
-      while (j < v.clen) {
-        h = h * 31 + (int32_t)(int8_t)kbyte(coff + j);
-        j++;
-      }
-      part = ((uint32_t)h & 0x7fffffff) % P;
-    }
+    if (d_part) part = (uint32_t)d_part[i];
+    else part = (uint32_t)((d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P);
     uint64_t key;
     if (ser_mode) {
       /* the reference prefix keeps only proxy >>> (bitcount(P)+1) bits
          (PipelinedSorter.java:457): ties on the TRUNCATED proxy fall through
          to the serialized compare, so the composite must truncate too */
-      uint32_t proxy = ((v.clen > 0 ? kbyte(coff) : 0u) << 16) |
-                       ((v.clen > 1 ? kbyte(coff + 1) : 0u) << 8) |
-                       (v.clen > 2 ? kbyte(coff + 2) : 0u);
+      uint32_t proxy = ((v.clen > 0 ? (uint32_t)v.content[0] : 0u) << 16) |
+                       ((v.clen > 1 ? (uint32_t)v.content[1] : 0u) << 8) |
+                       (v.clen > 2 ? (uint32_t)v.content[2] : 0u);
       int pw = 24 - ref_pb;            /* surviving proxy bits */
       if (pw < 0) pw = 0;
       uint32_t proxy_t = pw ? (proxy >> (24 - pw)) : 0;
       uint64_t ser = 0;
       uint32_t m = v.klen < 8 ? v.klen : 8;
-      for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)kbyte(b) << (56 - 8 * b);
+      for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)v.key[b] << (56 - 8 * b);
       key = ((uint64_t)part << (64 - pbits))
             | ((uint64_t)proxy_t << (64 - pbits - pw))
             | (ser >> (pbits + pw));
     } else {
       uint64_t c = 0;
       uint32_t m = v.clen < 8 ? v.clen : 8;
-      for (uint32_t b = 0; b < m; b++) c |= (uint64_t)kbyte(coff + b) << (56 - 8 * b);
+      for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
       key = pbits ? (((uint64_t)part << (64 - pbits)) | (c >> pbits)) : c;
     }
     d_key[i] = key & mask;
