@@ -217,7 +217,9 @@ def main():
         if not tdist.is_initialized():
             tdist.init_process_group("nccl", rank=rank, world_size=world)
 
-    n_local = args.records // n_gpus
+    # WEAK scaling (as reported in the JSON): every rank processes the full
+    # per-GPU batch; whole-job bytes grow with N.  (records is per rank.)
+    n_local = args.records
     conf = tez_amd.make_conf(PARTS)
     c5_part = None
     adopt = False

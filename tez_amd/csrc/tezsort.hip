@@ -1117,6 +1117,137 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
   }
 }
 
+/* Staged-span emit: each WAVE owns 64 consecutive sorted records — within a
+ * partition their output bytes are one contiguous span.  Every lane stages
+ * its own record (header regs + funneled u32 window loads) into a wave-local
+ * LDS image laid out so image words map to 4-aligned OUTPUT words, then the
+ * wave stores the span as dense aligned u32s.  Wave-synchronous (no
+ * barriers); LDS byte writes are byte-granular so record boundaries need no
+ * read-modify-write.  Cross-partition / short / oversized waves fall back to
+ * the per-record loop.  Fixes the two measured failure modes: scattered
+ * partial-line stores (lane-per-record v1) and per-rotate register pressure
+ * (word-funnel v2), while keeping 64 records of gather MLP in flight. */
+#define SPAN_BYTES (64 * 144 + 16)
+__global__ __launch_bounds__(BLOCK) void k_emit_span(
+    const RecDesc* desc, const uint8_t* same, const uint64_t* scan,
+    const uint32_t* parts, const uint64_t* seg_payload_start,
+    const uint64_t* part_scan_base, uint8_t* out, uint32_t n) {
+  __shared__ __attribute__((aligned(16))) uint8_t img8[WPB][SPAN_BYTES];
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  uint32_t wv = threadIdx.x / WAVE;
+  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
+  uint8_t* img = img8[wv];
+  uint32_t* img32 = (uint32_t*)img;
+  for (uint64_t base = (uint64_t)wave * WAVE; base < n;
+       base += (uint64_t)nwaves * WAVE) {
+    uint32_t i = (uint32_t)base + lane;
+    uint64_t my_src = 0, my_dst = 0, my_h0 = 0, my_h1 = 0;
+    uint32_t my_len = 0, my_hdr = 0, my_p = 0;
+    if (i < n) {
+      RecDesc v = desc[i];
+      my_p = parts[i];
+      my_dst = seg_payload_start[my_p] + (scan[i] - part_scan_base[my_p]);
+      uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
+      uint8_t hdrbuf[16] = {0};
+      uint32_t hdr = 0;
+      if (same[i]) {
+        if (!prev_same) hdrbuf[hdr++] = 0xFE;
+        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
+        my_src = v.src + v.klen;
+        my_len = v.vlen;
+      } else {
+        if (prev_same) hdrbuf[hdr++] = 0xFD;
+        hdr += d_vint_write(hdrbuf + hdr, v.klen);
+        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
+        my_src = v.src;
+        my_len = v.klen + v.vlen;
+      }
+      my_hdr = hdr;
+      for (int b = 0; b < 8; b++) my_h0 |= (uint64_t)hdrbuf[b] << (8 * b);
+      for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
+    }
+    uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
+    uint32_t p0 = __shfl(my_p, 0);
+    uint64_t okb = __ballot(i >= n || my_p == p0);
+    uint64_t span0 = __shfl(my_dst, 0);
+    uint64_t span_end = __shfl(my_dst + my_hdr + my_len, WAVE - 1);
+    uint32_t s = (uint32_t)(span0 & 3);
+    uint64_t span_len = span_end - span0;
+    if (nvalid == WAVE && okb == ~0ull && s + span_len <= SPAN_BYTES - 8) {
+      uint32_t io = (uint32_t)(my_dst - span0) + s;
+      uint32_t tot = my_hdr + my_len;
+      const uint8_t* sp = (const uint8_t*)(uintptr_t)my_src;
+      /* A: bytes up to the first aligned word boundary past the header */
+      uint32_t B0 = (io + my_hdr + 3) & ~3u;
+      if (B0 > io + tot) B0 = io + tot;
+      for (uint32_t q = io; q < B0; q++) {
+        uint32_t k = q - io;
+        img[q] = (k < my_hdr) ? ((k < 8) ? (uint8_t)(my_h0 >> (8 * k))
+                                         : (uint8_t)(my_h1 >> (8 * (k - 8))))
+                              : sp[k - my_hdr];
+      }
+      /* B: full image words from funneled source window words, in batches of
+         8 so the global loads issue independently (a serial prev/next funnel
+         chain measured 3x slower: one vmcnt stall per word) */
+      uint32_t B1 = (io + tot) & ~3u;
+      if (B1 > B0) {
+        uint32_t off0 = B0 - io - my_hdr;
+        uint64_t S = my_src + off0;
+        const uint32_t* sw = (const uint32_t*)(S & ~3ull);
+        uint32_t sh = (uint32_t)(S & 3), nw = (B1 - B0) >> 2;
+        uint32_t jw = B0 >> 2;
+        uint32_t j = 0;
+        while (j < nw) {
+          uint32_t cnt = nw - j;
+          if (cnt > 8) cnt = 8;
+          uint32_t r[9];
+          #pragma unroll
+          for (uint32_t k = 0; k < 9; k++)
+            if (k <= cnt) r[k] = sw[j + k];
+          #pragma unroll
+          for (uint32_t k = 0; k < 8; k++)
+            if (k < cnt)
+              img32[jw + j + k] =
+                  sh ? ((r[k] >> (8 * sh)) | (r[k + 1] << (8 * (4 - sh)))) : r[k];
+          j += cnt;
+        }
+      }
+      /* C: tail bytes */
+      for (uint32_t q = (B1 > B0 ? B1 : B0); q < io + tot; q++)
+        img[q] = sp[q - io - my_hdr];
+      /* wave store: head bytes, dense aligned words, tail bytes */
+      uint32_t hb = (4 - s) & 3;
+      if (hb > span_len) hb = (uint32_t)span_len;
+      if (lane < hb) out[span0 + lane] = img[s + lane];
+      uint32_t j0 = (s + hb) >> 2;
+      uint32_t j1 = (uint32_t)((s + span_len) >> 2);
+      for (uint32_t j = j0 + lane; j < j1; j += WAVE)
+        *(uint32_t*)(out + span0 - s + 4ull * j) = img32[j];
+      uint32_t tpos = 4u * j1;
+      uint32_t tend = (uint32_t)(s + span_len);
+      if (lane < tend - tpos) out[span0 - s + tpos + lane] = img[tpos + lane];
+    } else {
+      /* fallback: one record at a time, whole wave cooperates */
+      for (uint32_t r = 0; r < nvalid; r++) {
+        uint64_t src = __shfl(my_src, r);
+        uint64_t dsto = __shfl(my_dst, r);
+        uint64_t h0 = __shfl(my_h0, r);
+        uint64_t h1 = __shfl(my_h1, r);
+        uint32_t len = __shfl(my_len, r);
+        uint32_t hdr = __shfl(my_hdr, r);
+        uint8_t* w = out + dsto;
+        if (lane < hdr)
+          w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
+                               : (uint8_t)(h1 >> (8 * (lane - 8)));
+        w += hdr;
+        const uint8_t* sp2 = (const uint8_t*)(uintptr_t)src;
+        for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp2[b];
+      }
+    }
+  }
+}
+
 /* Uniform-record emit fast path: when every record serializes to the same
  * length and RLE is off, the IFile body is a constant-stride stream
  * (hdr vints ‖ key ‖ val per record).  One LANE per record: the 88-byte
@@ -2629,11 +2760,30 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
                        (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
                        (const uint64_t*)d_scanbase.p, d_out, n, uh_len, uh_word, uni_rec);
   } else {
-    hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
-                       (const RecDesc*)descbuf.p,
-                       (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
-                       (const uint32_t*)s->parts_sorted.p, (const uint64_t*)d_paystart.p,
-                       (const uint64_t*)d_scanbase.p, d_out, n, force_simple);
+    /* k_emit_span (LDS span image + dense aligned stores) measured 29-32 ms
+       vs 10.3 ms at C2 — the per-lane staging loops and LDS round trip cost
+       more than the wide stores save; L2 already merges the byte path's
+       consecutive stores.  Kept for experiments under TZS_EMIT_SPAN=1. */
+    static int use_span = -1;
+    if (use_span < 0) {
+      const char* e = getenv("TZS_EMIT_SPAN");
+      use_span = (e && e[0] == '1') ? 1 : 0;
+    }
+    if (use_span && !force_simple) {
+      hipLaunchKernelGGL(k_emit_span, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
+                         (const RecDesc*)descbuf.p,
+                         (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
+                         (const uint32_t*)s->parts_sorted.p,
+                         (const uint64_t*)d_paystart.p,
+                         (const uint64_t*)d_scanbase.p, d_out, n);
+    } else {
+      hipLaunchKernelGGL(k_emit_records, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
+                         (const RecDesc*)descbuf.p,
+                         (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
+                         (const uint32_t*)s->parts_sorted.p,
+                         (const uint64_t*)d_paystart.p,
+                         (const uint64_t*)d_scanbase.p, d_out, n, force_simple);
+    }
   }
   (void)hipEventRecord(ev[5]);
 
