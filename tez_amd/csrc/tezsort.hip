@@ -513,14 +513,15 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
 #define OS_CNT_MASK ((1u << 30) - 1)
 typedef unsigned int __attribute__((address_space(1))) os_gu32;
 
-__global__ void k_global_hist_all(const uint64_t* keys, uint32_t n, int first_byte,
+template <typename KeyT>
+__global__ void k_global_hist_all(const KeyT* keys, uint32_t n, int first_byte,
                                   int npasses, uint32_t* counts /* [npasses][256] */) {
   extern __shared__ uint32_t lh[]; /* npasses * 256 */
   for (int i = threadIdx.x; i < npasses * RADIX; i += blockDim.x) lh[i] = 0;
   __syncthreads();
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
-    uint64_t k = keys[i];
+    KeyT k = keys[i];
     for (int p = 0; p < npasses; p++)
       atomicAdd(&lh[p * RADIX + ((uint32_t)(k >> (8 * (first_byte + p))) & 0xFF)], 1u);
   }
@@ -1939,15 +1940,22 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
   uint64_t* b64out = (uint64_t*)tb64.p;
   int passes = 0;
 
-  /* onesweep path (u64 keys, base sort shape): one global-histogram read for
-     all passes, then a single lookback kernel per pass.  Falls back to the
-     classic 3-kernel pass on a lookback timeout. */
+  /* onesweep path (any key width / payload shape, incl. the refinement
+     seg-sorts): one global-histogram read for all passes, then a single
+     lookback kernel per pass.  Falls back to the classic 3-kernel pass on a
+     lookback timeout. */
   static int use_onesweep = -1;
   if (use_onesweep < 0) {
     const char* e = getenv("TZS_ONESWEEP");
     use_onesweep = (e && e[0] == '0') ? 0 : 1;
   }
-  if (use_onesweep && sizeof(KeyT) == 8 && !has_a1 && !has_b64 && n >= 20000) {
+  /* payload-carrying onesweep (a1/b64 staged through LDS) measured SLOWER
+     than the classic 3-kernel path on C3's refinement seg-sorts (48.9 ->
+     44.9 GB/s): the 50-57 KB LDS footprint halves resident blocks and each
+     seg-sort pays a host histogram sync.  The dispatch below supports every
+     shape; the gate keeps onesweep to the base (key + record-id) sort where
+     it wins. */
+  if (use_onesweep && !has_a1 && !has_b64 && n >= 20000) {
     int npasses = nbytes_key - first_byte;
     static thread_local DBuf gh, gbases, st, tick;
     if (gh.alloc(4u * npasses * RADIX)) return -12;
@@ -1956,8 +1964,8 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     if (tick.alloc(16)) return -12;
     HIP_CHECK(hipMemsetAsync(tick.p, 0, 16));  /* err word cleared once */
     HIP_CHECK(hipMemsetAsync(gh.p, 0, 4u * npasses * RADIX));
-    hipLaunchKernelGGL(k_global_hist_all, dim3(grid1d(n)), dim3(BLOCK),
-                       (uint32_t)(4 * npasses * RADIX), 0, (const uint64_t*)kin, n,
+    hipLaunchKernelGGL((k_global_hist_all<KeyT>), dim3(grid1d(n)), dim3(BLOCK),
+                       (uint32_t)(4 * npasses * RADIX), 0, kin, n,
                        first_byte, npasses, (uint32_t*)gh.p);
     std::vector<uint32_t> h_cnt(npasses * RADIX), h_base(npasses * RADIX);
     HIP_CHECK(hipMemcpy(h_cnt.data(), gh.p, 4u * npasses * RADIX,
@@ -1979,14 +1987,32 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
       HIP_CHECK(hipMemsetAsync(tick.p, 0, 4));
       if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                       (void)hipEventRecord(evs[nev]); }
-      hipLaunchKernelGGL((k_onesweep_pass<uint64_t, false>), dim3(nb), dim3(BLOCK), 0, 0,
-                         (const uint64_t*)kin, (uint64_t*)kout, a0in, a0out,
-                         nullptr, nullptr, nullptr, nullptr, n, b,
-                         (const uint32_t*)((uint32_t*)gbases.p + p * RADIX),
-                         (uint32_t*)st.p, (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      const uint32_t* pbases = (const uint32_t*)((uint32_t*)gbases.p + p * RADIX);
+      if (has_b64 && has_a1)
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, true, true>), dim3(nb), dim3(BLOCK),
+                           0, 0, kin, kout, a0in, a0out, a1in, a1out, b64in, b64out,
+                           n, b, pbases, (uint32_t*)st.p, (uint32_t*)tick.p,
+                           (uint32_t*)tick.p + 1);
+      else if (has_b64)
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, true>), dim3(nb), dim3(BLOCK),
+                           0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
+                           b64in, b64out, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      else if (has_a1)
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, true, false>), dim3(nb), dim3(BLOCK),
+                           0, 0, kin, kout, a0in, a0out, a1in, a1out,
+                           nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      else
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false>), dim3(nb), dim3(BLOCK),
+                           0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
+                           nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
       if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
       std::swap(kin, kout);
       std::swap(a0in, a0out);
+      if (has_a1) std::swap(a1in, a1out);
+      if (has_b64) std::swap(b64in, b64out);
       passes++;
     }
     {
