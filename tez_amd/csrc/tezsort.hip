@@ -793,20 +793,21 @@ __global__ void k_eq_init(const uint64_t* skeys, uint8_t* eq, uint32_t n) {
        i += gridDim.x * blockDim.x)
     eq[i] = (i > 0 && skeys[i] == skeys[i - 1]) ? 1 : 0;
 }
-/* inrun[i] = eq[i] || eq[i+1]; runstart[i] = inrun && !eq[i] */
-__global__ void k_run_flags(const uint8_t* eq, uint64_t* inrun, uint64_t* runstart,
-                            uint32_t n) {
+/* inrun[i] = eq[i] || eq[i+1]; runstart[i] = inrun && !eq[i].
+ * Both 0/1 flags ride ONE u64 as (inrun << 32) | runstart so a single
+ * scan_u64 produces both prefix sums — each field's total is <= n <= 2^32-1
+ * so the low field can never carry into the high one. */
+__global__ void k_run_flags(const uint8_t* eq, uint64_t* packed, uint32_t n) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
-    uint8_t in = eq[i] || (i + 1 < n && eq[i + 1]);
-    inrun[i] = in;
-    runstart[i] = (in && !eq[i]) ? 1 : 0;
+    uint64_t in = (eq[i] || (i + 1 < n && eq[i + 1])) ? 1 : 0;
+    packed[i] = (in << 32) | (uint64_t)(in && !eq[i]);
   }
 }
 /* compact ambiguous elements; seg = run rank; gather level key */
 __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
                                  const uint8_t* eq,
-                                 const uint64_t* inrun_scan, const uint64_t* runstart_scan,
+                                 const uint64_t* packed_scan,
                                  uint32_t n, int level_byte0, int use_len_level,
                                  int ser_mode,
                                  uint64_t* lkey, uint32_t* seg, uint32_t* pos) {
@@ -814,10 +815,11 @@ __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
        i += gridDim.x * blockDim.x) {
     uint8_t in = eq[i] || (i + 1 < n && eq[i + 1]);
     if (!in) continue;
-    uint32_t j = (uint32_t)inrun_scan[i];
+    uint64_t ps = packed_scan[i];
+    uint32_t j = (uint32_t)(ps >> 32);
     /* run rank via the EXCLUSIVE runstart scan: rank = scan + is_start - 1 */
     uint32_t is_start = (eq[i] == 0) ? 1u : 0u;
-    uint32_t sg = (uint32_t)runstart_scan[i] + is_start - 1;
+    uint32_t sg = (uint32_t)(ps & 0xFFFFFFFFu) + is_start - 1;
     RecView v = rt_view(rt, sidx[i]);
     const uint8_t* src = ser_mode ? v.key : v.content;
     uint32_t slen = ser_mode ? v.klen : v.clen;
@@ -2472,7 +2474,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     HIP_CHECK(hipMemcpy(&max_klen, dmax.p, 4, hipMemcpyDeviceToHost));
   }
   int max_clen = (int)max_klen; /* upper bound on content length */
-  static thread_local DBuf inrun, runstart, inrun_scan, runstart_scan, eqcnt;
+  static thread_local DBuf inrun, inrun_scan, eqcnt;
   static thread_local DBuf lkey, seg, pos, slotpos;
   if (eqcnt.alloc(4)) return -12;
 
@@ -2488,16 +2490,13 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     if (neq == 0) break;
     int use_len = (li == max_levels); /* final tiebreak: content length */
     if (inrun.alloc(sizeof(uint64_t) * n)) return -12;
-    if (runstart.alloc(sizeof(uint64_t) * n)) return -12;
     if (inrun_scan.alloc(sizeof(uint64_t) * n)) return -12;
-    if (runstart_scan.alloc(sizeof(uint64_t) * n)) return -12;
     hipLaunchKernelGGL(k_run_flags, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq,
-                       (uint64_t*)inrun.p, (uint64_t*)runstart.p, n);
-    uint64_t m64 = 0;
-    if (scan_u64((uint64_t*)inrun.p, (uint64_t*)inrun_scan.p, n, &m64)) return -12;
-    uint64_t nruns = 0;
-    if (scan_u64((uint64_t*)runstart.p, (uint64_t*)runstart_scan.p, n, &nruns)) return -12;
-    uint32_t m = (uint32_t)m64;
+                       (uint64_t*)inrun.p, n);
+    uint64_t ptotal = 0;
+    if (scan_u64((uint64_t*)inrun.p, (uint64_t*)inrun_scan.p, n, &ptotal)) return -12;
+    uint64_t nruns = ptotal & 0xFFFFFFFFu;
+    uint32_t m = (uint32_t)(ptotal >> 32);
     if (m == 0) break;
     if (lkey.alloc(sizeof(uint64_t) * m)) return -12;
     if (seg.alloc(sizeof(uint32_t) * m)) return -12;
@@ -2505,7 +2504,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     if (slotpos.alloc(sizeof(uint32_t) * m)) return -12;
     int lb0 = c0 + 8 * li;
     hipLaunchKernelGGL(k_compact_refine, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
-                       d_eq, (uint64_t*)inrun_scan.p, (uint64_t*)runstart_scan.p, n,
+                       d_eq, (uint64_t*)inrun_scan.p, n,
                        lb0, use_len, ser_mode, (uint64_t*)lkey.p, (uint32_t*)seg.p,
                        (uint32_t*)pos.p);
     /* slotpos = copy of pos (ascending) before sort */
@@ -2541,7 +2540,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   /* large-n headroom: refinement scratch and radix ping-pong temps are dead
      from here; return them to the pool before the output-stream allocation
      (at C3's 1e9 records these hold ~60 GB) */
-  inrun.release(); runstart.release(); inrun_scan.release(); runstart_scan.release();
+  inrun.release(); inrun_scan.release();
   lkey.release(); seg.release(); pos.release(); slotpos.release();
   radix_release_temps();
 
