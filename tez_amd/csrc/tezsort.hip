@@ -847,6 +847,21 @@ __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
  * order).  So: new_sidx[ pos_of_slot[j] ] = old_sidx[ pos[j] ] where
  * pos_of_slot = the ORIGINAL pos array in its pre-sort (position-ascending)
  * order.  We pass both. */
+/* two-phase in-place variant: {pos} and {slotpos} are the same position SET
+ * (the in-run slots), so gathering old values first then scattering touches
+ * only m slots — no full-n index copy */
+__global__ void k_refine_gather(const uint32_t* pos_sorted, const uint32_t* sidx,
+                                uint32_t* tmp, uint32_t m) {
+  for (uint32_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m;
+       j += gridDim.x * blockDim.x)
+    tmp[j] = sidx[pos_sorted[j]];
+}
+__global__ void k_refine_scatter(const uint32_t* slot_positions, const uint32_t* tmp,
+                                 uint32_t* sidx, uint32_t m) {
+  for (uint32_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m;
+       j += gridDim.x * blockDim.x)
+    sidx[slot_positions[j]] = tmp[j];
+}
 __global__ void k_refine_apply(const uint32_t* pos_sorted_elements /* pos[j] after sort */,
                                const uint32_t* slot_positions /* ascending positions */,
                                const uint32_t* old_sidx, uint32_t* new_sidx,
@@ -855,9 +870,15 @@ __global__ void k_refine_apply(const uint32_t* pos_sorted_elements /* pos[j] aft
        j += gridDim.x * blockDim.x)
     new_sidx[slot_positions[j]] = old_sidx[pos_sorted_elements[j]];
 }
-/* update eq inside runs after a refine level */
+/* update eq inside runs after a refine level; counts survivors so the next
+ * level can skip its full-n ambiguity pass */
 __global__ void k_eq_update(const uint64_t* lkey_sorted, const uint32_t* seg_sorted,
-                            const uint32_t* slot_positions, uint8_t* eq, uint32_t m) {
+                            const uint32_t* slot_positions, uint8_t* eq, uint32_t m,
+                            uint32_t* survivors) {
+  __shared__ uint32_t s;
+  if (threadIdx.x == 0) s = 0;
+  __syncthreads();
+  uint32_t loc = 0;
   for (uint32_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m;
        j += gridDim.x * blockDim.x) {
     uint8_t e = 0;
@@ -865,7 +886,11 @@ __global__ void k_eq_update(const uint64_t* lkey_sorted, const uint32_t* seg_sor
       e = 1;
     /* an element that was a run start keeps eq=0; others get refined eq */
     eq[slot_positions[j]] = e;
+    loc += e;
   }
+  atomicAdd(&s, loc);
+  __syncthreads();
+  if (threadIdx.x == 0 && s) atomicAdd(survivors, s);
 }
 __global__ void k_count_nonzero_u8(const uint8_t* a, uint32_t n, uint32_t* out) {
   __shared__ uint32_t s;
@@ -948,17 +973,26 @@ struct RecDesc {
   uint32_t klen;
   uint32_t vlen;
 };
-__global__ void k_build_desc(RecTable rt, const uint32_t* sidx, RecDesc* desc,
-                             uint32_t n) {
+__global__ void k_build_desc(RecTable rt, const uint32_t* sidx /* NULL: identity */,
+                             RecDesc* desc, uint32_t n) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
-    RecView v = rt_view(rt, sidx[i]);
+    RecView v = rt_view(rt, sidx ? sidx[i] : i);
     RecDesc d;
     d.src = (uint64_t)(uintptr_t)v.key;
     d.klen = v.klen;
     d.vlen = v.vlen;
     desc[i] = d;
   }
+}
+/* permute descriptors into sorted order: one 16B random read per record vs
+ * k_build_desc's 3-5 scattered off/klen/data reads through sidx (2x less
+ * line traffic for variable-length tables) */
+__global__ void k_permute_desc(const RecDesc* src, const uint32_t* sidx,
+                               RecDesc* dst, uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x)
+    dst[i] = src[sidx[i]];
 }
 /* same-as-prev full-key flags are exactly the final eq[] array.
  * writer-sameness (what the IFile stream encodes as RLE) additionally
@@ -2480,13 +2514,16 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
 
   int level = 1;
   const int max_levels = (max_clen > c0) ? (max_clen - c0 + 7) / 8 : 0;
-  for (int li = 0; li <= max_levels; li++) {
-    /* count remaining ambiguity */
+  /* initial ambiguity count; later levels get it from k_eq_update's
+     survivor counter (saves a full-n read per level) */
+  uint32_t neq = 0;
+  {
     HIP_CHECK(hipMemsetAsync(eqcnt.p, 0, 4));
     hipLaunchKernelGGL(k_count_nonzero_u8, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq, n,
                        (uint32_t*)eqcnt.p);
-    uint32_t neq = 0;
     HIP_CHECK(hipMemcpy(&neq, eqcnt.p, 4, hipMemcpyDeviceToHost));
+  }
+  for (int li = 0; li <= max_levels; li++) {
     if (neq == 0) break;
     int use_len = (li == max_levels); /* final tiebreak: content length */
     if (inrun.alloc(sizeof(uint64_t) * n)) return -12;
@@ -2520,20 +2557,20 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     rc = radix_sort<uint32_t>((uint32_t*)seg.p, (uint32_t*)pos.p, nullptr, m, segbytes,
                               0, nullptr, nullptr, nullptr, (uint64_t*)lkey.p, &lkey);
     if (rc) return rc;
-    /* apply permutation to d_idx and update eq */
-    static thread_local DBuf idx_new;
-    if (idx_new.alloc(sizeof(uint32_t) * n)) return -12;
-    HIP_CHECK(hipMemcpyAsync(idx_new.p, d_idx, sizeof(uint32_t) * n,
-                             hipMemcpyDeviceToDevice));
-    hipLaunchKernelGGL(k_refine_apply, dim3(grid1d(m)), dim3(BLOCK), 0, 0,
-                       (const uint32_t*)pos.p, (const uint32_t*)slotpos.p, d_idx,
-                       (uint32_t*)idx_new.p, m);
-    std::swap(s->sidx, idx_new);
-    idx_new.release(); /* old sidx buffer — back to the pool */
-    d_idx = (uint32_t*)s->sidx.p;
+    /* apply permutation to d_idx in place ({pos} == {slotpos} as position
+       sets, so only the m in-run slots change) and update eq */
+    static thread_local DBuf idx_tmp;
+    if (idx_tmp.alloc(sizeof(uint32_t) * m)) return -12;
+    hipLaunchKernelGGL(k_refine_gather, dim3(grid1d(m)), dim3(BLOCK), 0, 0,
+                       (const uint32_t*)pos.p, d_idx, (uint32_t*)idx_tmp.p, m);
+    hipLaunchKernelGGL(k_refine_scatter, dim3(grid1d(m)), dim3(BLOCK), 0, 0,
+                       (const uint32_t*)slotpos.p, (const uint32_t*)idx_tmp.p,
+                       d_idx, m);
+    HIP_CHECK(hipMemsetAsync(eqcnt.p, 0, 4));
     hipLaunchKernelGGL(k_eq_update, dim3(grid1d(m)), dim3(BLOCK), 0, 0,
                        (const uint64_t*)lkey.p, (const uint32_t*)seg.p,
-                       (const uint32_t*)slotpos.p, d_eq, m);
+                       (const uint32_t*)slotpos.p, d_eq, m, (uint32_t*)eqcnt.p);
+    HIP_CHECK(hipMemcpy(&neq, eqcnt.p, 4, hipMemcpyDeviceToHost));
     level++;
   }
   (void)hipEventRecord(ev[3]);
@@ -2544,12 +2581,9 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   lkey.release(); seg.release(); pos.release(); slotpos.release();
   radix_release_temps();
 
-  /* 4. writer-rle decision + same flags */
-  uint32_t neq_final = 0;
-  HIP_CHECK(hipMemsetAsync(eqcnt.p, 0, 4));
-  hipLaunchKernelGGL(k_count_nonzero_u8, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq, n,
-                     (uint32_t*)eqcnt.p);
-  HIP_CHECK(hipMemcpy(&neq_final, eqcnt.p, 4, hipMemcpyDeviceToHost));
+  /* 4. writer-rle decision + same flags (neq carried out of the refinement
+     loop — after the final level eq[i] means full-key-equal) */
+  uint32_t neq_final = neq;
   int writer_rle;
   if (s->conf.rle >= 0) writer_rle = s->conf.rle;
   else writer_rle = ((uint64_t)neq_final * 10 > n) ? 1 : 0;
@@ -2625,8 +2659,26 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   }
   static thread_local DBuf descbuf;
   if (descbuf.alloc(sizeof(RecDesc) * n)) return -12;
-  hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
-                     (RecDesc*)descbuf.p, n);
+  {
+    bool direct = rt.key_type != 1;
+    for (int sp2 = 0; sp2 < rt.nspills && direct; sp2++)
+      if (!rt.rec_u[sp2]) direct = false;
+    if (direct) {
+      /* uniform BytesWritable tables: rt_view is pure arithmetic */
+      hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
+                         (RecDesc*)descbuf.p, n);
+    } else {
+      /* variable-length / Text: build in original order (coalesced off/klen/
+         vint reads), then one 16B-per-record permute */
+      static thread_local DBuf desc0;
+      if (desc0.alloc(sizeof(RecDesc) * n)) return -12;
+      hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                         (const uint32_t*)nullptr, (RecDesc*)desc0.p, n);
+      hipLaunchKernelGGL(k_permute_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                         (const RecDesc*)desc0.p, d_idx, (RecDesc*)descbuf.p, n);
+      desc0.release();
+    }
+  }
   if (s->sizes.alloc(sizeof(uint64_t) * n)) return -12;
   if (s->scan.alloc(sizeof(uint64_t) * n)) return -12;
   hipLaunchKernelGGL(k_emit_sizes, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
