@@ -804,12 +804,37 @@ __global__ void k_run_flags(const uint8_t* eq, uint64_t* packed, uint32_t n) {
     packed[i] = (in << 32) | (uint64_t)(in && !eq[i]);
   }
 }
-/* compact ambiguous elements; seg = run rank; gather level key */
+/* level keys for EVERY record in original-record order (coalesced off/klen/
+ * data reads): used when most elements are still ambiguous — a random
+ * per-element gather through sidx is latency-bound, the dense build + one
+ * 8B permuted read is not */
+__global__ void k_build_lkeys(RecTable rt, int level_byte0, int use_len_level,
+                              int ser_mode, uint64_t* lk0, uint32_t n) {
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n;
+       g += gridDim.x * blockDim.x) {
+    RecView v = rt_view(rt, g);
+    const uint8_t* src = ser_mode ? v.key : v.content;
+    uint32_t slen = ser_mode ? v.klen : v.clen;
+    uint64_t k = 0;
+    if (use_len_level) {
+      k = slen;
+    } else {
+      for (int b = 0; b < 8; b++) {
+        uint32_t cb = (uint32_t)(level_byte0 + b);
+        uint8_t byte = (cb < slen) ? src[cb] : 0;
+        k |= (uint64_t)byte << (56 - 8 * b);
+      }
+    }
+    lk0[g] = k;
+  }
+}
+/* compact ambiguous elements; seg = run rank; gather level key (from lk0
+ * when prebuilt, else straight from the record table) */
 __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
                                  const uint8_t* eq,
                                  const uint64_t* packed_scan,
                                  uint32_t n, int level_byte0, int use_len_level,
-                                 int ser_mode,
+                                 int ser_mode, const uint64_t* lk0,
                                  uint64_t* lkey, uint32_t* seg, uint32_t* pos) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
@@ -820,13 +845,16 @@ __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
     /* run rank via the EXCLUSIVE runstart scan: rank = scan + is_start - 1 */
     uint32_t is_start = (eq[i] == 0) ? 1u : 0u;
     uint32_t sg = (uint32_t)(ps & 0xFFFFFFFFu) + is_start - 1;
-    RecView v = rt_view(rt, sidx[i]);
-    const uint8_t* src = ser_mode ? v.key : v.content;
-    uint32_t slen = ser_mode ? v.klen : v.clen;
     uint64_t k = 0;
-    if (use_len_level) {
-      k = slen;
+    if (lk0) {
+      k = lk0[sidx[i]];
+    } else if (use_len_level) {
+      RecView v = rt_view(rt, sidx[i]);
+      k = ser_mode ? v.klen : v.clen;
     } else {
+      RecView v = rt_view(rt, sidx[i]);
+      const uint8_t* src = ser_mode ? v.key : v.content;
+      uint32_t slen = ser_mode ? v.klen : v.clen;
       for (int b = 0; b < 8; b++) {
         uint32_t cb = (uint32_t)(level_byte0 + b);
         uint8_t byte = (cb < slen) ? src[cb] : 0;
@@ -2540,10 +2568,23 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     if (pos.alloc(sizeof(uint32_t) * m)) return -12;
     if (slotpos.alloc(sizeof(uint32_t) * m)) return -12;
     int lb0 = c0 + 8 * li;
-    hipLaunchKernelGGL(k_compact_refine, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
-                       d_eq, (uint64_t*)inrun_scan.p, n,
-                       lb0, use_len, ser_mode, (uint64_t*)lkey.p, (uint32_t*)seg.p,
-                       (uint32_t*)pos.p);
+    {
+      /* dense original-order level-key build when most records are still
+         ambiguous (m >= ~n/2): coalesced reads beat the random gather */
+      static thread_local DBuf lk0;
+      const uint64_t* lk0p = nullptr;
+      if ((uint64_t)m * 2 >= n) {
+        if (lk0.alloc(8ull * n)) return -12;
+        hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                           lb0, use_len, ser_mode, (uint64_t*)lk0.p, n);
+        lk0p = (const uint64_t*)lk0.p;
+      }
+      hipLaunchKernelGGL(k_compact_refine, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                         d_idx, d_eq, (uint64_t*)inrun_scan.p, n,
+                         lb0, use_len, ser_mode, lk0p, (uint64_t*)lkey.p,
+                         (uint32_t*)seg.p, (uint32_t*)pos.p);
+      lk0.release();
+    }
     /* slotpos = copy of pos (ascending) before sort */
     HIP_CHECK(hipMemcpyAsync(slotpos.p, pos.p, sizeof(uint32_t) * m,
                              hipMemcpyDeviceToDevice));
