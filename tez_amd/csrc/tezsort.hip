@@ -2003,6 +2003,7 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
   uint64_t* b64in = d_b64;
   uint64_t* b64out = (uint64_t*)tb64.p;
   int passes = 0;
+  bool pass_skip[16] = {false};
 
   /* onesweep path (any key width / payload shape, incl. the refinement
      seg-sorts): one global-histogram read for all passes, then a single
@@ -2093,7 +2094,28 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     goto finish;
   }
 
+  /* classic path: one upfront all-pass histogram detects single-digit
+     passes (Zipf/short keys leave low level-key bytes constant or zero) so
+     they skip entirely — one extra read of the keys + one sync buys up to
+     nbytes_key-1 saved passes */
+  if (n >= 100000 && nbytes_key - first_byte > 1) {
+    int npasses = nbytes_key - first_byte;
+    static thread_local DBuf gh2;
+    if (gh2.alloc(4u * npasses * RADIX) == 0) {
+      HIP_CHECK(hipMemsetAsync(gh2.p, 0, 4u * npasses * RADIX));
+      hipLaunchKernelGGL((k_global_hist_all<KeyT>), dim3(grid1d(n)), dim3(BLOCK),
+                         (uint32_t)(4 * npasses * RADIX), 0, kin, n,
+                         first_byte, npasses, (uint32_t*)gh2.p);
+      std::vector<uint32_t> h_cnt2(npasses * RADIX);
+      HIP_CHECK(hipMemcpy(h_cnt2.data(), gh2.p, 4u * npasses * RADIX,
+                          hipMemcpyDeviceToHost));
+      for (int p = 0; p < npasses && p < 16; p++)
+        for (int d = 0; d < RADIX; d++)
+          if (h_cnt2[p * RADIX + d] == n) { pass_skip[p] = true; break; }
+    }
+  }
   for (int b = first_byte; b < nbytes_key; b++) {
+    if (b - first_byte < 16 && pass_skip[b - first_byte]) continue;
     hipLaunchKernelGGL((k_radix_hist<KeyT>), dim3(nb), dim3(BLOCK), 0, 0, kin, n, b,
                        (uint32_t*)counts.p);
     hipLaunchKernelGGL(k_radix_scan_blocks, dim3(RADIX), dim3(BLOCK), 0, 0,
