@@ -1121,10 +1121,10 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
     }
     if (nvalid == WAVE && maxlen <= 32 * HB && !force_simple) {
       /* half h handles records 2r+h; 2-deep rotate pipeline per half.
-       * (A word-funnel variant — aligned u32 body stores assembled from
-       * funneled u32 loads — measured 17.7 ms vs 10.3 ms here: the extra
-       * per-rotate live state and doubled loads cost more than the wider
-       * stores saved.  Byte-granular moves + L2 write merging win.) */
+       * Byte-granular moves are the measured optimum here: a word-funnel
+       * STORE variant ran 17.7 vs 10.3 ms (register pressure), and a
+       * one-window-load + shuffle-redistribution GATHER variant was
+       * neutral (10.6 ms) — L1 already serves the byte re-reads. */
       uint32_t r0 = half;           /* first record index for this half */
       uint64_t src0 = __shfl(my_src, r0), dst0 = __shfl(my_dst, r0);
       uint64_t h00 = __shfl(my_h0, r0), h10 = __shfl(my_h1, r0);
@@ -1810,10 +1810,11 @@ static std::unordered_map<size_t, std::vector<void*>>& pool_map() {
 }
 static size_t pool_class(size_t n) {
   /* pow2 classes up to 1 GiB; 256 MiB steps beyond (a 105 GB stream must
-     not round to 128 GB on a 288 GB device).  +64 B pad: word-granular
-     kernels may read a few bytes past a logical end (funnel loads, CRC
-     trailer word) — the pad keeps every such read inside the allocation. */
-  n += 64;
+     not round to 128 GB on a 288 GB device).  +128 B pad: word- and
+     window-granular kernels may read past a logical end (funnel loads, CRC
+     trailer word, the emit gather's 128 B record window) — the pad keeps
+     every such read inside the allocation. */
+  n += 128;
   const size_t GB = 1ull << 30;
   if (n > GB) return (n + (256ull << 20) - 1) & ~((256ull << 20) - 1);
   size_t c = 1 << 16;
