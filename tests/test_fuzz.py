@@ -128,10 +128,11 @@ def test_gpu_randomized_config_sweep():
     if not tez_amd.device_available():
         pytest.skip("no GPU")
     rng = random.Random(0xF1122)
-    for trial in range(20):
+    for trial in range(32):
         P = rng.choice([1, 2, 3, 7, 16, 63, 200])
         text = rng.random() < 0.5
         dup = rng.random() < 0.4
+        send_empty = 1 if rng.random() < 0.8 else 0
         nspill = rng.choice([1, 1, 2, 3])
         # variable-length TezBytes keys: multi-spill merge order is unpinned
         # in the reference (segments not comparator-sorted — DESIGN.md §3);
@@ -147,7 +148,8 @@ def test_gpu_randomized_config_sweep():
                          for _ in range(rng.randrange(0 if not text else 1, 20)))
                    for _ in range(max(1, n_per // (4 if dup else 1)))]
         conf = tez_amd.make_conf(P, key_type=key_type, comparator=comparator,
-                                 combiner=combiner)
+                                 combiner=combiner,
+                                 send_empty_partition_details=send_empty)
         s = tez_amd.Sorter(conf)
         spills = []
         for sp in range(nspill):
@@ -163,13 +165,15 @@ def test_gpu_randomized_config_sweep():
             s.spill()
             d, f, kl = o.build_records(pairs)
             spills.append(o.spill(d, f, kl, P, key_type=key_type,
-                                  comparator=comparator, combiner=combiner))
+                                  comparator=comparator, combiner=combiner,
+                                  send_empty=bool(send_empty)))
         s.flush()
         got, gidx = s.output()
         s.close()
         gate = combiner if len(spills) >= 3 else 0
         want = (spills[0] if len(spills) == 1 else
-                o.final_merge(spills, P, comparator=comparator, combiner=gate))
+                o.final_merge(spills, P, comparator=comparator, combiner=gate,
+                              send_empty=bool(send_empty)))
         ctx = f"trial={trial} P={P} text={text} dup={dup} nspill={nspill} comb={combiner} n={n_per}"
         assert gidx == o.index_decode(want["index"], P), ctx
         assert got == want["data"], ctx
