@@ -84,14 +84,14 @@ def run_step_c5(tez_amd, d, off, kl, part, n):
     return ctr, tms
 
 
-def run_step_single(tez_amd, conf, d, off, kl, n, adopt=False):
+def run_step_single(tez_amd, conf, d, off, kl, n, adopt=False, part=None):
     t0 = time.perf_counter()
     s = tez_amd.Sorter(conf)
     t1 = time.perf_counter()
     if adopt:
-        s.write_batch_device_adopt(d, off, kl, None, n)
+        s.write_batch_device_adopt(d, off, kl, part, n)
     else:
-        s.write_batch_device(d, off, kl, None, n)
+        s.write_batch_device(d, off, kl, part, n)
     t2 = time.perf_counter()
     s.flush()
     t3 = time.perf_counter()
@@ -106,21 +106,24 @@ def run_step_single(tez_amd, conf, d, off, kl, n, adopt=False):
     return ctr, tms
 
 
-def run_step_multi(tez_amd, rank, world, device, d, off, kl, n, adopt=False):
+def run_step_multi(tez_amd, rank, world, device, d, off, kl, n, adopt=False,
+                   part=None, nparts=PARTS):
     from tez_amd import exchange as ex
-    conf = tez_amd.make_conf(PARTS, world_size=world, rank=rank)
+    conf = tez_amd.make_conf(nparts, world_size=world, rank=rank)
     m = tez_amd.Sorter(conf)
     if adopt:
-        m.write_batch_device_adopt(d, off, kl, None, n)
+        m.write_batch_device_adopt(d, off, kl, part, n)
     else:
-        m.write_batch_device(d, off, kl, None, n)
+        m.write_batch_device(d, off, kl, part, n)
     m.flush()
     ctr = m.counters()
     d_data, d_off2, d_klen2, rec_ranges, byte_ranges = m.sorted_columnar()
     plan = ex.plan_send(rec_ranges, byte_ranges, world)
     sd, sr, sk = ex.pack_send_tensors(m, plan, device)
     rd, rrl, rkl = ex.exchange(plan, sd, sr, sk)
-    red = ex.reduce_merge(lambda: tez_amd.Sorter(tez_amd.make_conf(PARTS)), rd, rrl, rkl)
+    rparts = ex.exchange_parts(plan, nparts, device)
+    red = ex.reduce_merge(lambda: tez_amd.Sorter(tez_amd.make_conf(nparts)),
+                          rd, rrl, rkl, rparts)
     tms = red.times()
     red.close()
     m.close()
@@ -171,10 +174,13 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--records", type=int, default=100_000_000,
                     help="total records across all ranks (C2 default 1e8)")
-    ap.add_argument("--workload", choices=["c2", "c3", "c5"], default="c2",
+    ap.add_argument("--workload", choices=["c2", "c3", "c4", "c5"], default="c2",
                     help="c3: 1 GPU, Text/Zipf keys, 256 partitions, forced "
                          "spills merged at flush (BASELINE configs[2]; "
                          "--records total, --spills segments). "
+                         "c4: 199 partitions with Zipf(1.0)-skewed partition "
+                         "SIZES (configs[3]); 2 GiB per rank default; at "
+                         "world>1 the skew drives the all-to-all-v. "
                          "c5: TeraSort-shaped 10B key + 90B value, range "
                          "partitions (single-GPU slice of configs[4])")
     ap.add_argument("--spills", type=int, default=32)
@@ -243,26 +249,44 @@ def main():
             tez_amd.free_device(part)
             gen_batches.append((d, off, kl, per))
     else:
-        # C2: one input set per step so the sorter can ADOPT it zero-copy
+        # C2/C4: one input set per step so the sorter can ADOPT it zero-copy
         # (one materialization of the records, like the reference's collect
         # serializing into its sort buffer).  Falls back to one shared input
         # + copying absorb when the pre-generated sets would not fit HBM.
+        # C4 (configs[3]): 199 partitions with Zipf(1.0)-skewed SIZES via the
+        # generator's inverse-CDF LUT; explicit per-record partitions ride
+        # through the C-ABI; 2 GiB per rank by default.
+        c4 = args.workload == "c4"
+        if c4:
+            if args.records == 100_000_000:
+                args.records = 20_000_000  # ~2 GiB of 100B records per rank
+                n_local = args.records
+            klen_w, vlen_w, kind_w, parts_w = 10, 90, 3, 199
+            conf = tez_amd.make_conf(parts_w)
+        else:
+            klen_w, vlen_w, kind_w, parts_w = KLEN, VLEN, 0, PARTS
+        rec_ser_w = 4 + klen_w + 4 + vlen_w
         total_sets = args.warmup + args.steps
         adopt = (not args.no_adopt
-                 and total_sets * n_local * REC_SER <= 120e9)
+                 and total_sets * n_local * rec_ser_w <= 120e9)
         if adopt:
             input_sets = []
             for sidx in range(total_sets):
                 d, off, kl, part = tez_amd.generate(
-                    seed=SEED + rank + 7919 * sidx, n=n_local, kind=0,
-                    klen=KLEN, vlen=VLEN, conf=conf)
-                tez_amd.free_device(part)
-                input_sets.append((d, off, kl))
+                    seed=SEED + rank + 7919 * sidx, n=n_local, kind=kind_w,
+                    klen=klen_w, vlen=vlen_w, conf=conf)
+                if not c4:
+                    tez_amd.free_device(part)
+                    part = None
+                input_sets.append((d, off, kl, part))
             set_cursor = [0]
         else:
-            d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local, kind=0,
-                                                klen=KLEN, vlen=VLEN, conf=conf)
-            tez_amd.free_device(part)
+            d, off, kl, part = tez_amd.generate(seed=SEED + rank, n=n_local,
+                                                kind=kind_w, klen=klen_w,
+                                                vlen=vlen_w, conf=conf)
+            if not c4:
+                tez_amd.free_device(part)
+                part = None
 
     def barrier_sync():
         if dist:
@@ -292,15 +316,18 @@ def main():
         if args.workload == "c3":
             return run_step_c3(tez_amd, gen_batches, c3_free_inputs)
         if adopt:
-            sd, so, sk = input_sets[set_cursor[0]]
+            sd, so, sk, spart = input_sets[set_cursor[0]]
             set_cursor[0] += 1
             if use_exchange:
                 return run_step_multi(tez_amd, rank, world, device, sd, so, sk,
-                                      n_local, adopt=True)
-            return run_step_single(tez_amd, conf, sd, so, sk, n_local, adopt=True)
+                                      n_local, adopt=True, part=spart,
+                                      nparts=parts_w)
+            return run_step_single(tez_amd, conf, sd, so, sk, n_local, adopt=True,
+                                   part=spart)
         if use_exchange:
-            return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local)
-        return run_step_single(tez_amd, conf, d, off, kl, n_local)
+            return run_step_multi(tez_amd, rank, world, device, d, off, kl, n_local,
+                                  part=part, nparts=parts_w)
+        return run_step_single(tez_amd, conf, d, off, kl, n_local, part=part)
 
     last_ctr = last_tms = None
     if c3_free_inputs and args.steps > 1:
@@ -331,6 +358,8 @@ def main():
         tez_amd.free_device(d, off, kl, c5_part)
     elif not adopt:
         tez_amd.free_device(d, off, kl)
+        if part is not None:
+            tez_amd.free_device(part)
     # adopt mode: every input set was consumed (ownership moved to the sorter)
 
     total_bytes_per_step = last_ctr["output_bytes"] * n_gpus  # whole-job Σ
@@ -376,12 +405,17 @@ def main():
                              "C3: Text/Zipf keys 4-32B + 64B vals, 256 partitions, "
                              f"{args.spills} spills k-way merged"
                              if args.workload == "c3" else
+                             "C4: 100B records, 199 partitions, Zipf(1.0)-skewed "
+                             "partition sizes (all-to-all-v shape)"
+                             if args.workload == "c4" else
                              "C2: 1e8 rec x (16B unique key + 64B val), 64 partitions,"
                              " BytesWritable/TezBytesComparator, ordered shuffle"),
                 "records": args.records,
-                "key_bytes": KLEN,
-                "value_bytes": VLEN,
-                "partitions": PARTS,
+                "key_bytes": 10 if args.workload in ("c4", "c5") else KLEN,
+                "value_bytes": 90 if args.workload in ("c4", "c5") else VLEN,
+                "partitions": (199 if args.workload == "c4" else
+                               128 if args.workload == "c5" else
+                               256 if args.workload == "c3" else PARTS),
                 "parallelism": f"shuffle-shard p%%{n_gpus}" if n_gpus > 1 else "single",
             },
             "roofline": roofline,

@@ -73,6 +73,13 @@ def _worker(rank, world, tmpdir):
     assert got == want, f"rank {rank} data mismatch"
     assert rrl.tolist() == want_lens
     assert rkl.tolist() == [1] * len(want_lens)
+    # per-record partition reconstruction (explicit-partitioner provenance)
+    rparts = ex.exchange_parts(plan, P, rd.device)
+    want_parts = []
+    for src in range(world):
+        for p in ex.parts_for_dest(P, world, rank):
+            want_parts += [p] * (src + 1 + p)
+    assert rparts.tolist() == want_parts, f"rank {rank} parts mismatch"
     dist.destroy_process_group()
 
 

@@ -110,7 +110,9 @@ def test_exchange_world1_on_device(engine):
     plan = ex.plan_send(rr, br, 1)
     sd, srl, skl = ex.pack_send_tensors(m, plan, device)
     rd, rrl, rkl = ex.exchange(plan, sd, srl, skl)
-    red = ex.reduce_merge(lambda: engine.Sorter(engine.make_conf(P)), rd, rrl, rkl)
+    rparts = ex.exchange_parts(plan, P, device)
+    red = ex.reduce_merge(lambda: engine.Sorter(engine.make_conf(P)), rd, rrl, rkl,
+                          rparts)
     got_data, got_idx = red.output()
     red.close()
     m.close()

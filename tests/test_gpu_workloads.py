@@ -62,6 +62,32 @@ def test_c5_terasort_shape_range_partitions(engine):
     assert got == want["data"]
 
 
+def test_c4_zipf_skewed_partitions(engine):
+    """kind 3 (BASELINE configs[3] shape): uniform 10B keys whose PARTITION
+    SIZES follow Zipf(1.0) over 199 partitions via the generator's
+    inverse-CDF LUT; byte parity of the skewed sort vs the oracle."""
+    import collections
+    n, P = 30000, 199
+    conf = engine.make_conf(P)
+    d, off, kl, part = engine.generate(seed=0xC4, n=n, kind=3, klen=10, vlen=90,
+                                       conf=conf)
+    data, offs, klen, parts = pull_generated(engine, d, off, kl, part, n)
+    cnt = collections.Counter(parts.tolist())
+    # Zipf(1.0): partition 0 holds ~1/H(199) ~ 17%; heavy head, thin tail
+    assert cnt[0] > n * 0.10
+    assert cnt[0] > 5 * max(cnt.get(p, 0) for p in range(100, 199))
+    s = engine.Sorter(conf)
+    s.write_batch_device(d, off, kl, part, n)
+    s.flush()
+    got, gidx = s.output()
+    s.close()
+    engine.free_device(d, off, kl, part)
+    want = o.spill(data, offs, klen, P, key_type=o.KEY_BYTES,
+                   comparator=o.CMP_TEZBYTES, partitions=parts)
+    assert gidx == o.index_decode(want["index"], P)
+    assert got == want["data"]
+
+
 def test_c3_text_zipf_multi_spill(engine):
     """kind 1: variable-length Text keys (Zipf word + unique suffix),
     4 spills merged at flush — the C3 shape at oracle scale."""

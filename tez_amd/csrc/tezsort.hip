@@ -1671,6 +1671,19 @@ __global__ void k_range_partition(const uint8_t* data, uint64_t rec_bytes, int32
     d_part[i] = (int32_t)((uint64_t)v * (uint32_t)P >> 16);
   }
 }
+/* kind 3 (C4): uniform keys mapped through a host-built inverse-CDF LUT so
+ * PARTITION SIZES follow Zipf(1.0) — BASELINE configs[3]'s skewed
+ * 199-partition all-to-all-v shape. */
+__global__ void k_lut_partition(const uint8_t* data, uint64_t rec_bytes,
+                                const int32_t* lut /* [65536] */,
+                                int32_t* d_part, int64_t n) {
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* k = data + (uint64_t)i * rec_bytes + 4;
+    uint32_t v = ((uint32_t)k[0] << 8) | k[1];
+    d_part[i] = lut[v];
+  }
+}
 
 /* kind 1 (C3): Text keys, content = zipf(1.1)-drawn dict word + unique
  * base-36 record-id suffix, total length in [4,32]; 64B BytesWritable
@@ -3330,14 +3343,14 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
                             int32_t vlen, const tzs_conf* conf, void** d_data,
                             uint64_t** d_off, uint32_t** d_klen, int32_t** d_part) {
   if (ensure_device_constants()) return -70;
-  if (kind < 0 || kind > 2) FAIL(-22, "generator kind %d not implemented", kind);
+  if (kind < 0 || kind > 3) FAIL(-22, "generator kind %d not implemented", kind);
   void* dd = nullptr;
   uint64_t* doff = nullptr;
   uint32_t* dkl = nullptr;
   if (pool_alloc_raw(8 * (n + 1), (void**)&doff)) return -12;
   if (pool_alloc_raw(4 * n, (void**)&dkl)) return -12;
   uint64_t rec = 0;
-  if (kind == 0 || kind == 2) {
+  if (kind == 0 || kind == 2 || kind == 3) {
     rec = 4 + (uint64_t)klen + 4 + (uint64_t)vlen;
     if (pool_alloc_raw(rec * n, &dd)) return -12;
     hipLaunchKernelGGL(k_generate_fixed, dim3(grid1d(n)), dim3(BLOCK), 0, 0, seed, n,
@@ -3360,6 +3373,25 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
     if (kind == 2) {
       hipLaunchKernelGGL(k_range_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
                          (const uint8_t*)dd, rec, conf ? conf->num_partitions : 1, dp, n);
+    } else if (kind == 3) {
+      /* inverse-CDF LUT: partition p's share ∝ 1/(p+1) (Zipf s=1.0) */
+      int P = conf ? conf->num_partitions : 1;
+      std::vector<double> cdf(P + 1, 0.0);
+      for (int p = 0; p < P; p++) cdf[p + 1] = cdf[p] + 1.0 / (p + 1);
+      double tot = cdf[P];
+      std::vector<int32_t> lut(65536);
+      int p = 0;
+      for (int v = 0; v < 65536; v++) {
+        double x = (v + 0.5) / 65536.0 * tot;
+        while (p + 1 < P && cdf[p + 1] < x) p++;
+        lut[v] = p;
+      }
+      static thread_local DBuf dlut;
+      if (dlut.alloc(4 * 65536)) return -12;
+      HIP_CHECK(hipMemcpyAsync(dlut.p, lut.data(), 4 * 65536, hipMemcpyHostToDevice));
+      hipLaunchKernelGGL(k_lut_partition, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                         (const uint8_t*)dd, rec, (const int32_t*)dlut.p, dp, n);
+      HIP_CHECK(hipDeviceSynchronize());  /* lut is reused next call */
     } else {
       RecTable rt = {};
       rt.nspills = 1;

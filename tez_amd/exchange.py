@@ -113,10 +113,45 @@ def pack_send_tensors(sorter, plan: SendPlan, device):
     return send_data[:total_b], send_reclen[:total_r], send_klen[:total_r]
 
 
-def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen):
+def exchange_parts(plan: SendPlan, P: int, device, group=None):
+    """All-gather the per-partition record-count matrix (the 24B index
+    triples' record counts — the ShuffleHeader-level metadata) and rebuild
+    the per-record PARTITION ids of the records this rank receives, in
+    arrival order (src rank asc, then owned partitions asc — matching
+    all_to_all_single's concatenation and plan_send's packing order).
+    Preserves explicit partitioners (range/LUT) across the exchange; for
+    HashPartitioner data it also saves the reduce side a re-hash."""
+    import torch
+    import torch.distributed as dist
+    world = dist.get_world_size(group)
+    me = dist.get_rank(group)
+    counts = [0] * P
+    for i, p in enumerate(plan.order):
+        counts[p] = plan.seg_recs[i]
+    pc = torch.tensor(counts, dtype=torch.int64, device=device)
+    all_pc = [torch.empty_like(pc) for _ in range(world)]
+    dist.all_gather(all_pc, pc, group=group)
+    ids, cnts = [], []
+    for src in range(world):
+        src_pc = all_pc[src].tolist()
+        for p in range(P):
+            if p % world == me and src_pc[p]:
+                ids.append(p)
+                cnts.append(int(src_pc[p]))
+    if not ids:
+        return torch.empty(0, dtype=torch.int32, device=device)
+    return torch.repeat_interleave(
+        torch.tensor(ids, dtype=torch.int32, device=device),
+        torch.tensor(cnts, dtype=torch.int64, device=device))
+
+
+def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen,
+                 recv_parts=None):
     """Feed received columnar records into a reduce-side sorter (the
     MergeManager/TezMerger replacement) and flush: the final merged IFile for
-    this rank's owned partitions."""
+    this rank's owned partitions.  recv_parts (from exchange_parts) carries
+    the original partition ids; without it the sorter recomputes
+    HashPartitioner placement (wrong for explicit partitioners)."""
     import torch
     n = int(recv_reclen.numel())
     sorter = conf_factory()
@@ -124,8 +159,13 @@ def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen):
         off = torch.zeros(n + 1, dtype=torch.int64, device=recv_data.device)
         torch.cumsum(recv_reclen.to(torch.int64), 0, out=off[1:])
         klen_u32 = recv_klen.contiguous()
+        parts_ptr = None
+        parts_t = None
+        if recv_parts is not None and int(recv_parts.numel()) == n:
+            parts_t = recv_parts.contiguous()
+            parts_ptr = parts_t.data_ptr()
         sorter.write_batch_device(recv_data.data_ptr(), off.data_ptr(),
-                                  klen_u32.data_ptr(), None, n)
-        sorter._exchange_keepalive = (recv_data, off, klen_u32)
+                                  klen_u32.data_ptr(), parts_ptr, n)
+        sorter._exchange_keepalive = (recv_data, off, klen_u32, parts_t)
     sorter.flush()
     return sorter
