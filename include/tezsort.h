@@ -8,8 +8,13 @@
  * Conventions: all functions return 0 on success or a negative errno-style code;
  * tzs_last_error() returns a thread-local message for the last failure.
  * Pointers prefixed d_ are HIP device pointers; everything else is host memory.
- * Handles are single-threaded except where noted (mirrors ExternalSorter threading,
- * ExternalSorter.java:74-92 / PipelinedSorter.java:399 "synchronized collect").
+ * Threading: each handle is used by one thread AT A TIME, but the thread may
+ * change between calls (producer thread != flush thread — SURVEY §8b,
+ * mirrors ExternalSorter threading, ExternalSorter.java:74-92 /
+ * PipelinedSorter.java:399 "synchronized collect").  The device-buffer pool
+ * and allocation registry behind every handle are mutex-protected, so
+ * distinct handles may be driven from distinct threads concurrently; all
+ * device work runs on the null HIP stream (serialized per process).
  */
 #ifndef TEZSORT_H
 #define TEZSORT_H

@@ -17,6 +17,7 @@
 #include <cstring>
 #include <string>
 #include <vector>
+#include <mutex>
 #include <sys/stat.h>
 #include <sys/types.h>
 
@@ -1817,6 +1818,13 @@ static inline uint32_t grid_waves(uint64_t nrec) {
  * engine's buffer sizes recur every spill/step, so pooling removes that
  * entirely (288 GB HBM makes holding the pool cheap). */
 #include <unordered_map>
+/* host-side pool/registry lock: the C-ABI contract allows the producer
+   thread to differ from the flush thread; device work is still
+   one stream, but the allocator bookkeeping must not race */
+static std::mutex& pool_mu() {
+  static std::mutex m;
+  return m;
+}
 static std::unordered_map<size_t, std::vector<void*>>& pool_map() {
   static std::unordered_map<size_t, std::vector<void*>> m;
   return m;
@@ -1836,6 +1844,7 @@ static size_t pool_class(size_t n) {
 }
 static int pool_alloc(size_t n, void** out, size_t* cls_out) {
   size_t cls = pool_class(n);
+  std::lock_guard<std::mutex> lk(pool_mu());
   auto& fl = pool_map()[cls];
   if (!fl.empty()) { *out = fl.back(); fl.pop_back(); *cls_out = cls; return 0; }
   if (hipMalloc(out, cls) != hipSuccess) {
@@ -1851,7 +1860,11 @@ static int pool_alloc(size_t n, void** out, size_t* cls_out) {
   *cls_out = cls;
   return 0;
 }
-static void pool_free(void* p, size_t cls) { if (p) pool_map()[cls].push_back(p); }
+static void pool_free(void* p, size_t cls) {
+  if (!p) return;
+  std::lock_guard<std::mutex> lk(pool_mu());
+  pool_map()[cls].push_back(p);
+}
 
 /* raw pool allocations (generator buffers): class tracked in a registry so
  * tzs_free_device can return them to the pool */
@@ -1862,18 +1875,26 @@ static std::unordered_map<void*, size_t>& pool_registry() {
 static int pool_alloc_raw(size_t n, void** out) {
   size_t cls = 0;
   int rc = pool_alloc(n, out, &cls);
-  if (rc == 0) pool_registry()[*out] = cls;
+  if (rc == 0) {
+    std::lock_guard<std::mutex> lk(pool_mu());
+    pool_registry()[*out] = cls;
+  }
   return rc;
 }
 static void pool_free_raw(void* p) {
   if (!p) return;
-  auto it = pool_registry().find(p);
-  if (it != pool_registry().end()) {
-    pool_free(p, it->second);
+  size_t cls = 0;
+  {
+    std::lock_guard<std::mutex> lk(pool_mu());
+    auto it = pool_registry().find(p);
+    if (it == pool_registry().end()) {
+      (void)hipFree(p);
+      return;
+    }
+    cls = it->second;
     pool_registry().erase(it);
-  } else {
-    (void)hipFree(p);
   }
+  pool_free(p, cls);
 }
 
 struct DBuf {
@@ -2420,12 +2441,16 @@ extern "C" int tzs_sorter_write_batch_device_adopt(tzs_sorter* s, void* d_data,
   if (n > 4000000000ll) FAIL(-22, "batch too large (u32 record ids)");
   if (s->cur_n != 0 || !s->host_klen.empty())
     FAIL(-22, "adopt requires an empty current buffer (first batch of spill)");
+  std::unique_lock<std::mutex> reg_lk(pool_mu());
   auto& reg = pool_registry();
   auto id = reg.find(d_data), io = reg.find(d_off), ik = reg.find(d_klen);
   auto ip = d_part ? reg.find(d_part) : reg.end();
   if (id == reg.end() || io == reg.end() || ik == reg.end() ||
       (d_part && ip == reg.end()))
     FAIL(-22, "adopt requires buffers allocated by tzs_malloc_device/tzs_generate");
+  size_t cls_d = id->second, cls_o = io->second, cls_k = ik->second;
+  size_t cls_p = d_part ? ip->second : 0;
+  reg_lk.unlock();
   uint64_t nbytes = 0, first = 0;
   HIP_CHECK(hipMemcpy(&first, d_off, 8, hipMemcpyDeviceToHost));
   HIP_CHECK(hipMemcpy(&nbytes, d_off + n, 8, hipMemcpyDeviceToHost));
@@ -2447,11 +2472,18 @@ extern "C" int tzs_sorter_write_batch_device_adopt(tzs_sorter* s, void* d_data,
     s->cur_first_batch = false;
   }
   /* take ownership: registry entry -> DBuf (released to the pool later) */
-  s->cur_data.release(); s->cur_data.p = d_data; s->cur_data.sz = id->second; reg.erase(id);
-  s->cur_off.release();  s->cur_off.p = d_off;   s->cur_off.sz = io->second;  reg.erase(io);
-  s->cur_klen.release(); s->cur_klen.p = d_klen; s->cur_klen.sz = ik->second; reg.erase(ik);
+  {
+    std::lock_guard<std::mutex> lk(pool_mu());
+    reg.erase(d_data);
+    reg.erase(d_off);
+    reg.erase(d_klen);
+    if (d_part) reg.erase(d_part);
+  }
+  s->cur_data.release(); s->cur_data.p = d_data; s->cur_data.sz = cls_d;
+  s->cur_off.release();  s->cur_off.p = d_off;   s->cur_off.sz = cls_o;
+  s->cur_klen.release(); s->cur_klen.p = d_klen; s->cur_klen.sz = cls_k;
   if (d_part) {
-    s->cur_part.release(); s->cur_part.p = d_part; s->cur_part.sz = ip->second; reg.erase(ip);
+    s->cur_part.release(); s->cur_part.p = d_part; s->cur_part.sz = cls_p;
     s->have_explicit_parts = true;
   } else {
     if (s->cur_part.alloc(sizeof(int32_t) * n)) return -12;
