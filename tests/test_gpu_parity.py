@@ -259,6 +259,35 @@ def test_parity_adopted_batch(engine):
     s3.close()
 
 
+def test_cross_thread_write_then_flush(engine):
+    """the producer thread may differ from the flush thread
+    (ExternalSorter threading).  Writes happen on a worker thread, flush on
+    the main thread; output must equal the single-thread path."""
+    import threading
+    pairs = _mk_fixed(2000, 16, 32, seed=71)
+    conf = engine.make_conf(8)
+    s = engine.Sorter(conf)
+    err = []
+
+    def producer():
+        try:
+            for k, v in pairs:
+                s.write(k, v, -1)
+        except Exception as e:   # pragma: no cover
+            err.append(e)
+
+    t = threading.Thread(target=producer)
+    t.start()
+    t.join()
+    assert not err
+    s.flush()
+    got, gidx = s.output()
+    s.close()
+    want = _oracle_single_spill(pairs, 8, o.KEY_BYTES, o.CMP_TEZBYTES)
+    assert gidx == o.index_decode(want["index"], 8)
+    assert got == want["data"]
+
+
 def test_write_files_reference_layout(engine, tmp_path):
     pairs = _mk_fixed(100, 16, 16, seed=23)
     conf = engine.make_conf(4)
