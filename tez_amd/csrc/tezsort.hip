@@ -1881,11 +1881,20 @@ static int pool_alloc_raw(size_t n, void** out) {
   }
   return rc;
 }
+/* generator-produced uniformity hints, keyed by the offsets pointer: lets
+ * the zero-copy absorb skip its full-n uniformity scan (the producer KNOWS
+ * the records are fixed-stride) */
+struct UniformHint { uint32_t rec_u, klen_u; uint64_t nbytes; };
+static std::unordered_map<void*, UniformHint>& uniform_hints() {
+  static std::unordered_map<void*, UniformHint> m;
+  return m;
+}
 static void pool_free_raw(void* p) {
   if (!p) return;
   size_t cls = 0;
   {
     std::lock_guard<std::mutex> lk(pool_mu());
+    uniform_hints().erase(p);  /* a reused pointer must not inherit a hint */
     auto it = pool_registry().find(p);
     if (it == pool_registry().end()) {
       (void)hipFree(p);
@@ -2451,12 +2460,26 @@ extern "C" int tzs_sorter_write_batch_device_adopt(tzs_sorter* s, void* d_data,
   size_t cls_d = id->second, cls_o = io->second, cls_k = ik->second;
   size_t cls_p = d_part ? ip->second : 0;
   reg_lk.unlock();
-  uint64_t nbytes = 0, first = 0;
-  HIP_CHECK(hipMemcpy(&first, d_off, 8, hipMemcpyDeviceToHost));
-  HIP_CHECK(hipMemcpy(&nbytes, d_off + n, 8, hipMemcpyDeviceToHost));
-  if (first != 0) FAIL(-22, "d_off must start at 0");
-  /* uniform-record detection (same rule as the copy path) */
+  uint64_t nbytes = 0;
+  bool hinted = false;
   {
+    std::lock_guard<std::mutex> lk(pool_mu());
+    auto hh = uniform_hints().find(d_off);
+    if (hh != uniform_hints().end() && !getenv("TZS_NO_UNIFORM")) {
+      s->cur_rec_u = hh->second.rec_u;
+      s->cur_klen_u = hh->second.klen_u;
+      nbytes = hh->second.nbytes;
+      s->cur_first_batch = false;
+      uniform_hints().erase(hh);
+      hinted = true;
+    }
+  }
+  if (!hinted) {
+    uint64_t first = 0;
+    HIP_CHECK(hipMemcpy(&first, d_off, 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(&nbytes, d_off + n, 8, hipMemcpyDeviceToHost));
+    if (first != 0) FAIL(-22, "d_off must start at 0");
+    /* uniform-record detection (same rule as the copy path) */
     static thread_local DBuf mm;
     if (mm.alloc(32)) return -12;
     uint64_t init[4] = {~0ull, 0, ~0ull, 0};
@@ -3438,6 +3461,13 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
     *d_part = dp;
   }
   HIP_CHECK(hipDeviceSynchronize());
+  if (kind == 0 || kind == 2 || kind == 3) {
+    /* fixed-stride kinds: record uniformity so a zero-copy adopt can skip
+       its full-n uniformity scan */
+    std::lock_guard<std::mutex> lk(pool_mu());
+    uniform_hints()[doff] = UniformHint{(uint32_t)rec, (uint32_t)(4 + klen),
+                                        rec * (uint64_t)n};
+  }
   *d_data = dd;
   *d_off = doff;
   *d_klen = dkl;
