@@ -789,10 +789,17 @@ __global__ __launch_bounds__(BLOCK) void k_scan_lookback(
 
 /* ---- refinement ---- */
 /* eq[i] = 1 if sorted element i has same (still-ambiguous) key prefix as i-1 */
-__global__ void k_eq_init(const uint64_t* skeys, uint8_t* eq, uint32_t n) {
+__global__ void k_eq_init(const uint64_t* skeys, uint8_t* eq, int pbits,
+                          uint32_t* parts, uint32_t n) {
+  /* also materializes the per-position partition ids (composite top bits) —
+     skey positions are stable under refinement (intra-run permutations swap
+     EQUAL composites), so this single read serves both */
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x)
-    eq[i] = (i > 0 && skeys[i] == skeys[i - 1]) ? 1 : 0;
+       i += gridDim.x * blockDim.x) {
+    uint64_t k = skeys[i];
+    eq[i] = (i > 0 && k == skeys[i - 1]) ? 1 : 0;
+    parts[i] = pbits ? (uint32_t)(k >> (64 - pbits)) : 0;
+  }
 }
 /* inrun[i] = eq[i] || eq[i+1]; runstart[i] = inrun && !eq[i].
  * Both 0/1 flags ride ONE u64 as (inrun << 32) | runstart so a single
@@ -1002,8 +1009,20 @@ struct RecDesc {
   uint32_t klen;
   uint32_t vlen;
 };
+/* IFile body bytes of sorted record i (a run never crosses a partition
+ * segment: the previous record's RLE state is invisible to this partition's
+ * stream) — fused into the descriptor builders below */
+__device__ __forceinline__ uint64_t d_rec_emit_size(
+    const RecDesc& v, const uint8_t* same, const uint32_t* parts, uint32_t i) {
+  uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
+  if (same[i])
+    return (prev_same ? 0 : 1) /* RLE marker */ + d_vint_size(v.vlen) + v.vlen;
+  return (prev_same ? 1 : 0) /* V_END */ + d_vint_size(v.klen) + d_vint_size(v.vlen)
+         + v.klen + v.vlen;
+}
 __global__ void k_build_desc(RecTable rt, const uint32_t* sidx /* NULL: identity */,
-                             RecDesc* desc, uint32_t n) {
+                             RecDesc* desc, const uint8_t* same,
+                             const uint32_t* parts, uint64_t* sizes, uint32_t n) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += gridDim.x * blockDim.x) {
     RecView v = rt_view(rt, sidx ? sidx[i] : i);
@@ -1012,16 +1031,21 @@ __global__ void k_build_desc(RecTable rt, const uint32_t* sidx /* NULL: identity
     d.klen = v.klen;
     d.vlen = v.vlen;
     desc[i] = d;
+    if (sizes) sizes[i] = d_rec_emit_size(d, same, parts, i);
   }
 }
 /* permute descriptors into sorted order: one 16B random read per record vs
  * k_build_desc's 3-5 scattered off/klen/data reads through sidx (2x less
  * line traffic for variable-length tables) */
 __global__ void k_permute_desc(const RecDesc* src, const uint32_t* sidx,
-                               RecDesc* dst, uint32_t n) {
+                               RecDesc* dst, const uint8_t* same,
+                               const uint32_t* parts, uint64_t* sizes, uint32_t n) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x)
-    dst[i] = src[sidx[i]];
+       i += gridDim.x * blockDim.x) {
+    RecDesc d = src[sidx[i]];
+    dst[i] = d;
+    sizes[i] = d_rec_emit_size(d, same, parts, i);
+  }
 }
 /* same-as-prev full-key flags are exactly the final eq[] array.
  * writer-sameness (what the IFile stream encodes as RLE) additionally
@@ -1044,31 +1068,7 @@ __global__ void k_writer_same(RecTable rt, const uint32_t* sidx, const uint8_t* 
     same[i] = s;
   }
 }
-__global__ void k_emit_sizes(const RecDesc* desc, const uint8_t* same,
-                             const uint32_t* parts, uint64_t* sizes, uint32_t n) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x) {
-    RecDesc v = desc[i];
-    uint64_t sz;
-    /* a run never crosses a partition segment: the previous record's RLE
-       state is invisible to this partition's IFile stream */
-    uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
-    if (same[i]) {
-      sz = (prev_same ? 0 : 1) /* RLE marker */ + d_vint_size(v.vlen) + v.vlen;
-    } else {
-      sz = (prev_same ? 1 : 0) /* V_END */ + d_vint_size(v.klen) + d_vint_size(v.vlen)
-           + v.klen + v.vlen;
-    }
-    sizes[i] = sz;
-  }
-}
-/* partition id of sorted element i (from composite top bits) */
-__global__ void k_sorted_parts(const uint64_t* skeys, int pbits, uint32_t* parts,
-                               uint32_t n) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x)
-    parts[i] = pbits ? (uint32_t)(skeys[i] >> (64 - pbits)) : 0;
-}
+
 /* per-record IFile emit (replaces the IFile.Writer.append loop,
  * IFile.java:444-615): batch of 64 records per wave iteration — each lane
  * fetches ONE record's descriptor in parallel, then two records are kept in
@@ -2598,8 +2598,10 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
 
   /* 3. refinement levels */
   if (s->eq.alloc(n)) return -12;
+  if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
   uint8_t* d_eq = (uint8_t*)s->eq.p;
-  hipLaunchKernelGGL(k_eq_init, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, d_eq, n);
+  hipLaunchKernelGGL(k_eq_init, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, d_eq,
+                     pbits, (uint32_t*)s->parts_sorted.p, n);
   /* max content length: for refinement level count */
   /* refinement start byte within the comparator's byte source: sorted-covered
      bits are 8*SB - pbits (content mode) or 8*SB - pbits - 24 past the proxy
@@ -2723,10 +2725,7 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   /* combiner stage: replace the sorted view with folded records */
   s->combine_applied = false;
   if (apply_combine && s->conf.combiner == 1) {
-    /* partitions of the sorted records (needed for parts2) */
-    if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
-    hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
-                       (uint32_t*)s->parts_sorted.p, n);
+    /* partitions of the sorted records (parts_sorted, from k_eq_init) */
     static thread_local DBuf rs, rs_scan, pos2, lens2, off2, parts2, idx2;
     if (rs.alloc(8ull * n) || rs_scan.alloc(8ull * n)) return -12;
     hipLaunchKernelGGL(k_combine_mark, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq,
@@ -2783,40 +2782,40 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
   hipLaunchKernelGGL(k_writer_same, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx, d_eq,
                      writer_rle, (const uint8_t*)d_sprle.p, (uint8_t*)s->same.p, n);
 
-  /* 5. emit sizes + partition layout */
-  if (!s->combined_parts_valid) {
-    if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
-    hipLaunchKernelGGL(k_sorted_parts, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, pbits,
-                       (uint32_t*)s->parts_sorted.p, n);
-  }
+  /* 5. emit sizes + partition layout (parts_sorted came from k_eq_init; the
+     combiner path swapped in its own parts2) */
   static thread_local DBuf descbuf;
   if (descbuf.alloc(sizeof(RecDesc) * n)) return -12;
+  if (s->sizes.alloc(sizeof(uint64_t) * n)) return -12;
+  if (s->scan.alloc(sizeof(uint64_t) * n)) return -12;
   {
     bool direct = rt.key_type != 1;
     for (int sp2 = 0; sp2 < rt.nspills && direct; sp2++)
       if (!rt.rec_u[sp2]) direct = false;
     if (direct) {
-      /* uniform BytesWritable tables: rt_view is pure arithmetic */
+      /* uniform BytesWritable tables: rt_view is pure arithmetic; emit
+         sizes are computed in the same pass */
       hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt, d_idx,
-                         (RecDesc*)descbuf.p, n);
+                         (RecDesc*)descbuf.p, (const uint8_t*)s->same.p,
+                         (const uint32_t*)s->parts_sorted.p,
+                         (uint64_t*)s->sizes.p, n);
     } else {
       /* variable-length / Text: build in original order (coalesced off/klen/
-         vint reads), then one 16B-per-record permute */
+         vint reads), then one 16B-per-record permute + fused sizes */
       static thread_local DBuf desc0;
       if (desc0.alloc(sizeof(RecDesc) * n)) return -12;
       hipLaunchKernelGGL(k_build_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                         (const uint32_t*)nullptr, (RecDesc*)desc0.p, n);
+                         (const uint32_t*)nullptr, (RecDesc*)desc0.p,
+                         (const uint8_t*)nullptr, (const uint32_t*)nullptr,
+                         (uint64_t*)nullptr, n);
       hipLaunchKernelGGL(k_permute_desc, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
-                         (const RecDesc*)desc0.p, d_idx, (RecDesc*)descbuf.p, n);
+                         (const RecDesc*)desc0.p, d_idx, (RecDesc*)descbuf.p,
+                         (const uint8_t*)s->same.p,
+                         (const uint32_t*)s->parts_sorted.p,
+                         (uint64_t*)s->sizes.p, n);
       desc0.release();
     }
   }
-  if (s->sizes.alloc(sizeof(uint64_t) * n)) return -12;
-  if (s->scan.alloc(sizeof(uint64_t) * n)) return -12;
-  hipLaunchKernelGGL(k_emit_sizes, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
-                     (const RecDesc*)descbuf.p,
-                     (const uint8_t*)s->same.p, (const uint32_t*)s->parts_sorted.p,
-                     (uint64_t*)s->sizes.p, n);
   uint64_t total_body = 0;
   if (scan_u64((uint64_t*)s->sizes.p, (uint64_t*)s->scan.p, n, &total_body)) return -12;
   s->sizes.release();
