@@ -1831,11 +1831,11 @@ static std::unordered_map<size_t, std::vector<void*>>& pool_map() {
 }
 static size_t pool_class(size_t n) {
   /* pow2 classes up to 1 GiB; 256 MiB steps beyond (a 105 GB stream must
-     not round to 128 GB on a 288 GB device).  +128 B pad: word- and
-     window-granular kernels may read past a logical end (funnel loads, CRC
-     trailer word, the emit gather's 128 B record window) — the pad keeps
-     every such read inside the allocation. */
-  n += 128;
+     not round to 128 GB on a 288 GB device).  The class is computed on the
+     UNPADDED size — the +256 B overread pad is added to the physical
+     hipMalloc instead (pool_alloc), so exactly-pow2 buffers do not jump a
+     class (jumping inflated the 1e9-record working set past the drop-pool
+     retry threshold: C3 1e9 fell 53 -> 14 GB/s). */
   const size_t GB = 1ull << 30;
   if (n > GB) return (n + (256ull << 20) - 1) & ~((256ull << 20) - 1);
   size_t c = 1 << 16;
@@ -1847,12 +1847,16 @@ static int pool_alloc(size_t n, void** out, size_t* cls_out) {
   std::lock_guard<std::mutex> lk(pool_mu());
   auto& fl = pool_map()[cls];
   if (!fl.empty()) { *out = fl.back(); fl.pop_back(); *cls_out = cls; return 0; }
-  if (hipMalloc(out, cls) != hipSuccess) {
+  /* +256 B: word/window-granular kernels (funnel loads, CRC trailer word,
+     the emit gather window) may read a little past a logical end; padding
+     the physical allocation keeps every such read in bounds without
+     inflating the size class */
+  if (hipMalloc(out, cls + 256) != hipSuccess) {
     /* under pressure: drop the whole pool and retry once */
     for (auto& kv : pool_map())
       for (void* q : kv.second) (void)hipFree(q);
     pool_map().clear();
-    if (hipMalloc(out, cls) != hipSuccess) {
+    if (hipMalloc(out, cls + 256) != hipSuccess) {
       snprintf(g_err, sizeof(g_err), "hipMalloc(%zu) failed", cls);
       return -12;
     }
@@ -2663,10 +2667,13 @@ static int sort_and_emit(tzs_sorter* s, RecTable& rt, uint32_t n,
     int lb0 = c0 + 8 * li;
     {
       /* dense original-order level-key build when most records are still
-         ambiguous (m >= ~n/2): coalesced reads beat the random gather */
+         ambiguous (m >= ~n/2): coalesced reads beat the random gather.
+         Gated to n <= 3e8: the 8n-byte lk0 buffer at a 1e9-record merge
+         pushed the peak working set into pool-drop/hipMalloc churn
+         (C3 1e9 53 -> 14 GB/s). */
       static thread_local DBuf lk0;
       const uint64_t* lk0p = nullptr;
-      if ((uint64_t)m * 2 >= n) {
+      if ((uint64_t)m * 2 >= n && n <= 300000000u) {
         if (lk0.alloc(8ull * n)) return -12;
         hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
                            lb0, use_len, ser_mode, (uint64_t*)lk0.p, n);
