@@ -422,6 +422,8 @@ def main():
             "cpu_baseline": cpu,
             "phase_ms": {k: round(v / 1e6, 2) for k, v in last_tms.items()
                          if k.endswith("_ns")},
+            "pool": {k: (round(v / 1e9, 2) if k != "drops" else v)
+                     for k, v in tez_amd.pool_stats().items()},
         }
         print(json.dumps(out))
     if dist:

@@ -189,6 +189,12 @@ int tzs_generate(uint64_t seed, int64_t n, int32_t kind,
                  void** d_data, uint64_t** d_off, uint32_t** d_klen, int32_t** d_part);
 void tzs_free_device(void* d_ptr);
 
+/* Device-pool telemetry: out = {in-use bytes, held bytes, peak in-use bytes,
+ * drop-and-retry count}.  A non-zero drop count means the working set
+ * approached HBM capacity and the pool discarded idle buffers (hipMalloc
+ * churn follows — a performance smell at large configs). */
+void tzs_pool_stats(uint64_t out[4]);
+
 /* ---- introspection ------------------------------------------------------ */
 /* Per-phase HIP-event times of the last flush, nanoseconds. */
 typedef struct tzs_times {

@@ -8,6 +8,7 @@ OrderedPartitionedKVOutput / OrderedGroupedKVInput plugin surface
 from ._engine import (  # noqa: F401
     KEY_BYTES, KEY_TEXT, CMP_TEZBYTES, CMP_TEXT,
     Sorter, make_conf, upload_records, generate, free_device, device_available,
+    pool_stats,
     merge_segments, read_device,
 )
 from .conf import conf_from_tez_properties  # noqa: F401

@@ -121,6 +121,8 @@ def lib():
     L.tzs_memcpy_d2h.argtypes = [c.c_void_p, c.c_void_p, c.c_uint64]
     L.tzs_memcpy_h2d.argtypes = [c.c_void_p, c.c_void_p, c.c_uint64]
     L.tzs_malloc_device.argtypes = [c.c_uint64, c.POINTER(c.c_void_p)]
+    L.tzs_pool_stats.argtypes = [c.POINTER(c.c_uint64)]
+    L.tzs_pool_stats.restype = None
     L.tzs_device_available.restype = c.c_int
     L.tzs_test_crc_combine.restype = c.c_uint32
     L.tzs_test_crc_combine.argtypes = [c.c_uint32, c.c_uint32, c.c_uint64]
@@ -329,6 +331,13 @@ def merge_segments(conf, segments):
     if out.value:
         lib().tzs_free_device(out)
     return data, (rec.start_offset, rec.raw_length, rec.part_length)
+
+
+def pool_stats():
+    """{inuse, held, peak, drops} of the device buffer pool (bytes/counts)."""
+    arr = (ctypes.c_uint64 * 4)()
+    lib().tzs_pool_stats(arr)
+    return {"inuse": arr[0], "held": arr[1], "peak": arr[2], "drops": arr[3]}
 
 
 def device_available():

@@ -1829,6 +1829,11 @@ static std::unordered_map<size_t, std::vector<void*>>& pool_map() {
   static std::unordered_map<size_t, std::vector<void*>> m;
   return m;
 }
+/* telemetry (bytes): in-use = handed to callers; held = hipMalloc'd total;
+ * peak tracks in-use; drops counts drop-the-pool retries.  Guarded by
+ * pool_mu like the maps. */
+static size_t g_pool_inuse = 0, g_pool_held = 0, g_pool_peak = 0;
+static uint64_t g_pool_drops = 0;
 static size_t pool_class(size_t n) {
   /* pow2 classes up to 1 GiB; 256 MiB steps beyond (a 105 GB stream must
      not round to 128 GB on a 288 GB device).  The class is computed on the
@@ -1846,21 +1851,37 @@ static int pool_alloc(size_t n, void** out, size_t* cls_out) {
   size_t cls = pool_class(n);
   std::lock_guard<std::mutex> lk(pool_mu());
   auto& fl = pool_map()[cls];
-  if (!fl.empty()) { *out = fl.back(); fl.pop_back(); *cls_out = cls; return 0; }
+  if (!fl.empty()) {
+    *out = fl.back(); fl.pop_back(); *cls_out = cls;
+    g_pool_inuse += cls;
+    if (g_pool_inuse > g_pool_peak) g_pool_peak = g_pool_inuse;
+    return 0;
+  }
   /* +256 B: word/window-granular kernels (funnel loads, CRC trailer word,
      the emit gather window) may read a little past a logical end; padding
      the physical allocation keeps every such read in bounds without
      inflating the size class */
   if (hipMalloc(out, cls + 256) != hipSuccess) {
     /* under pressure: drop the whole pool and retry once */
+    size_t freed = 0;
     for (auto& kv : pool_map())
-      for (void* q : kv.second) (void)hipFree(q);
+      for (void* q : kv.second) { (void)hipFree(q); freed += kv.first; }
     pool_map().clear();
+    g_pool_held -= freed > g_pool_held ? g_pool_held : freed;
+    g_pool_drops++;
+    fprintf(stderr,
+            "[tzs pool] drop-and-retry #%llu: want %.2f GB, in-use %.2f GB, "
+            "freed %.2f GB of idle pool\n",
+            (unsigned long long)g_pool_drops, cls / 1e9, g_pool_inuse / 1e9,
+            freed / 1e9);
     if (hipMalloc(out, cls + 256) != hipSuccess) {
       snprintf(g_err, sizeof(g_err), "hipMalloc(%zu) failed", cls);
       return -12;
     }
   }
+  g_pool_held += cls;
+  g_pool_inuse += cls;
+  if (g_pool_inuse > g_pool_peak) g_pool_peak = g_pool_inuse;
   *cls_out = cls;
   return 0;
 }
@@ -1868,6 +1889,7 @@ static void pool_free(void* p, size_t cls) {
   if (!p) return;
   std::lock_guard<std::mutex> lk(pool_mu());
   pool_map()[cls].push_back(p);
+  g_pool_inuse -= cls > g_pool_inuse ? g_pool_inuse : cls;
 }
 
 /* raw pool allocations (generator buffers): class tracked in a registry so
@@ -3481,6 +3503,15 @@ extern "C" int tzs_generate(uint64_t seed, int64_t n, int32_t kind, int32_t klen
 }
 
 extern "C" void tzs_free_device(void* p) { pool_free_raw(p); }
+
+extern "C" void tzs_pool_stats(uint64_t out[4]) {
+  /* {in-use bytes, held bytes, peak in-use bytes, drop-and-retry count} */
+  std::lock_guard<std::mutex> lk(pool_mu());
+  out[0] = g_pool_inuse;
+  out[1] = g_pool_held;
+  out[2] = g_pool_peak;
+  out[3] = g_pool_drops;
+}
 
 /* ---- reduce-side merge over columnar segments ---- */
 extern "C" int tzs_merge_segments(const tzs_conf* conf, const tzs_segment* segs,
