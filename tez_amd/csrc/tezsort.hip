@@ -18,6 +18,7 @@
 #include <string>
 #include <vector>
 #include <mutex>
+#include <algorithm>
 #include <sys/stat.h>
 #include <sys/types.h>
 
@@ -1847,6 +1848,15 @@ static size_t pool_class(size_t n) {
   while (c < n) c <<= 1;
   return c;
 }
+/* NOTE on oversubscription: hipMalloc on this stack silently overcommits
+ * (301 GB "held" succeeded on a 288 GB device at C3 1e9) — harmless while
+ * the TOUCHED working set stays under physical HBM (in-use peak 218 GB
+ * there), but FIRST TOUCHES of fresh allocations beyond capacity fault at
+ * page-migration speed (~2 GB/s): an extra 8 GB scratch buffer cost ~4 s.
+ * Alloc-time eviction of idle buffers was tried and measured WORSE (24 GB/s
+ * vs 53 — hipFree device-syncs inside the hot path); the controls that work
+ * are (a) not allocating n-scaled scratch at the large-config peak (the lk0
+ * gate) and (b) the telemetry below to catch it. */
 static int pool_alloc(size_t n, void** out, size_t* cls_out) {
   size_t cls = pool_class(n);
   std::lock_guard<std::mutex> lk(pool_mu());
