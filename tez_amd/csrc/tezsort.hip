@@ -532,8 +532,8 @@ __global__ void k_global_hist_all(const KeyT* keys, uint32_t n, int first_byte,
     if (lh[i]) atomicAdd(&counts[i], lh[i]);
 }
 
-template <typename KeyT, bool HAS_A1, bool HAS_B64 = false>
-__global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
+template <typename KeyT, bool HAS_A1, bool HAS_B64 = false, int BLK = BLOCK>
+__global__ __launch_bounds__(BLK) void k_onesweep_pass(
     const KeyT* keys_in, KeyT* keys_out,
     const uint32_t* a0_in, uint32_t* a0_out,
     const uint32_t* a1_in, uint32_t* a1_out,
@@ -541,22 +541,24 @@ __global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
     uint32_t n, int byte_idx,
     const uint32_t* bases /* [256] exclusive digit bases */,
     uint32_t* status /* [ntiles*256] */, uint32_t* ticket, uint32_t* error) {
+  constexpr uint32_t OT = (uint32_t)TILE_ROUNDS * BLK;  /* elements per tile */
+  constexpr int OWPB = BLK / WAVE;
   __shared__ uint32_t tilecnt[RADIX];
-  __shared__ uint32_t wavehist[WPB][RADIX];
+  __shared__ uint32_t wavehist[OWPB][RADIX];
   __shared__ uint32_t tileoff[RADIX];
   __shared__ uint32_t excl[RADIX];
   __shared__ uint32_t s_tile;
-  __shared__ KeyT ls_key[TILE];
-  __shared__ uint32_t ls_a0[TILE];
-  __shared__ uint32_t ls_a1[HAS_A1 ? TILE : 1];
-  __shared__ uint64_t ls_b64[HAS_B64 ? TILE : 1];
-  __shared__ uint8_t ls_dig[TILE];
+  __shared__ KeyT ls_key[OT];
+  __shared__ uint32_t ls_a0[OT];
+  __shared__ uint32_t ls_a1[HAS_A1 ? OT : 1];
+  __shared__ uint64_t ls_b64[HAS_B64 ? OT : 1];
+  __shared__ uint8_t ls_dig[OT];
   if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
   for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
   __syncthreads();
   const uint32_t tile = s_tile;
-  const uint32_t start = tile * TILE;
-  const uint32_t end = min(start + TILE, n);
+  const uint32_t start = tile * OT;
+  const uint32_t end = min(start + OT, n);
   const uint32_t count = (start < n) ? (end - start) : 0;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wv = threadIdx.x / WAVE;
@@ -568,13 +570,13 @@ __global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
   uint32_t my_a1v[TILE_ROUNDS];
   uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
   int round = 0;
-  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
     uint32_t i = r0 + threadIdx.x;
     bool active = i < end;
     KeyT key = active ? keys_in[i] : (KeyT)0;
     uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0xFFFFFFFFu;
     __syncthreads();
-    for (int j = threadIdx.x; j < WPB * RADIX; j += blockDim.x)
+    for (int j = threadIdx.x; j < OWPB * RADIX; j += blockDim.x)
       ((uint32_t*)wavehist)[j] = 0;
     __syncthreads();
     uint64_t m = ~0ull;
@@ -602,7 +604,7 @@ __global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
     __syncthreads();
     for (int j = threadIdx.x; j < RADIX; j += blockDim.x) {
       uint32_t sum = 0;
-      for (int w = 0; w < WPB; w++) sum += wavehist[w][j];
+      for (int w = 0; w < OWPB; w++) sum += wavehist[w][j];
       tilecnt[j] += sum;
     }
   }
@@ -637,7 +639,7 @@ __global__ __launch_bounds__(BLOCK) void k_onesweep_pass(
   }
   /* place into LDS grouped by digit, then digit-contiguous global writes */
   round = 0;
-  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
     uint32_t i = r0 + threadIdx.x;
     if (i < end) {
       uint32_t slot = tileoff[my_dig[round]] + my_seq[round];
@@ -2102,10 +2104,21 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
      it wins. */
   if (use_onesweep && !has_a1 && !has_b64 && n >= 20000) {
     int npasses = nbytes_key - first_byte;
+    /* 512-thread blocks double the tile (4096 elements) WITHOUT losing
+       occupancy (2 resident blocks x 8 waves = the 256-thread config's 16
+       waves/CU) so digit runs double and partial-line scatter waste halves */
+    static int os_blk = -1;
+    if (os_blk < 0) {
+      const char* e = getenv("TZS_OS_BLK");
+      os_blk = e ? atoi(e) : 512;
+      if (os_blk != 256 && os_blk != 512) os_blk = 512;
+    }
+    uint32_t os_tile = (uint32_t)TILE_ROUNDS * (uint32_t)os_blk;
+    uint32_t nb_os = nblocks_for(n, os_tile);
     static thread_local DBuf gh, gbases, st, tick;
     if (gh.alloc(4u * npasses * RADIX)) return -12;
     if (gbases.alloc(4u * npasses * RADIX)) return -12;
-    if (st.alloc(4ull * nb * RADIX)) return -12;
+    if (st.alloc(4ull * nb_os * RADIX)) return -12;
     if (tick.alloc(16)) return -12;
     HIP_CHECK(hipMemsetAsync(tick.p, 0, 16));  /* err word cleared once */
     HIP_CHECK(hipMemsetAsync(gh.p, 0, 4u * npasses * RADIX));
@@ -2126,31 +2139,22 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                              hipMemcpyHostToDevice));
     for (int b = first_byte; b < nbytes_key; b++) {
       int p = b - first_byte;
-      HIP_CHECK(hipMemsetAsync(st.p, 0, 4ull * nb * RADIX));
+      HIP_CHECK(hipMemsetAsync(st.p, 0, 4ull * nb_os * RADIX));
       /* reset only the ticket word — the error word ([1]) accumulates across
          passes and is checked once after the loop */
       HIP_CHECK(hipMemsetAsync(tick.p, 0, 4));
       if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                       (void)hipEventRecord(evs[nev]); }
       const uint32_t* pbases = (const uint32_t*)((uint32_t*)gbases.p + p * RADIX);
-      if (has_b64 && has_a1)
-        hipLaunchKernelGGL((k_onesweep_pass<KeyT, true, true>), dim3(nb), dim3(BLOCK),
-                           0, 0, kin, kout, a0in, a0out, a1in, a1out, b64in, b64out,
-                           n, b, pbases, (uint32_t*)st.p, (uint32_t*)tick.p,
-                           (uint32_t*)tick.p + 1);
-      else if (has_b64)
-        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, true>), dim3(nb), dim3(BLOCK),
-                           0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
-                           b64in, b64out, n, b, pbases, (uint32_t*)st.p,
-                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
-      else if (has_a1)
-        hipLaunchKernelGGL((k_onesweep_pass<KeyT, true, false>), dim3(nb), dim3(BLOCK),
-                           0, 0, kin, kout, a0in, a0out, a1in, a1out,
+      /* base shape only (gate above): dispatch on block size */
+      if (os_blk == 512)
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false, 512>), dim3(nb_os),
+                           dim3(512), 0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
                            nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
                            (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
       else
-        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false>), dim3(nb), dim3(BLOCK),
-                           0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false, 256>), dim3(nb_os),
+                           dim3(256), 0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
                            nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
                            (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
       if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
