@@ -222,6 +222,48 @@ def test_text_comparator_order():
     assert got == sorted(contents)
 
 
+def test_spill_explicit_partitions():
+    """Explicit per-record partitions (range/LUT partitioners — C4/C5 shapes)
+    override HashPartitioner placement; order within a partition still follows
+    the comparator, and the independent model agrees."""
+    n, P = 800, 9
+    pairs = make_bytes_records(n, klen=10, vlen=4, seed=5, nparts=P)
+    import random as _r
+    rng = _r.Random(5)
+    parts = np.array([rng.randrange(P) for _ in range(n)], dtype=np.int32)
+    data, off, klen = o.build_records(pairs)
+    res = o.spill(data, off, klen, P, key_type=o.KEY_BYTES,
+                  comparator=o.CMP_TEZBYTES, partitions=parts, want_order=True)
+    # independent model with the EXPLICIT placement
+    import functools
+    items = []
+    for i, (k, v) in enumerate(pairs):
+        content = k[4:]
+        pref = py_prefix(parts[i], P, content, True)
+        items.append((pref, k, i, parts[i]))
+
+    def cmp(a, b):
+        if a[0] != b[0]:
+            return -1 if a[0] < b[0] else 1
+        c = py_cmp_key_tezbytes(a[1], b[1])
+        return c if c else a[2] - b[2]
+
+    items.sort(key=functools.cmp_to_key(cmp))
+    assert list(res["order"]) == [i for _, _, i, _ in items]
+    idx = o.index_decode(res["index"], P)
+    # every record landed in ITS partition
+    cursor = 0
+    import collections
+    by_part = collections.Counter(parts)
+    for p in range(P):
+        start, raw, plen = idx[p]
+        assert start == cursor
+        seg = res["data"][start:start + plen]
+        recs = o.ifile_read(seg, with_header=True) if plen else []
+        assert len(recs) == by_part.get(p, 0)
+        cursor += plen
+
+
 def test_spill_mt_matches_single_thread():
     """The partition-parallel spill (the multi-core CPU baseline) must be
     byte-identical to the single-threaded restatement."""
