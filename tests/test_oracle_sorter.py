@@ -239,8 +239,8 @@ def test_spill_explicit_partitions():
     items = []
     for i, (k, v) in enumerate(pairs):
         content = k[4:]
-        pref = py_prefix(parts[i], P, content, True)
-        items.append((pref, k, i, parts[i]))
+        pref = py_prefix(int(parts[i]), P, content, True)
+        items.append((pref, k, i, int(parts[i])))
 
     def cmp(a, b):
         if a[0] != b[0]:
