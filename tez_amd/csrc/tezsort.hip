@@ -2104,14 +2104,16 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
      it wins. */
   if (use_onesweep && !has_a1 && !has_b64 && n >= 20000) {
     int npasses = nbytes_key - first_byte;
-    /* 512-thread blocks double the tile (4096 elements) WITHOUT losing
-       occupancy (2 resident blocks x 8 waves = the 256-thread config's 16
-       waves/CU) so digit runs double and partial-line scatter waste halves */
+    /* bigger blocks scale the tile while KEEPING 16 waves/CU: 512 threads =
+       4096-elem tile (2 blocks x 8 waves), 1024 = 8192-elem tile (1 block x
+       16 waves, ~135 KB LDS) — digit runs grow to ~32 x 12 B = 384 B so
+       partial-line scatter waste nearly vanishes (measured 289/293/301 GB/s
+       for 256/512/1024 at C2; TZS_OS_BLK overrides) */
     static int os_blk = -1;
     if (os_blk < 0) {
       const char* e = getenv("TZS_OS_BLK");
-      os_blk = e ? atoi(e) : 512;
-      if (os_blk != 256 && os_blk != 512) os_blk = 512;
+      os_blk = e ? atoi(e) : 1024;
+      if (os_blk != 256 && os_blk != 512 && os_blk != 1024) os_blk = 1024;
     }
     uint32_t os_tile = (uint32_t)TILE_ROUNDS * (uint32_t)os_blk;
     uint32_t nb_os = nblocks_for(n, os_tile);
@@ -2147,7 +2149,12 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
                       (void)hipEventRecord(evs[nev]); }
       const uint32_t* pbases = (const uint32_t*)((uint32_t*)gbases.p + p * RADIX);
       /* base shape only (gate above): dispatch on block size */
-      if (os_blk == 512)
+      if (os_blk == 1024)
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false, 1024>), dim3(nb_os),
+                           dim3(1024), 0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
+                           nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      else if (os_blk == 512)
         hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false, 512>), dim3(nb_os),
                            dim3(512), 0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
                            nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
