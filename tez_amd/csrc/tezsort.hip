@@ -323,14 +323,15 @@ __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
  * hist: per-block digit counts over its contiguous tile.
  * scan: off[b][d] = digit_base[d] + sum_{b'<b} cnt[b'][d]  (digit-major scan)
  * scatter: stable rank within tile via wave ballots + LDS wave histograms. */
-template <typename KeyT>
+template <typename KeyT, int BLK = BLOCK>
 __global__ void k_radix_hist(const KeyT* keys, uint32_t n, int byte_idx,
                              uint32_t* counts /* [nblocks*RADIX] */) {
+  constexpr uint32_t OT = (uint32_t)TILE_ROUNDS * BLK;
   __shared__ uint32_t h[RADIX];
   for (int i = threadIdx.x; i < RADIX; i += blockDim.x) h[i] = 0;
   __syncthreads();
-  uint32_t start = blockIdx.x * TILE;
-  uint32_t end = min(start + TILE, n);
+  uint32_t start = blockIdx.x * OT;
+  uint32_t end = min(start + OT, n);
   for (uint32_t i = start + threadIdx.x; i < end; i += blockDim.x) {
     uint32_t d = (uint32_t)(keys[i] >> (8 * byte_idx)) & 0xFF;
     atomicAdd(&h[d], 1u);
@@ -388,25 +389,27 @@ __global__ void k_radix_scan_digits(uint32_t* totals, uint32_t* bases) {
  * global writes — a direct per-element scatter write-allocates a 64B line
  * per 12B element (PMC: 22 GB written for 6 GB of payload); digit runs of
  * TILE/256 elements restore coalescing. */
-template <typename KeyT, bool HAS_A1, bool HAS_B64 = false>
-__global__ __launch_bounds__(BLOCK) void k_radix_scatter(
+template <typename KeyT, bool HAS_A1, bool HAS_B64 = false, int BLK = BLOCK>
+__global__ __launch_bounds__(BLK) void k_radix_scatter(
     const KeyT* keys_in, KeyT* keys_out,
     const uint32_t* a0_in, uint32_t* a0_out,
     const uint32_t* a1_in, uint32_t* a1_out,
     const uint64_t* b64_in, uint64_t* b64_out,
     uint32_t n, int byte_idx,
     const uint32_t* offsets, const uint32_t* bases) {
+  constexpr uint32_t OT = (uint32_t)TILE_ROUNDS * BLK;
+  constexpr int OWPB = BLK / WAVE;
   __shared__ uint32_t tilecnt[RADIX];       /* per-digit running count in tile */
-  __shared__ uint32_t wavehist[WPB][RADIX];
+  __shared__ uint32_t wavehist[OWPB][RADIX];
   __shared__ uint32_t tileoff[RADIX];       /* exclusive scan of final counts */
-  __shared__ KeyT ls_key[TILE];
-  __shared__ uint32_t ls_a0[TILE];
-  __shared__ uint32_t ls_a1[HAS_A1 ? TILE : 1];
-  __shared__ uint64_t ls_b64[HAS_B64 ? TILE : 1];
-  __shared__ uint8_t ls_dig[TILE];
+  __shared__ KeyT ls_key[OT];
+  __shared__ uint32_t ls_a0[OT];
+  __shared__ uint32_t ls_a1[HAS_A1 ? OT : 1];
+  __shared__ uint64_t ls_b64[HAS_B64 ? OT : 1];
+  __shared__ uint8_t ls_dig[OT];
   for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
-  uint32_t start = blockIdx.x * TILE;
-  uint32_t end = min(start + TILE, n);
+  uint32_t start = blockIdx.x * OT;
+  uint32_t end = min(start + OT, n);
   uint32_t count = end - start;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wv = threadIdx.x / WAVE;
@@ -418,13 +421,13 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
   uint32_t my_a1v[TILE_ROUNDS];
   uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
   int round = 0;
-  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
     uint32_t i = r0 + threadIdx.x;
     bool active = i < end;
     KeyT key = active ? keys_in[i] : (KeyT)0;
     uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0xFFFFFFFFu;
     __syncthreads();
-    for (int j = threadIdx.x; j < WPB * RADIX; j += blockDim.x)
+    for (int j = threadIdx.x; j < OWPB * RADIX; j += blockDim.x)
       ((uint32_t*)wavehist)[j] = 0;
     __syncthreads();
     uint64_t m = ~0ull;
@@ -452,7 +455,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
     __syncthreads();
     for (int j = threadIdx.x; j < RADIX; j += blockDim.x) {
       uint32_t sum = 0;
-      for (int w = 0; w < WPB; w++) sum += wavehist[w][j];
+      for (int w = 0; w < OWPB; w++) sum += wavehist[w][j];
       tilecnt[j] += sum;
     }
   }
@@ -476,7 +479,7 @@ __global__ __launch_bounds__(BLOCK) void k_radix_scatter(
   }
   /* place elements into LDS grouped by digit */
   round = 0;
-  for (uint32_t r0 = start; r0 < start + TILE; r0 += blockDim.x, round++) {
+  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
     uint32_t i = r0 + threadIdx.x;
     if (i < end) {
       uint32_t slot = tileoff[my_dig[round]] + my_seq[round];
@@ -2205,34 +2208,88 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
           if (h_cnt2[p * RADIX + d] == n) { pass_skip[p] = true; break; }
     }
   }
+  /* classic path at 512-thread blocks: 4096-element tiles keep 8 waves/CU
+     (1-2 resident blocks) while doubling digit runs; the u64-key + u64-
+     payload variant would exceed the 160 KB LDS at 512 and stays at 256
+     (it is never dispatched at runtime — b64 rides u32 seg keys). */
+  {
+    static int cl_blk = -1;
+    if (cl_blk < 0) {
+      const char* e = getenv("TZS_CL_BLK");
+      cl_blk = e ? atoi(e) : 256;   /* 512 measured neutral on the
+                                        refinement shapes (55.7 vs 56.0
+                                        GB/s C3) — they are bound by the
+                                        compact/gather pattern, not by
+                                        digit-run length */
+      if (cl_blk != 256 && cl_blk != 512) cl_blk = 256;
+    }
+    uint32_t cl_tile = (uint32_t)TILE_ROUNDS * (uint32_t)cl_blk;
+    uint32_t nb_cl = nblocks_for(n, cl_tile);
+    if (nb_cl > nb) nb_cl = nb;  /* counts/offsets were sized for nb */
+    bool use512 = (cl_blk == 512);
+    if (sizeof(KeyT) == 8 && has_b64) use512 = false;  /* LDS > 160 KB */
   for (int b = first_byte; b < nbytes_key; b++) {
     if (b - first_byte < 16 && pass_skip[b - first_byte]) continue;
-    hipLaunchKernelGGL((k_radix_hist<KeyT>), dim3(nb), dim3(BLOCK), 0, 0, kin, n, b,
-                       (uint32_t*)counts.p);
+    uint32_t nbl = use512 ? nb_cl : nb;
+    if (use512)
+      hipLaunchKernelGGL((k_radix_hist<KeyT, 512>), dim3(nbl), dim3(512), 0, 0,
+                         kin, n, b, (uint32_t*)counts.p);
+    else
+      hipLaunchKernelGGL((k_radix_hist<KeyT>), dim3(nbl), dim3(BLOCK), 0, 0, kin, n, b,
+                         (uint32_t*)counts.p);
     hipLaunchKernelGGL(k_radix_scan_blocks, dim3(RADIX), dim3(BLOCK), 0, 0,
-                       (uint32_t*)counts.p, nb, (uint32_t*)offsets.p, (uint32_t*)totals.p);
+                       (uint32_t*)counts.p, nbl, (uint32_t*)offsets.p, (uint32_t*)totals.p);
     hipLaunchKernelGGL(k_radix_scan_digits, dim3(1), dim3(RADIX), 0, 0,
                        (uint32_t*)totals.p, (uint32_t*)bases.p);
     if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                     (void)hipEventRecord(evs[nev]); }
-    if (has_b64)
-      hipLaunchKernelGGL((k_radix_scatter<KeyT, false, true>), dim3(nb), dim3(BLOCK),
-                         0, 0, kin, kout, a0in, a0out, nullptr, nullptr, b64in,
-                         b64out, n, b, (uint32_t*)offsets.p, (uint32_t*)bases.p);
-    else if (has_a1)
-      hipLaunchKernelGGL((k_radix_scatter<KeyT, true>), dim3(nb), dim3(BLOCK), 0, 0,
-                         kin, kout, a0in, a0out, a1in, a1out, nullptr, nullptr, n, b,
-                         (uint32_t*)offsets.p, (uint32_t*)bases.p);
-    else
-      hipLaunchKernelGGL((k_radix_scatter<KeyT, false>), dim3(nb), dim3(BLOCK), 0, 0,
-                         kin, kout, a0in, a0out, nullptr, nullptr, nullptr, nullptr,
-                         n, b, (uint32_t*)offsets.p, (uint32_t*)bases.p);
+    if (has_b64) {
+      if constexpr (sizeof(KeyT) == 4) {
+        if (use512) {
+          hipLaunchKernelGGL((k_radix_scatter<KeyT, false, true, 512>), dim3(nbl),
+                             dim3(512), 0, 0, kin, kout, a0in, a0out, nullptr,
+                             nullptr, b64in, b64out, n, b, (uint32_t*)offsets.p,
+                             (uint32_t*)bases.p);
+        } else {
+          hipLaunchKernelGGL((k_radix_scatter<KeyT, false, true>), dim3(nbl),
+                             dim3(BLOCK), 0, 0, kin, kout, a0in, a0out, nullptr,
+                             nullptr, b64in, b64out, n, b, (uint32_t*)offsets.p,
+                             (uint32_t*)bases.p);
+        }
+      } else {
+        hipLaunchKernelGGL((k_radix_scatter<KeyT, false, true>), dim3(nbl),
+                           dim3(BLOCK), 0, 0, kin, kout, a0in, a0out, nullptr,
+                           nullptr, b64in, b64out, n, b, (uint32_t*)offsets.p,
+                           (uint32_t*)bases.p);
+      }
+    } else if (has_a1) {
+      if (use512)
+        hipLaunchKernelGGL((k_radix_scatter<KeyT, true, false, 512>), dim3(nbl),
+                           dim3(512), 0, 0, kin, kout, a0in, a0out, a1in, a1out,
+                           nullptr, nullptr, n, b, (uint32_t*)offsets.p,
+                           (uint32_t*)bases.p);
+      else
+        hipLaunchKernelGGL((k_radix_scatter<KeyT, true>), dim3(nbl), dim3(BLOCK),
+                           0, 0, kin, kout, a0in, a0out, a1in, a1out, nullptr,
+                           nullptr, n, b, (uint32_t*)offsets.p, (uint32_t*)bases.p);
+    } else {
+      if (use512)
+        hipLaunchKernelGGL((k_radix_scatter<KeyT, false, false, 512>), dim3(nbl),
+                           dim3(512), 0, 0, kin, kout, a0in, a0out, nullptr,
+                           nullptr, nullptr, nullptr, n, b, (uint32_t*)offsets.p,
+                           (uint32_t*)bases.p);
+      else
+        hipLaunchKernelGGL((k_radix_scatter<KeyT, false>), dim3(nbl), dim3(BLOCK),
+                           0, 0, kin, kout, a0in, a0out, nullptr, nullptr, nullptr,
+                           nullptr, n, b, (uint32_t*)offsets.p, (uint32_t*)bases.p);
+    }
     if (nev < 16) { (void)hipEventRecord(eve[nev]); nev++; }
     std::swap(kin, kout);
     std::swap(a0in, a0out);
     if (has_a1) std::swap(a1in, a1out);
     if (has_b64) std::swap(b64in, b64out);
     passes++;
+  }
   }
 finish:
   (void)hipDeviceSynchronize();
