@@ -62,6 +62,42 @@ def test_c5_terasort_shape_range_partitions(engine):
     assert got == want["data"]
 
 
+def test_overflow_spill_count_coalesce(engine):
+    """More spills than the RecTable's 32-slot merge table: the flush
+    coalesces record sets (the reference reaches any count via multipass
+    merges, getPassFactor/TezMerger.java:921-931).  Unique keys => byte-exact
+    vs the oracle's merge of all 40 segments."""
+    import random
+    rng = random.Random(0x40)
+    P, nspill, per = 8, 40, 120
+    conf = engine.make_conf(P)
+    s = engine.Sorter(conf)
+    spills = []
+    seen = set()
+    for sp in range(nspill):
+        pairs = []
+        while len(pairs) < per:
+            k = bytes(rng.randrange(256) for _ in range(12))
+            if k in seen:
+                continue
+            seen.add(k)
+            pairs.append((o.serialize_bytes_writable(k),
+                          o.serialize_bytes_writable(b"v%02d.%03d" % (sp, len(pairs)))))
+        for k, v in pairs:
+            s.write(k, v, -1)
+        s.spill()
+        d, f, kl = o.build_records(pairs)
+        spills.append(o.spill(d, f, kl, P))
+    s.flush()
+    got, gidx = s.output()
+    ctr = s.counters()
+    s.close()
+    assert ctr["num_spills"] == nspill
+    want = o.final_merge(spills, P)
+    assert gidx == o.index_decode(want["index"], P)
+    assert got == want["data"]
+
+
 def test_c4_zipf_skewed_partitions(engine):
     """kind 3 (BASELINE configs[3] shape): uniform 10B keys whose PARTITION
     SIZES follow Zipf(1.0) over 199 partitions via the generator's

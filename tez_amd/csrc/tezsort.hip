@@ -3336,7 +3336,7 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
    * duplicates the SAME_KEY provenance rule is applied in k_writer_same —
    * DESIGN.md §4/§5.) */
   int nsp = (int)s->spills.size();
-  if (nsp > MAX_SPILLS) FAIL(-22, "too many spills (%d > %d)", nsp, MAX_SPILLS);
+  int orig_nsp = nsp;  /* the combiner gate follows the ORIGINAL spill count */
   /* the merge reads the columnar record sets only; the per-spill IFile
      streams are dead weight now (the reference deletes spill files after the
      final merge, PipelinedSorter.java:844-849) — free them up front so the
@@ -3344,6 +3344,56 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
   for (auto* sp2 : s->spills) {
     sp2->ifile.release();
     sp2->ifile_len = 0;
+  }
+  if (nsp > MAX_SPILLS) {
+    /* >32 spills: coalesce all spill record sets into one (the union re-sort
+       does not need pre-sorted inputs, so concatenation is enough).  The
+       coalesced set's rle flag is the OR of its parts plus 1 when several
+       parts combine — mirroring the reference's multipass intermediate
+       merges, whose segments carry SAME_KEY runs for duplicates they
+       collapsed (getPassFactor/TezMerger.java:921-931, SAME_KEY :598-653). */
+    SpillData* comb = new SpillData();
+    uint64_t cn = 0, cbytes = 0;
+    for (auto* sp2 : s->spills) {
+      cn += (uint64_t)sp2->n;
+      uint64_t b0 = 0;
+      HIP_CHECK(hipMemcpy(&b0, (const uint64_t*)sp2->off.p + sp2->n, 8,
+                          hipMemcpyDeviceToHost));
+      cbytes += b0;
+    }
+    if (cn > 4000000000ull) FAIL(-22, "coalesced spills exceed u32 record ids");
+    if (comb->data.alloc(cbytes ? cbytes : 1)) return -12;
+    if (comb->off.alloc(8 * (cn + 1))) return -12;
+    if (comb->klen.alloc(4 * (cn ? cn : 1))) return -12;
+    uint64_t rn = 0, rb = 0;
+    uint32_t ru = s->spills[0]->rec_u, ku = s->spills[0]->klen_u;
+    uint8_t crle = 1; /* conservative: intermediate-merge SAME_KEY semantics */
+    for (auto* sp2 : s->spills) {
+      uint64_t b0 = 0;
+      HIP_CHECK(hipMemcpy(&b0, (const uint64_t*)sp2->off.p + sp2->n, 8,
+                          hipMemcpyDeviceToHost));
+      if (b0)
+        HIP_CHECK(hipMemcpyAsync((uint8_t*)comb->data.p + rb, sp2->data.p, b0,
+                                 hipMemcpyDeviceToDevice));
+      hipLaunchKernelGGL(k_shift_offsets, dim3(grid1d(sp2->n + 1)), dim3(BLOCK), 0, 0,
+                         (const uint64_t*)sp2->off.p,
+                         (uint64_t*)comb->off.p + rn, rb, sp2->n + 1);
+      if (sp2->n)
+        HIP_CHECK(hipMemcpyAsync((uint32_t*)comb->klen.p + rn, sp2->klen.p,
+                                 4ull * sp2->n, hipMemcpyDeviceToDevice));
+      if (!sp2->rec_u || sp2->rec_u != ru || sp2->klen_u != ku) { ru = 0; ku = 0; }
+      rn += (uint64_t)sp2->n;
+      rb += b0;
+    }
+    HIP_CHECK(hipDeviceSynchronize());
+    comb->n = (int64_t)cn;
+    comb->rec_u = ru;
+    comb->klen_u = ku;
+    comb->rle = crle;
+    for (auto* sp2 : s->spills) { sp2->release(); delete sp2; }
+    s->spills.clear();
+    s->spills.push_back(comb);
+    nsp = 1;
   }
   RecTable rt = {};
   rt.nspills = nsp;
@@ -3370,7 +3420,7 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
     FAIL(-22, "explicit partitions require a single spill in round 1");
   SpillData finalsp;
   bool combine_at_merge = s->conf.combiner != 0 &&
-                          nsp >= s->conf.min_spills_for_combine;
+                          orig_nsp >= s->conf.min_spills_for_combine;
   rc = sort_and_emit(s, rt, (uint32_t)total_n, nullptr,
                      spill_rle.data(), nsp, &finalsp, combine_at_merge);
   if (rc) return rc;
