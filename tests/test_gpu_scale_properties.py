@@ -96,6 +96,10 @@ def test_c2_full_size_properties(engine):
         checked += 1
     assert checked == 2
     s.close()
+    # memory-health guard: the full-size C2 flow must never hit the pool's
+    # drop-and-retry path (hipMalloc churn — DESIGN.md C3 memory note)
+    ps = engine.pool_stats()
+    assert ps["drops"] == 0, ps
 
 
 def test_merge_segments_cabi(engine):
