@@ -204,6 +204,10 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     n_gpus = max(args.gpus, world)
+    if args.gpus > 1 and world == 1:
+        sys.exit("--gpus N>1 must be launched via torch.distributed.run "
+                 "(one rank per GPU); a single-process run would report an "
+                 "N-inflated value")
 
     import __graft_entry__
     __graft_entry__.build()
