@@ -120,8 +120,7 @@ def run_step_multi(tez_amd, rank, world, device, d, off, kl, n, adopt=False,
     d_data, d_off2, d_klen2, rec_ranges, byte_ranges = m.sorted_columnar()
     plan = ex.plan_send(rec_ranges, byte_ranges, world)
     sd, sr, sk = ex.pack_send_tensors(m, plan, device)
-    rd, rrl, rkl = ex.exchange(plan, sd, sr, sk)
-    rparts = ex.exchange_parts(plan, nparts, device)
+    rd, rrl, rkl, rparts = ex.exchange(plan, sd, sr, sk, nparts=nparts)
     red = ex.reduce_merge(lambda: tez_amd.Sorter(tez_amd.make_conf(nparts)),
                           rd, rrl, rkl, rparts)
     tms = red.times()
@@ -271,8 +270,13 @@ def main():
             klen_w, vlen_w, kind_w, parts_w = KLEN, VLEN, 0, PARTS
         rec_ser_w = 4 + klen_w + 4 + vlen_w
         total_sets = args.warmup + args.steps
+        # 230 GB cap: pre-generated inputs + ~15 GB step working set must
+        # stay under the touched-page capacity of the 288 GB HBM (hipMalloc
+        # overcommits silently; first touches past capacity fault at
+        # page-migration speed — DESIGN §7a).  21 sets of the C2 shape
+        # (driver --steps 20 --warmup 1) = 185 GB: adopt survives it.
         adopt = (not args.no_adopt
-                 and total_sets * n_local * rec_ser_w <= 120e9)
+                 and total_sets * n_local * rec_ser_w <= 230e9)
         if adopt:
             input_sets = []
             for sidx in range(total_sets):

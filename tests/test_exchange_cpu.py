@@ -80,6 +80,14 @@ def _worker(rank, world, tmpdir):
         for p in ex.parts_for_dest(P, world, rank):
             want_parts += [p] * (src + 1 + p)
     assert rparts.tolist() == want_parts, f"rank {rank} parts mismatch"
+    # fused form: identical results with ONE meta all_gather
+    # + two all_to_all collectives (packed lengths, data)
+    rd2, rrl2, rkl2, rp2 = ex.exchange(plan, send_data, send_reclen, send_klen,
+                                       nparts=P)
+    assert bytes(rd2.numpy().tobytes()) == want
+    assert rrl2.tolist() == want_lens
+    assert rkl2.tolist() == [1] * len(want_lens)
+    assert rp2.tolist() == want_parts, f"rank {rank} fused parts mismatch"
     dist.destroy_process_group()
 
 

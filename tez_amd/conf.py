@@ -46,7 +46,16 @@ def conf_from_tez_properties(props: dict, num_partitions: int):
                          "supports BytesWritable and Text)")
     key_type, cmp_default = _KEY_CLASSES[key_cls]
     cmp_cls = props.get("tez.runtime.key.comparator.class")
-    comparator = _COMPARATORS.get(cmp_cls, cmp_default) if cmp_cls else cmp_default
+    if cmp_cls:
+        if cmp_cls not in _COMPARATORS:
+            # mirror the key-class handling: a silently-wrong sort order is
+            # worse than an error
+            raise ValueError(f"unsupported key.comparator.class {cmp_cls} "
+                             "(engine supports TezBytesComparator and "
+                             "Text.Comparator)")
+        comparator = _COMPARATORS[cmp_cls]
+    else:
+        comparator = cmp_default
 
     combiner = 0
     comb_cls = props.get("tez.runtime.combiner.class")

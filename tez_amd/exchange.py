@@ -52,30 +52,86 @@ def plan_send(rec_ranges, byte_ranges, world: int) -> SendPlan:
     return SendPlan(order, byte_splits, rec_splits, seg_bytes, seg_recs)
 
 
-def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None):
-    """all-to-all-v of (data bytes, per-record lengths, per-record klens).
-    send_* are torch tensors already laid out in plan order (uint8/u32/u32).
-    Returns (recv_data, recv_reclen, recv_klen) tensors."""
+def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None,
+             nparts=None):
+    """all-to-all-v of (data bytes, per-record lengths+klens).  send_* are
+    torch tensors already laid out in plan order (uint8/i32/i32).
+
+    Exactly TWO data-path collectives per step: one all_to_all_single for the
+    record bytes and one for the per-record (reclen ‖ klen) lengths packed
+    into a single i32 tensor (per-dest chunk = reclens then klens).  The
+    size matrix rides ONE small all_gather up front (the 24B index triples'
+    counts — SURVEY §8e); when ``nparts`` is given the per-partition record
+    counts ride the same all_gather and the received records' original
+    partition ids are reconstructed here (what exchange_parts previously did
+    with its own collective).
+
+    Returns (recv_data, recv_reclen, recv_klen) or, with nparts,
+    (recv_data, recv_reclen, recv_klen, recv_parts)."""
     import torch
     import torch.distributed as dist
     world = dist.get_world_size(group)
+    me = dist.get_rank(group)
+    dev = send_data.device
     in_b = plan.byte_splits
     in_r = plan.rec_splits
-    # exchange split sizes first (the size-matrix all-gather; the 24B index
-    # triples' byte counts — SURVEY §8e)
-    my_sizes = torch.tensor([in_b, in_r], dtype=torch.int64,
-                            device=send_data.device)
-    all_sizes = [torch.empty_like(my_sizes) for _ in range(world)]
-    dist.all_gather(all_sizes, my_sizes, group=group)
-    out_b = [int(all_sizes[src][0][dist.get_rank(group)]) for src in range(world)]
-    out_r = [int(all_sizes[src][1][dist.get_rank(group)]) for src in range(world)]
-    recv_data = torch.empty(sum(out_b), dtype=torch.uint8, device=send_data.device)
-    recv_reclen = torch.empty(sum(out_r), dtype=torch.int32, device=send_data.device)
-    recv_klen = torch.empty(sum(out_r), dtype=torch.int32, device=send_data.device)
+    P = int(nparts) if nparts is not None else 0
+    counts = [0] * P
+    if P:
+        for i, p in enumerate(plan.order):
+            counts[p] = plan.seg_recs[i]
+    my_meta = torch.tensor(in_b + in_r + counts, dtype=torch.int64, device=dev)
+    all_meta = [torch.empty_like(my_meta) for _ in range(world)]
+    dist.all_gather(all_meta, my_meta, group=group)
+    all_meta = [m.tolist() for m in all_meta]
+    out_b = [int(all_meta[src][me]) for src in range(world)]
+    out_r = [int(all_meta[src][world + me]) for src in range(world)]
+    # collective 1: packed lengths — per-dest chunk [reclen_d ‖ klen_d]
+    total_in_r = sum(in_r)
+    total_out_r = sum(out_r)
+    send_len = torch.empty(2 * total_in_r, dtype=torch.int32, device=dev)
+    pos = rpos = 0
+    for d in range(world):
+        r = in_r[d]
+        if r:
+            send_len[pos:pos + r] = send_reclen[rpos:rpos + r]
+            send_len[pos + r:pos + 2 * r] = send_klen[rpos:rpos + r]
+        pos += 2 * r
+        rpos += r
+    recv_len = torch.empty(2 * total_out_r, dtype=torch.int32, device=dev)
+    dist.all_to_all_single(recv_len, send_len, [2 * x for x in out_r],
+                           [2 * x for x in in_r], group=group)
+    recv_reclen = torch.empty(total_out_r, dtype=torch.int32, device=dev)
+    recv_klen = torch.empty(total_out_r, dtype=torch.int32, device=dev)
+    pos = rpos = 0
+    for src in range(world):
+        r = out_r[src]
+        if r:
+            recv_reclen[rpos:rpos + r] = recv_len[pos:pos + r]
+            recv_klen[rpos:rpos + r] = recv_len[pos + r:pos + 2 * r]
+        pos += 2 * r
+        rpos += r
+    # collective 2: the record bytes
+    recv_data = torch.empty(sum(out_b), dtype=torch.uint8, device=dev)
     dist.all_to_all_single(recv_data, send_data, out_b, in_b, group=group)
-    dist.all_to_all_single(recv_reclen, send_reclen, out_r, in_r, group=group)
-    dist.all_to_all_single(recv_klen, send_klen, out_r, in_r, group=group)
-    return recv_data, recv_reclen, recv_klen
+    if nparts is None:
+        return recv_data, recv_reclen, recv_klen
+    # reconstruct received records' partition ids in arrival order
+    # (src rank asc, then owned partitions asc — all_to_all concatenation)
+    ids, cnts = [], []
+    for src in range(world):
+        for p in range(P):
+            c = int(all_meta[src][2 * world + p])
+            if p % world == me and c:
+                ids.append(p)
+                cnts.append(c)
+    if not ids:
+        recv_parts = torch.empty(0, dtype=torch.int32, device=dev)
+    else:
+        recv_parts = torch.repeat_interleave(
+            torch.tensor(ids, dtype=torch.int32, device=dev),
+            torch.tensor(cnts, dtype=torch.int64, device=dev))
+    return recv_data, recv_reclen, recv_klen, recv_parts
 
 
 def pack_send_tensors(sorter, plan: SendPlan, device):

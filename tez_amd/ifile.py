@@ -10,6 +10,8 @@ import zlib
 
 
 def vint_read(b, pos):
+    if pos >= len(b):
+        raise ValueError("truncated IFile segment (vint past end)")
     first = b[pos]
     first_s = first - 256 if first >= 128 else first
     if first_s >= -112:
@@ -73,9 +75,15 @@ def read_stream(stream, with_header=True, verify_crc=True):
         if klen == -1 and vlen == -1:
             break
         same = klen == -2
+        if klen < -1 and not same:
+            raise ValueError(f"corrupt IFile segment (klen marker {klen})")
         if not same:
+            if klen < 0 or pos + klen > len(body):
+                raise ValueError("truncated IFile segment (key past end)")
             cur_key = bytes(body[pos: pos + klen])
             pos += klen
+        if vlen < 0 or pos + vlen > len(body):
+            raise ValueError("truncated IFile segment (value past end)")
         val = bytes(body[pos: pos + vlen])
         pos += vlen
         out.append((cur_key, val, same))

@@ -43,7 +43,7 @@ class OrderedGroupedKVInput:
             comparator=_engine.CMP_TEXT if self._text_keys else _engine.CMP_TEZBYTES)
         s = _engine.Sorter(conf)
         for k, v in pairs:
-            s.write(k, v, 0)
+            s.write(k, v, -1)  # hash%1 == 0; avoids the explicit-partition path
         s.flush()
         data, index = s.output()
         s.close()

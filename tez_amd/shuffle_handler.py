@@ -91,14 +91,21 @@ class ShuffleHandlerServer:
                 chunks = []
                 try:
                     for map_id in maps:
+                        # a client-supplied id must not escape local_dir
+                        # (the reference resolves ids through the NM's
+                        # path allowlist)
+                        if "/" in map_id or "\\" in map_id or ".." in map_id:
+                            return self._err(400, "Invalid map id")
                         base = f"{outer.local_dir}/output/{map_id}/file.out"
                         idx = decode_index(open(base + ".index", "rb").read())
-                        data = open(base, "rb").read()
                         chunks.append(ifile.vint_write(r1 - r0 + 1))
-                        for r in range(r0, r1 + 1):
-                            st, raw, cl = idx[r]
-                            chunks.append(encode_shuffle_header(map_id, cl, raw, r))
-                            chunks.append(data[st:st + cl])
+                        with open(base, "rb") as f:
+                            for r in range(r0, r1 + 1):
+                                st, raw, cl = idx[r]
+                                chunks.append(
+                                    encode_shuffle_header(map_id, cl, raw, r))
+                                f.seek(st)
+                                chunks.append(f.read(cl))
                 except FileNotFoundError:
                     return self._err(404, "map output not found")
                 body = b"".join(chunks)
