@@ -122,7 +122,9 @@ def run_step_multi(tez_amd, rank, world, device, d, off, kl, n, adopt=False,
     sd, sr, sk = ex.pack_send_tensors(m, plan, device)
     rd, rrl, rkl, rparts = ex.exchange(plan, sd, sr, sk, nparts=nparts)
     red = ex.reduce_merge(lambda: tez_amd.Sorter(tez_amd.make_conf(nparts)),
-                          rd, rrl, rkl, rparts)
+                          rd, rrl, rkl, rparts,
+                          src_rec_splits=plan.recv_rec_splits,
+                          src_byte_splits=plan.recv_byte_splits)
     tms = red.times()
     red.close()
     m.close()

@@ -117,6 +117,20 @@ int tzs_sorter_write_batch_device_adopt(tzs_sorter* s, void* d_data,
  * (PipelinedSorter.spill, PipelinedSorter.java:559-648). Returns spill id >= 0. */
 int tzs_sorter_spill(tzs_sorter* s);
 
+/* Reduce-side ingestion of an ALREADY-SORTED columnar segment (the
+ * MergeManager admission path, MergeManager.java:423-519: fetched segments
+ * arrive sorted by the map side).  Registers the caller's device buffers as
+ * one sorted spill WITHOUT copying or re-sorting them (the caller keeps
+ * them alive until flush/close); flush() then runs the k-way merge-path
+ * merge over all segments (TezMerger.MergeQueue, TezMerger.java:466-706).
+ * d_part = per-record original partition ids (explicit partitioners ride
+ * through the exchange), or NULL to recompute HashPartitioner placement.
+ * Returns the segment id (>=0) or <0 on error; -22 if the segment is not
+ * sorted by (partition, key). */
+int tzs_sorter_add_sorted_segment(tzs_sorter* s, const void* d_data,
+                                  const uint64_t* d_off, const uint32_t* d_klen,
+                                  const int32_t* d_part, int64_t n);
+
 /* flush(): final spill + final merge (PipelinedSorter.flush, :665-851).
  * After this, final output/index are available. Blocking. */
 int tzs_sorter_flush(tzs_sorter* s);
@@ -201,6 +215,7 @@ typedef struct tzs_times {
   int64_t absorb_ns, composite_ns, sort_ns, permute_ns, emit_ns, crc_ns, dominant_kernel_elems, total_ns;
   int64_t sort_passes;        /* radix passes actually run (incl. refinement) */
   int64_t dominant_kernel_ns; /* scatter total */
+  int64_t merge_ns;           /* flush merge-path tree (k-way spill merge) */
 } tzs_times;
 int tzs_sorter_times(const tzs_sorter* s, tzs_times* out);
 

@@ -57,7 +57,7 @@ class TzsTimes(ctypes.Structure):
     _fields_ = [(n, ctypes.c_int64) for n in
                 ("absorb_ns", "composite_ns", "sort_ns", "permute_ns", "emit_ns",
                  "crc_ns", "dominant_kernel_elems", "total_ns", "sort_passes",
-                 "dominant_kernel_ns")]
+                 "dominant_kernel_ns", "merge_ns")]
 
 
 _lib = None
@@ -94,6 +94,8 @@ def lib():
                                                       c.c_void_p, c.c_void_p,
                                                       c.c_void_p, c.c_int64]
     L.tzs_sorter_spill.argtypes = [c.c_void_p]
+    L.tzs_sorter_add_sorted_segment.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p,
+                                                c.c_void_p, c.c_void_p, c.c_int64]
     L.tzs_sorter_flush.argtypes = [c.c_void_p]
     L.tzs_sorter_num_spills.argtypes = [c.c_void_p]
     L.tzs_sorter_output.argtypes = [c.c_void_p, c.POINTER(c.c_void_p),
@@ -172,6 +174,13 @@ class Sorter:
 
     def spill(self):
         return _ck(lib().tzs_sorter_spill(self.h), "spill")
+
+    def add_sorted_segment(self, d_data, d_off, d_klen, d_part, n):
+        """Ingest an already-sorted columnar segment without copy or re-sort
+        (reduce-side MergeManager admission).  Caller keeps the device
+        buffers alive until flush()/close()."""
+        return _ck(lib().tzs_sorter_add_sorted_segment(
+            self.h, d_data, d_off, d_klen, d_part, n), "add_sorted_segment")
 
     def flush(self):
         _ck(lib().tzs_sorter_flush(self.h), "flush")

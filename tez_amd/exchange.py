@@ -114,6 +114,10 @@ def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None,
     # collective 2: the record bytes
     recv_data = torch.empty(sum(out_b), dtype=torch.uint8, device=dev)
     dist.all_to_all_single(recv_data, send_data, out_b, in_b, group=group)
+    # per-source chunk boundaries (lets reduce_merge ingest each source's
+    # chunk as one pre-sorted segment — no re-sort on the reduce side)
+    plan.recv_rec_splits = out_r
+    plan.recv_byte_splits = out_b
     if nparts is None:
         return recv_data, recv_reclen, recv_klen
     # reconstruct received records' partition ids in arrival order
@@ -202,12 +206,20 @@ def exchange_parts(plan: SendPlan, P: int, device, group=None):
 
 
 def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen,
-                 recv_parts=None):
+                 recv_parts=None, src_rec_splits=None, src_byte_splits=None):
     """Feed received columnar records into a reduce-side sorter (the
     MergeManager/TezMerger replacement) and flush: the final merged IFile for
-    this rank's owned partitions.  recv_parts (from exchange_parts) carries
-    the original partition ids; without it the sorter recomputes
-    HashPartitioner placement (wrong for explicit partitioners)."""
+    this rank's owned partitions.  recv_parts carries the original partition
+    ids; without it the sorter recomputes HashPartitioner placement (wrong
+    for explicit partitioners).
+
+    With src_rec_splits/src_byte_splits (exchange() records them on the plan
+    as recv_rec_splits/recv_byte_splits), each source rank's chunk — already
+    sorted by (partition, key) by the map side — is ingested as one
+    PRE-SORTED segment and flush runs the k-way merge-path merge over the
+    segments, skipping the reduce-side re-sort (MergeManager admission +
+    TezMerger merge, MergeManager.java:423-519, TezMerger.java:466-706).
+    Without splits it falls back to the unsorted single-batch path."""
     import torch
     n = int(recv_reclen.numel())
     sorter = conf_factory()
@@ -215,13 +227,33 @@ def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen,
         off = torch.zeros(n + 1, dtype=torch.int64, device=recv_data.device)
         torch.cumsum(recv_reclen.to(torch.int64), 0, out=off[1:])
         klen_u32 = recv_klen.contiguous()
-        parts_ptr = None
         parts_t = None
         if recv_parts is not None and int(recv_parts.numel()) == n:
             parts_t = recv_parts.contiguous()
-            parts_ptr = parts_t.data_ptr()
-        sorter.write_batch_device(recv_data.data_ptr(), off.data_ptr(),
-                                  klen_u32.data_ptr(), parts_ptr, n)
-        sorter._exchange_keepalive = (recv_data, off, klen_u32, parts_t)
+        if src_rec_splits is not None:
+            r0 = 0
+            b0 = 0
+            keep = []
+            for src, r in enumerate(src_rec_splits):
+                if r == 0:
+                    continue
+                b = (int(src_byte_splits[src]) if src_byte_splits is not None
+                     else int(off[r0 + r] - off[r0]))
+                seg_off = (off[r0:r0 + r + 1] - off[r0]).contiguous()
+                keep.append(seg_off)
+                sorter.add_sorted_segment(
+                    recv_data.data_ptr() + b0,
+                    seg_off.data_ptr(),
+                    klen_u32.data_ptr() + 4 * r0,
+                    parts_t.data_ptr() + 4 * r0 if parts_t is not None else None,
+                    r)
+                r0 += r
+                b0 += b
+            sorter._exchange_keepalive = (recv_data, off, klen_u32, parts_t, keep)
+        else:
+            parts_ptr = parts_t.data_ptr() if parts_t is not None else None
+            sorter.write_batch_device(recv_data.data_ptr(), off.data_ptr(),
+                                      klen_u32.data_ptr(), parts_ptr, n)
+            sorter._exchange_keepalive = (recv_data, off, klen_u32, parts_t)
     sorter.flush()
     return sorter
