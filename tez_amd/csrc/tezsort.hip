@@ -2913,6 +2913,7 @@ static int sort_and_emit(tzs_sorter* s, HostRT& hrt, uint32_t n,
                          SpillData* outsp, bool apply_combine = false,
                          SpillData* retain = nullptr) {
   tzs_times& T = s->times;
+  g_scatter_ns = 0; g_scatter_launches = 0; g_scatter_elems = 0;
   hipEvent_t ev[4];
   for (auto& e : ev) (void)hipEventCreate(&e);
   (void)hipEventRecord(ev[0]);
@@ -2959,7 +2960,6 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                            SpillData* outsp, bool apply_combine,
                            SpillData* retain) {
   tzs_times& T = s->times;
-  g_scatter_ns = 0; g_scatter_launches = 0; g_scatter_elems = 0;
   hipEvent_t ev[10];
   for (auto& e : ev) (void)hipEventCreate(&e);
   int P = s->conf.num_partitions;
@@ -3719,6 +3719,7 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
     if (rc) return rc;
   } else {
     auto t0 = std::chrono::steady_clock::now();
+    g_scatter_ns = 0; g_scatter_launches = 0; g_scatter_elems = 0;
     /* common composite parameters across spills */
     SortParams prm = derive_sort_params(s, hrt.segs, (uint32_t)total_n);
     int common_ser = prm.ser_mode;
