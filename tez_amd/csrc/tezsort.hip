@@ -821,6 +821,16 @@ __global__ __launch_bounds__(BLOCK) void k_scan_lookback(
 
 /* ---- refinement ---- */
 /* eq[i] = 1 if sorted element i has same (still-ambiguous) key prefix as i-1 */
+__global__ void k_eq_init2(const uint64_t* skeys, const uint64_t* lo,
+                           uint8_t* eq, int pbits, uint32_t* parts,
+                           uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    uint64_t k = skeys[i];
+    eq[i] = (i > 0 && k == skeys[i - 1] && lo[i] == lo[i - 1]) ? 1 : 0;
+    parts[i] = pbits ? (uint32_t)(k >> (64 - pbits)) : 0;
+  }
+}
 __global__ void k_eq_init(const uint64_t* skeys, uint8_t* eq, int pbits,
                           uint32_t* parts, uint32_t n) {
   /* also materializes the per-position partition ids (composite top bits) —
@@ -1149,6 +1159,110 @@ __global__ void k_apply_leaf(const uint64_t* k, const uint32_t* p, uint32_t n,
     ko[i] = k[i] & mask;
     po[i] = p[i] + add;
   }
+}
+__global__ void k_apply_leaf2(const uint64_t* k, const uint64_t* lo,
+                              const uint32_t* p, uint32_t n,
+                              uint64_t mask, uint32_t add,
+                              uint64_t* ko, uint64_t* loo, uint32_t* po) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    ko[i] = k[i] & mask;
+    loo[i] = lo[i];
+    po[i] = p[i] + add;
+  }
+}
+
+/* 128-bit merge path: key = (hi & mask, lo).  Each spill retains lo = the
+ * next 8 comparator-source bytes past the composite (zero-padded), so the
+ * merged order resolves ~15 leading key bytes and the flush refinement only
+ * sees genuinely long shared prefixes (C3's Zipf words were re-refined over
+ * the whole union otherwise — the round-2 profile's dominant block). */
+#define MP2_IPT 8
+#define MP2_TILE (MP2_IPT * MP_BLOCK)
+__device__ __forceinline__ uint32_t d_mp2_diag(
+    const uint64_t* ka, const uint64_t* la, uint64_t maskA, uint32_t na,
+    const uint64_t* kb, const uint64_t* lb, uint64_t maskB, uint32_t nb,
+    uint64_t D) {
+  uint32_t lo_ = (D > nb) ? (uint32_t)(D - nb) : 0;
+  uint32_t hi_ = (D < na) ? (uint32_t)D : na;
+  while (lo_ < hi_) {
+    uint32_t mid = lo_ + ((hi_ - lo_) >> 1);
+    uint64_t ah = ka[mid] & maskA, bh = kb[D - 1 - mid] & maskB;
+    bool le = (ah < bh) || (ah == bh && la[mid] <= lb[D - 1 - mid]);
+    if (le) lo_ = mid + 1;
+    else hi_ = mid;
+  }
+  return lo_;
+}
+__global__ __launch_bounds__(MP_BLOCK) void k_merge_path2(
+    const uint64_t* ka, const uint64_t* la, const uint32_t* pa, uint32_t na,
+    uint64_t maskA, uint32_t addA,
+    const uint64_t* kb, const uint64_t* lb, const uint32_t* pb, uint32_t nb,
+    uint64_t maskB, uint32_t addB,
+    uint64_t* kout, uint64_t* loout, uint32_t* pout) {
+  __shared__ uint64_t ls_k[MP2_TILE];
+  __shared__ uint64_t ls_l[MP2_TILE];
+  __shared__ uint32_t ls_p[MP2_TILE];
+  __shared__ uint32_t s_sp[2];
+  uint64_t total = (uint64_t)na + nb;
+  uint64_t D0 = (uint64_t)blockIdx.x * MP2_TILE;
+  if (D0 >= total) return;
+  uint64_t D1 = min(D0 + (uint64_t)MP2_TILE, total);
+  if (threadIdx.x < 2) {
+    uint64_t D = threadIdx.x ? D1 : D0;
+    s_sp[threadIdx.x] = d_mp2_diag(ka, la, maskA, na, kb, lb, maskB, nb, D);
+  }
+  __syncthreads();
+  uint32_t a0 = s_sp[0], a1 = s_sp[1];
+  uint32_t b0 = (uint32_t)(D0 - a0), b1 = (uint32_t)(D1 - a1);
+  uint32_t nA = a1 - a0, nB = b1 - b0;
+  for (uint32_t i = threadIdx.x; i < nA; i += blockDim.x) {
+    ls_k[i] = ka[a0 + i] & maskA;
+    ls_l[i] = la[a0 + i];
+    ls_p[i] = pa[a0 + i] + addA;
+  }
+  for (uint32_t i = threadIdx.x; i < nB; i += blockDim.x) {
+    ls_k[nA + i] = kb[b0 + i] & maskB;
+    ls_l[nA + i] = lb[b0 + i];
+    ls_p[nA + i] = pb[b0 + i] + addB;
+  }
+  __syncthreads();
+  uint32_t r = threadIdx.x * MP2_IPT;
+  uint32_t tile_n = (uint32_t)(D1 - D0);
+  if (r >= tile_n) return;
+  uint32_t cnt = min(tile_n - r, (uint32_t)MP2_IPT);
+  uint32_t lo_ = (r > nB) ? r - nB : 0, hi_ = min(r, nA);
+  while (lo_ < hi_) {
+    uint32_t mid = lo_ + ((hi_ - lo_) >> 1);
+    bool le = (ls_k[mid] < ls_k[nA + r - 1 - mid]) ||
+              (ls_k[mid] == ls_k[nA + r - 1 - mid] &&
+               ls_l[mid] <= ls_l[nA + r - 1 - mid]);
+    if (le) lo_ = mid + 1;
+    else hi_ = mid;
+  }
+  uint32_t i = lo_, j = r - lo_;
+  uint64_t ok[MP2_IPT], ol[MP2_IPT];
+  uint32_t op[MP2_IPT];
+  #pragma unroll
+  for (uint32_t t = 0; t < MP2_IPT; t++) {
+    if (t < cnt) {
+      bool takeA = (j >= nB) ||
+                   (i < nA && (ls_k[i] < ls_k[nA + j] ||
+                               (ls_k[i] == ls_k[nA + j] &&
+                                ls_l[i] <= ls_l[nA + j])));
+      uint32_t src = takeA ? i++ : nA + (j++);
+      ok[t] = ls_k[src];
+      ol[t] = ls_l[src];
+      op[t] = ls_p[src];
+    }
+  }
+  #pragma unroll
+  for (uint32_t t = 0; t < MP2_IPT; t++)
+    if (t < cnt) {
+      kout[D0 + r + t] = ok[t];
+      loout[D0 + r + t] = ol[t];
+      pout[D0 + r + t] = op[t];
+    }
 }
 
 /* ---- emit ---- */
@@ -2531,10 +2645,14 @@ struct SpillData {
   uint8_t rle = 0;
   uint32_t rec_u = 0, klen_u = 0;
   /* retained sorted state: skey[i] = masked composite of the i-th sorted
-     record, sidxb[i] = its ORIGINAL local record index.  Lets flush merge
-     already-sorted spills (merge path) instead of re-sorting the union. */
+     record, skey2[i] = the next 8 comparator-source bytes (zero-padded)
+     from byte lo_c0, sidxb[i] = its ORIGINAL local record index.  Lets
+     flush merge already-sorted spills (128-bit merge path) instead of
+     re-sorting the union. */
   DBuf skey;   /* u64 [n] */
+  DBuf skey2;  /* u64 [n] */
   DBuf sidxb;  /* u32 [n] */
+  int lo_c0 = -1;
   int sort_sb = 0;        /* composite mask width (bytes) */
   int sort_ser_mode = 0;  /* 1 = serialized-byte composite (TezBytes var-len) */
   uint8_t sorted_valid = 0;
@@ -2546,7 +2664,7 @@ struct SpillData {
   const uint32_t* xklen = nullptr;
   void release() {
     data.release(); off.release(); klen.release(); ifile.release();
-    skey.release(); sidxb.release();
+    skey.release(); skey2.release(); sidxb.release();
   }
   /* emitted IFile bytes + host index */
   DBuf ifile;
@@ -2591,7 +2709,7 @@ struct tzs_sorter {
   tzs_times times = {};
   /* scratch kept across calls */
   DBuf skey, sidx, eq, same, sizes, scan, parts_sorted;
-  DBuf mkey2, midx2;                       /* merge-tree ping buffers */
+  DBuf skey_lo;                            /* merged lo keys (flush merge) */
   /* final-sort metadata for the exchange path */
   std::vector<uint64_t> final_rec_ranges;  /* [P+1] record index ranges */
   HostRT final_hrt;  /* multi-segment flush tables: owns the device SegDesc
@@ -2904,7 +3022,7 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                            RecTable rt, uint32_t n, int SB, int ser_mode,
                            const uint8_t* h_spill_rle, int nspills_rle,
                            SpillData* outsp, bool apply_combine,
-                           SpillData* retain);
+                           SpillData* retain, const uint64_t* d_lo);
 
 /* ---- the core: sort current buffer + emit IFile segments ---- */
 static int sort_and_emit(tzs_sorter* s, HostRT& hrt, uint32_t n,
@@ -2946,7 +3064,7 @@ static int sort_and_emit(tzs_sorter* s, HostRT& hrt, uint32_t n,
   for (auto& e : ev) (void)hipEventDestroy(e);
   return refine_and_emit(s, hrt.segs, hrt.base, rt, n, SB, ser_mode,
                          h_spill_rle, nspills_rle, outsp, apply_combine,
-                         retain);
+                         retain, nullptr);
 }
 
 /* refinement + combiner + IFile emit + CRC over an already-ordered view:
@@ -2958,7 +3076,7 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                            RecTable rt, uint32_t n, int SB, int ser_mode,
                            const uint8_t* h_spill_rle, int nspills_rle,
                            SpillData* outsp, bool apply_combine,
-                           SpillData* retain) {
+                           SpillData* retain, const uint64_t* d_lo) {
   tzs_times& T = s->times;
   hipEvent_t ev[10];
   for (auto& e : ev) (void)hipEventCreate(&e);
@@ -2977,8 +3095,12 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
   if (s->eq.alloc(n)) return -12;
   if (s->parts_sorted.alloc(sizeof(uint32_t) * n)) return -12;
   uint8_t* d_eq = (uint8_t*)s->eq.p;
-  hipLaunchKernelGGL(k_eq_init, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, d_eq,
-                     pbits, (uint32_t*)s->parts_sorted.p, n);
+  if (d_lo)
+    hipLaunchKernelGGL(k_eq_init2, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key,
+                       d_lo, d_eq, pbits, (uint32_t*)s->parts_sorted.p, n);
+  else
+    hipLaunchKernelGGL(k_eq_init, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_key, d_eq,
+                       pbits, (uint32_t*)s->parts_sorted.p, n);
   /* max content length: for refinement level count */
   /* refinement start byte within the comparator's byte source: sorted-covered
      bits are 8*SB - pbits (content mode) or 8*SB - pbits - 24 past the proxy
@@ -2986,6 +3108,8 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
      bytes repeat the content after the 4B length) */
   int c0 = ser_mode ? (8 * SB - pbits - proxy_w) / 8 : (8 * SB - pbits) / 8;
   if (c0 < 0) c0 = 0;
+  /* the merged-lo entry path already compared 8 more source bytes */
+  int c0_eff = d_lo ? c0 + 8 : c0;
   /* determine max clen lazily: use a safe cap by scanning klen on host?  We
      compute it from the conf: key_type BYTES => clen = klen-4 (max over
      spills).  For TEXT, clen <= klen-1.  Host keeps max_klen per spill. */
@@ -3020,12 +3144,12 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
     }
   }
   int max_clen = (int)max_klen; /* upper bound on content length */
-  static thread_local DBuf inrun, inrun_scan, eqcnt;
+  static thread_local DBuf inrun_scan, eqcnt;
   static thread_local DBuf lkey, seg, pos, slotpos;
   if (eqcnt.alloc(4)) return -12;
 
   int level = 1;
-  const int max_levels = (max_clen > c0) ? (max_clen - c0 + 7) / 8 : 0;
+  const int max_levels = (max_clen > c0_eff) ? (max_clen - c0_eff + 7) / 8 : 0;
   /* initial ambiguity count; later levels get it from k_eq_update's
      survivor counter (saves a full-n read per level) */
   uint32_t neq = 0;
@@ -3038,12 +3162,14 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
   for (int li = 0; li <= max_levels; li++) {
     if (neq == 0) break;
     int use_len = (li == max_levels); /* final tiebreak: content length */
-    if (inrun.alloc(sizeof(uint64_t) * n)) return -12;
     if (inrun_scan.alloc(sizeof(uint64_t) * n)) return -12;
     hipLaunchKernelGGL(k_run_flags, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq,
-                       (uint64_t*)inrun.p, n);
+                       (uint64_t*)inrun_scan.p, n);
     uint64_t ptotal = 0;
-    if (scan_u64((uint64_t*)inrun.p, (uint64_t*)inrun_scan.p, n, &ptotal)) return -12;
+    /* in-place scan: halves the refinement's largest scratch (8n bytes) —
+       the C3 1e9 shape was pool-drop bound (DESIGN 7a memory rule) */
+    if (scan_u64((uint64_t*)inrun_scan.p, (uint64_t*)inrun_scan.p, n, &ptotal))
+      return -12;
     uint64_t nruns = ptotal & 0xFFFFFFFFu;
     uint32_t m = (uint32_t)(ptotal >> 32);
     if (m == 0) break;
@@ -3051,7 +3177,7 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
     if (seg.alloc(sizeof(uint32_t) * m)) return -12;
     if (pos.alloc(sizeof(uint32_t) * m)) return -12;
     if (slotpos.alloc(sizeof(uint32_t) * m)) return -12;
-    int lb0 = c0 + 8 * li;
+    int lb0 = c0_eff + 8 * li;
     {
       /* dense original-order level-key build when most records are still
          ambiguous (m >= ~n/2): coalesced reads beat the random gather.
@@ -3105,7 +3231,7 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
   /* large-n headroom: refinement scratch and radix ping-pong temps are dead
      from here; return them to the pool before the output-stream allocation
      (at C3's 1e9 records these hold ~60 GB) */
-  inrun.release(); inrun_scan.release();
+  inrun_scan.release();
   lkey.release(); seg.release(); pos.release(); slotpos.release();
   radix_release_temps();
 
@@ -3119,6 +3245,19 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
     if (retain->sidxb.alloc(4ull * n)) return -12;
     HIP_CHECK(hipMemcpyAsync(retain->sidxb.p, s->sidx.p, 4ull * n,
                              hipMemcpyDeviceToDevice));
+    /* lo keys: comparator-source bytes [c0, c0+8) per sorted record —
+       dense build (coalesced) + one 8B gather into sorted order */
+    if (retain->skey2.alloc(8ull * n)) return -12;
+    {
+      static thread_local DBuf lo_tmp;
+      if (lo_tmp.alloc(8ull * n)) return -12;
+      hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                         c0, 0, ser_mode, (uint64_t*)lo_tmp.p, n);
+      hipLaunchKernelGGL(k_gather_merge_hi, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                         (const uint64_t*)lo_tmp.p, (const uint32_t*)s->sidx.p,
+                         (uint64_t*)retain->skey2.p, 0ull, n);
+    }
+    retain->lo_c0 = c0;
     retain->sort_sb = SB;
     retain->sort_ser_mode = ser_mode;
     retain->sorted_valid = 1;
@@ -3196,6 +3335,10 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                          rt, (const int32_t*)s->parts_sorted.p, P, pbits,
                          ref_pb, SB, ser_mode, (uint64_t*)retain->skey.p,
                          (uint32_t*)retain->sidxb.p, n);
+      if (retain->skey2.alloc(8ull * n)) return -12;
+      hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                         c0, 0, ser_mode, (uint64_t*)retain->skey2.p, n);
+      retain->lo_c0 = c0;
       retain->sort_sb = SB;
       retain->sort_ser_mode = ser_mode;
       retain->sorted_valid = 1;
@@ -3729,25 +3872,35 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
     for (auto* sp : live)
       if (sp->sort_sb < SBc) SBc = sp->sort_sb;
     uint64_t mask_c = (SBc >= 8) ? ~0ull : ~0ull << (8 * (8 - SBc));
+    /* common refinement start for the lo keys */
+    int c0c = common_ser ? (8 * SBc - s->pbits - prm.proxy_w) / 8
+                         : (8 * SBc - s->pbits) / 8;
+    if (c0c < 0) c0c = 0;
     /* spills whose retained composites used the content form must be
        rebuilt in serialized form when any spill needs it (their sorted
        order is unchanged: content order == serialized order for the
-       uniform-klen spills that chose the content form — DESIGN.md §3) */
+       uniform-klen spills that chose the content form — DESIGN.md §3);
+       spills whose lo byte range differs from the common one rebuild the
+       lo keys the same way.  (A fully-refined stream is sorted under ANY
+       prefix projection of the comparator order, so rebuilding values
+       never breaks per-stream sortedness.) */
     for (size_t i = 0; i < live.size(); i++) {
       SpillData* sp = live[i];
-      if (common_ser && !sp->sort_ser_mode) {
-        HostRT one;
-        one.add(sp->xdata ? sp->xdata : sp->data.p,
-                sp->xoff ? (const void*)sp->xoff : sp->off.p,
-                sp->xklen ? (const void*)sp->xklen : sp->klen.p,
-                sp->rec_u, sp->klen_u, (uint32_t)sp->n);
-        if (one.finish(s->conf.key_type)) return -12;
-        static thread_local DBuf rebuilt_idx;
+      bool ser_rebuild = common_ser && !sp->sort_ser_mode;
+      bool lo_rebuild = ser_rebuild || sp->lo_c0 != c0c;
+      if (!ser_rebuild && !lo_rebuild) continue;
+      HostRT one;
+      one.add(sp->xdata ? sp->xdata : sp->data.p,
+              sp->xoff ? (const void*)sp->xoff : sp->off.p,
+              sp->xklen ? (const void*)sp->xklen : sp->klen.p,
+              sp->rec_u, sp->klen_u, (uint32_t)sp->n);
+      if (one.finish(s->conf.key_type)) return -12;
+      static thread_local DBuf rebuilt_idx, ser0;
+      if (ser0.alloc(8ull * sp->n)) return -12;
+      if (ser_rebuild) {
         if (rebuilt_idx.alloc(4ull * sp->n)) return -12;
         /* build ser composites in ORIGINAL order, then gather into the
            retained sorted order through sidxb */
-        static thread_local DBuf ser0;
-        if (ser0.alloc(8ull * sp->n)) return -12;
         hipLaunchKernelGGL(k_build_composite, dim3(grid1d(sp->n)), dim3(BLOCK),
                            0, 0, one.rt, (const int32_t*)nullptr,
                            s->conf.num_partitions, s->pbits, prm.ref_pb, SBc,
@@ -3761,65 +3914,89 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
         sp->sort_ser_mode = 1;
         sp->sort_sb = SBc;
       }
+      if (lo_rebuild) {
+        hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(sp->n)), dim3(BLOCK),
+                           0, 0, one.rt, c0c, 0, common_ser,
+                           (uint64_t*)ser0.p, (uint32_t)sp->n);
+        hipLaunchKernelGGL(k_gather_merge_hi, dim3(grid1d(sp->n)), dim3(BLOCK),
+                           0, 0, (const uint64_t*)ser0.p,
+                           (const uint32_t*)sp->sidxb.p,
+                           (uint64_t*)sp->skey2.p, 0ull, (uint32_t)sp->n);
+        sp->lo_c0 = c0c;
+      }
     }
-    /* merge tree: stable pairwise rounds, final round lands in skey/sidx */
+    /* merge tree: stable pairwise rounds over (hi, lo) 128-bit keys,
+       final round lands in (skey, skey_lo, sidx) */
     struct MStream {
-      const uint64_t* k; const uint32_t* pidx; uint32_t n;
+      const uint64_t* k; const uint64_t* l; const uint32_t* pidx; uint32_t n;
       uint64_t mask; uint32_t add;
-      DBuf* own_k; DBuf* own_p;
+      DBuf* own_k; DBuf* own_l; DBuf* own_p;
     };
     std::vector<MStream> streams;
     for (size_t i = 0; i < live.size(); i++) {
       SpillData* sp = live[i];
       MStream m;
       m.k = (const uint64_t*)sp->skey.p;
+      m.l = (const uint64_t*)sp->skey2.p;
       m.pidx = (const uint32_t*)sp->sidxb.p;
       m.n = (uint32_t)sp->n;
       m.mask = mask_c;
       m.add = hrt.base[i];
       m.own_k = &sp->skey;
+      m.own_l = &sp->skey2;
       m.own_p = &sp->sidxb;
       streams.push_back(m);
     }
     if (s->skey.alloc(8ull * total_n)) return -12;
+    if (s->skey_lo.alloc(8ull * total_n)) return -12;
     if (s->sidx.alloc(4ull * total_n)) return -12;
     int R = 0;
     for (size_t m2 = streams.size(); m2 > 1; m2 = (m2 + 1) / 2) R++;
     if (R == 0) {
       MStream& m = streams[0];
-      hipLaunchKernelGGL(k_apply_leaf, dim3(grid1d(m.n)), dim3(BLOCK), 0, 0,
-                         m.k, m.pidx, m.n, m.mask, m.add,
-                         (uint64_t*)s->skey.p, (uint32_t*)s->sidx.p);
+      hipLaunchKernelGGL(k_apply_leaf2, dim3(grid1d(m.n)), dim3(BLOCK), 0, 0,
+                         m.k, m.l, m.pidx, m.n, m.mask, m.add,
+                         (uint64_t*)s->skey.p, (uint64_t*)s->skey_lo.p,
+                         (uint32_t*)s->sidx.p);
       HIP_CHECK(hipDeviceSynchronize());
-      if (m.own_k) { m.own_k->release(); m.own_p->release(); }
+      if (m.own_k) { m.own_k->release(); m.own_l->release(); m.own_p->release(); }
     } else {
-      DBuf& t1k = g_rs_tk64; DBuf& t1p = g_rs_ta0;
-      DBuf& t2k = s->mkey2; DBuf& t2p = s->midx2;
+      /* two buffers suffice: each round reads ONLY the previous round's
+         output (stray streams are materialized into the current round's
+         buffer), so rounds alternate X <-> Y with the parity chosen so the
+         final round lands in X = (s->skey, s->skey_lo, s->sidx). */
+      DBuf& t1k = g_rs_tk64; DBuf& t1l = g_rs_tb64; DBuf& t1p = g_rs_ta0;
+      if (R > 1 && (t1k.alloc(8ull * total_n) || t1l.alloc(8ull * total_n) ||
+                    t1p.alloc(4ull * total_n)))
+        return -12;
       for (int r = 1; r <= R; r++) {
         uint64_t* outk;
+        uint64_t* outl;
         uint32_t* outp;
-        if (r == R) {
-          outk = (uint64_t*)s->skey.p; outp = (uint32_t*)s->sidx.p;
-        } else if (r & 1) {
-          if (t1k.alloc(8ull * total_n) || t1p.alloc(4ull * total_n)) return -12;
-          outk = (uint64_t*)t1k.p; outp = (uint32_t*)t1p.p;
+        if ((R - r) % 2 == 0) {
+          outk = (uint64_t*)s->skey.p;
+          outl = (uint64_t*)s->skey_lo.p;
+          outp = (uint32_t*)s->sidx.p;
         } else {
-          if (t2k.alloc(8ull * total_n) || t2p.alloc(4ull * total_n)) return -12;
-          outk = (uint64_t*)t2k.p; outp = (uint32_t*)t2p.p;
+          outk = (uint64_t*)t1k.p;
+          outl = (uint64_t*)t1l.p;
+          outp = (uint32_t*)t1p.p;
         }
         std::vector<MStream> next;
         uint64_t cursor = 0;
         for (size_t i = 0; i + 1 < streams.size(); i += 2) {
           MStream &A = streams[i], &B = streams[i + 1];
           uint64_t on = (uint64_t)A.n + B.n;
-          uint32_t nblk = (uint32_t)((on + MP_TILE - 1) / MP_TILE);
-          hipLaunchKernelGGL(k_merge_path, dim3(nblk), dim3(MP_BLOCK), 0, 0,
-                             A.k, A.pidx, A.n, A.mask, A.add,
-                             B.k, B.pidx, B.n, B.mask, B.add,
-                             outk + cursor, outp + cursor);
+          uint32_t nblk = (uint32_t)((on + MP2_TILE - 1) / MP2_TILE);
+          hipLaunchKernelGGL(k_merge_path2, dim3(nblk), dim3(MP_BLOCK), 0, 0,
+                             A.k, A.l, A.pidx, A.n, A.mask, A.add,
+                             B.k, B.l, B.pidx, B.n, B.mask, B.add,
+                             outk + cursor, outl + cursor, outp + cursor);
           MStream m;
-          m.k = outk + cursor; m.pidx = outp + cursor; m.n = (uint32_t)on;
-          m.mask = ~0ull; m.add = 0; m.own_k = nullptr; m.own_p = nullptr;
+          m.k = outk + cursor; m.l = outl + cursor; m.pidx = outp + cursor;
+          m.n = (uint32_t)on;
+          m.mask = ~0ull; m.add = 0;
+          m.own_k = nullptr; m.own_l = nullptr; m.own_p = nullptr;
           next.push_back(m);
           cursor += on;
         }
@@ -3827,29 +4004,33 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
           /* stray stream: materialize into this round's output so no source
              buffer is read across round boundaries */
           MStream& A = streams.back();
-          hipLaunchKernelGGL(k_apply_leaf, dim3(grid1d(A.n)), dim3(BLOCK), 0, 0,
-                             A.k, A.pidx, A.n, A.mask, A.add,
-                             outk + cursor, outp + cursor);
+          hipLaunchKernelGGL(k_apply_leaf2, dim3(grid1d(A.n)), dim3(BLOCK), 0, 0,
+                             A.k, A.l, A.pidx, A.n, A.mask, A.add,
+                             outk + cursor, outl + cursor, outp + cursor);
           MStream m;
-          m.k = outk + cursor; m.pidx = outp + cursor; m.n = A.n;
-          m.mask = ~0ull; m.add = 0; m.own_k = nullptr; m.own_p = nullptr;
+          m.k = outk + cursor; m.l = outl + cursor; m.pidx = outp + cursor;
+          m.n = A.n;
+          m.mask = ~0ull; m.add = 0;
+          m.own_k = nullptr; m.own_l = nullptr; m.own_p = nullptr;
           next.push_back(m);
           cursor += A.n;
         }
         HIP_CHECK(hipDeviceSynchronize());
         for (auto& st : streams)
-          if (st.own_k) { st.own_k->release(); st.own_p->release(); }
+          if (st.own_k) { st.own_k->release(); st.own_l->release();
+                          st.own_p->release(); }
         streams.swap(next);
       }
-      t2k.release(); t2p.release();
     }
     for (auto* sp : live) sp->sorted_valid = 0;
     s->times.merge_ns += std::chrono::duration_cast<std::chrono::nanoseconds>(
         std::chrono::steady_clock::now() - t0).count();
     rc = refine_and_emit(s, hrt.segs, hrt.base, hrt.rt, (uint32_t)total_n,
                          SBc, common_ser, spill_rle.data(), lsp, &finalsp,
-                         combine_at_merge, nullptr);
+                         combine_at_merge, nullptr,
+                         (const uint64_t*)s->skey_lo.p);
     if (rc) return rc;
+    s->skey_lo.release();
   }
   s->final_index = finalsp.index;
   std::swap(s->final_ifile, finalsp.ifile);
@@ -4141,6 +4322,16 @@ extern "C" int tzs_sorter_add_sorted_segment(tzs_sorter* s, const void* d_data,
       FAIL(-22, "add_sorted_segment: segment is not sorted (%u inversions)",
            h_bad);
     }
+  }
+  {
+    int c0 = prm.ser_mode ? (8 * prm.SB - s->pbits - prm.proxy_w) / 8
+                          : (8 * prm.SB - s->pbits) / 8;
+    if (c0 < 0) c0 = 0;
+    if (sp->skey2.alloc(8ull * n)) { delete sp; return -12; }
+    hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                       one.rt, c0, 0, prm.ser_mode, (uint64_t*)sp->skey2.p,
+                       (uint32_t)n);
+    sp->lo_c0 = c0;
   }
   sp->sort_sb = prm.SB;
   sp->sort_ser_mode = prm.ser_mode;
