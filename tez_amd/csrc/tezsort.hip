@@ -409,12 +409,16 @@ __global__ void k_radix_scan_digits(uint32_t* totals, uint32_t* bases) {
   bases[t] = lds[t] - totals[t];
 }
 
-/* Stable scatter with up to two u32 payloads.  Two-phase LDS staging:
- * (1) per-tile stable ranking (wave ballots + wave histograms), elements
- * reordered into LDS grouped by digit; (2) digit-contiguous cooperative
- * global writes — a direct per-element scatter write-allocates a 64B line
- * per 12B element (PMC: 22 GB written for 6 GB of payload); digit runs of
- * TILE/256 elements restore coalescing. */
+/* Stable scatter with up to two u32 payloads — barrier-light ranking.
+ * Each (round, wave) pair owns a private u16 count slot, so the 8 ranking
+ * rounds run with NO barriers (the round-1 version re-used one wavehist
+ * row per round: ~3 barriers x rounds; PMC showed the scatters 61-71%
+ * wave-parked on barrier/latency, not bandwidth).  One in-place per-digit
+ * prefix over the (round,wave) slots then yields every element's stable
+ * tile rank; digit offsets come from a single wave-shuffle scan.  The count
+ * region is re-used for the key staging (consumed into registers first).
+ * Stores stay digit-contiguous (partial-line scatter waste was the round-1
+ * fix). */
 template <typename KeyT, bool HAS_A1, bool HAS_B64 = false, int BLK = BLOCK>
 __global__ __launch_bounds__(BLK) void k_radix_scatter(
     const KeyT* keys_in, KeyT* keys_out,
@@ -425,37 +429,37 @@ __global__ __launch_bounds__(BLK) void k_radix_scatter(
     const uint32_t* offsets, const uint32_t* bases) {
   constexpr uint32_t OT = (uint32_t)TILE_ROUNDS * BLK;
   constexpr int OWPB = BLK / WAVE;
-  __shared__ uint32_t tilecnt[RADIX];       /* per-digit running count in tile */
-  __shared__ uint32_t wavehist[OWPB][RADIX];
-  __shared__ uint32_t tileoff[RADIX];       /* exclusive scan of final counts */
-  __shared__ KeyT ls_key[OT];
+  constexpr int RW = OWPB * TILE_ROUNDS;
+  constexpr size_t CNT_BYTES = (size_t)RW * RADIX * 2;
+  constexpr size_t KEY_BYTES = (size_t)OT * sizeof(KeyT);
+  constexpr size_t UNION_BYTES = CNT_BYTES > KEY_BYTES ? CNT_BYTES : KEY_BYTES;
+  __shared__ __attribute__((aligned(16))) uint8_t u_raw[UNION_BYTES];
   __shared__ uint32_t ls_a0[OT];
   __shared__ uint32_t ls_a1[HAS_A1 ? OT : 1];
   __shared__ uint64_t ls_b64[HAS_B64 ? OT : 1];
-  __shared__ uint8_t ls_dig[OT];
-  for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
+  __shared__ uint32_t tileoff[RADIX];
+  uint16_t* cnt = (uint16_t*)u_raw;
+  KeyT* ls_key = (KeyT*)u_raw;
+  for (uint32_t i = threadIdx.x; i < (uint32_t)RW * RADIX / 2; i += blockDim.x)
+    ((uint32_t*)cnt)[i] = 0;
+  __syncthreads();
   uint32_t start = blockIdx.x * OT;
   uint32_t end = min(start + OT, n);
-  uint32_t count = end - start;
+  uint32_t count = (start < n) ? (end - start) : 0;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wv = threadIdx.x / WAVE;
   const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
-  uint32_t my_seq[TILE_ROUNDS];             /* rank within (tile, digit) */
-  uint32_t my_dig[TILE_ROUNDS];
   KeyT my_key[TILE_ROUNDS];
   uint32_t my_a0[TILE_ROUNDS];
-  uint32_t my_a1v[TILE_ROUNDS];
+  uint32_t my_a1v[HAS_A1 ? TILE_ROUNDS : 1];
   uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
-  int round = 0;
-  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
-    uint32_t i = r0 + threadIdx.x;
+  uint16_t my_rank[TILE_ROUNDS];
+  uint16_t my_dig[TILE_ROUNDS];
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
     bool active = i < end;
     KeyT key = active ? keys_in[i] : (KeyT)0;
-    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0xFFFFFFFFu;
-    __syncthreads();
-    for (int j = threadIdx.x; j < OWPB * RADIX; j += blockDim.x)
-      ((uint32_t*)wavehist)[j] = 0;
-    __syncthreads();
+    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0u;
     uint64_t m = ~0ull;
     for (int b = 0; b < 8; b++) {
       uint64_t bb = __ballot((d >> b) & 1);
@@ -464,64 +468,66 @@ __global__ __launch_bounds__(BLK) void k_radix_scatter(
     uint64_t act = __ballot(active);
     m &= act;
     uint32_t lane_rank = (uint32_t)__popcll(m & lt_mask);
-    if (active && lane_rank == 0) wavehist[wv][d] = (uint32_t)__popcll(m);
-    __syncthreads();
-    my_key[round] = key;
-    my_dig[round] = d;
-    my_a0[round] = active ? a0_in[i] : 0;
-    if (HAS_A1) my_a1v[round] = active ? a1_in[i] : 0;
-    if (HAS_B64) my_b64[round] = active ? b64_in[i] : 0;
-    if (active) {
-      uint32_t prior = 0;
-      for (int w = 0; w < wv; w++) prior += wavehist[w][d];
-      my_seq[round] = tilecnt[d] + prior + lane_rank;
-    } else {
-      my_seq[round] = 0;
-    }
-    __syncthreads();
-    for (int j = threadIdx.x; j < RADIX; j += blockDim.x) {
-      uint32_t sum = 0;
-      for (int w = 0; w < OWPB; w++) sum += wavehist[w][j];
-      tilecnt[j] += sum;
-    }
+    if (active && lane_rank == 0)
+      cnt[(r * OWPB + wv) * RADIX + d] = (uint16_t)__popcll(m);
+    my_key[r] = key;
+    my_a0[r] = active ? a0_in[i] : 0;
+    if (HAS_A1) my_a1v[r] = active ? a1_in[i] : 0;
+    if (HAS_B64) my_b64[r] = active ? b64_in[i] : 0;
+    my_rank[r] = (uint16_t)lane_rank;
+    my_dig[r] = (uint16_t)d;
   }
   __syncthreads();
-  /* exclusive scan of tilecnt -> tileoff (256 entries, one block) */
-  {
-    int t = threadIdx.x;
-    uint32_t v = (t < RADIX) ? tilecnt[t] : 0;
-    /* reuse wavehist row 0 as scan scratch */
-    uint32_t* sc = (uint32_t*)wavehist;
-    if (t < RADIX) sc[t] = v;
-    __syncthreads();
-    for (int st = 1; st < RADIX; st <<= 1) {
-      uint32_t add = (t >= st && t < RADIX) ? sc[t - st] : 0;
-      __syncthreads();
-      if (t < RADIX) sc[t] += add;
-      __syncthreads();
+  if (threadIdx.x < RADIX) {
+    uint32_t d = threadIdx.x;
+    uint32_t running = 0;
+    for (int rw = 0; rw < RW; rw++) {
+      uint32_t t = cnt[rw * RADIX + d];
+      cnt[rw * RADIX + d] = (uint16_t)running;
+      running += t;
     }
-    if (t < RADIX) tileoff[t] = sc[t] - v;
-    __syncthreads();
+    tileoff[d] = running;
   }
-  /* place elements into LDS grouped by digit */
-  round = 0;
-  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
-    uint32_t i = r0 + threadIdx.x;
+  __syncthreads();
+  if (wv == 0) {
+    uint32_t v0 = tileoff[4 * lane], v1 = tileoff[4 * lane + 1],
+             v2 = tileoff[4 * lane + 2], v3 = tileoff[4 * lane + 3];
+    uint32_t sum = v0 + v1 + v2 + v3;
+    uint32_t inc = sum;
+    for (int s2 = 1; s2 < WAVE; s2 <<= 1) {
+      uint32_t t = __shfl_up(inc, s2);
+      if (lane >= s2) inc += t;
+    }
+    uint32_t base0 = inc - sum;
+    tileoff[4 * lane] = base0;
+    tileoff[4 * lane + 1] = base0 + v0;
+    tileoff[4 * lane + 2] = base0 + v0 + v1;
+    tileoff[4 * lane + 3] = base0 + v0 + v1 + v2;
+  }
+  __syncthreads();
+  uint16_t my_slot[TILE_ROUNDS];
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t d = my_dig[r];
+    my_slot[r] = (uint16_t)(tileoff[d] + cnt[(r * OWPB + wv) * RADIX + d] +
+                            my_rank[r]);
+  }
+  __syncthreads(); /* cnt consumed; region becomes the key staging */
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
     if (i < end) {
-      uint32_t slot = tileoff[my_dig[round]] + my_seq[round];
-      ls_key[slot] = my_key[round];
-      ls_a0[slot] = my_a0[round];
-      if (HAS_A1) ls_a1[slot] = my_a1v[round];
-      if (HAS_B64) ls_b64[slot] = my_b64[round];
-      ls_dig[slot] = (uint8_t)my_dig[round];
+      uint32_t slot = my_slot[r];
+      ls_key[slot] = my_key[r];
+      ls_a0[slot] = my_a0[r];
+      if (HAS_A1) ls_a1[slot] = my_a1v[r];
+      if (HAS_B64) ls_b64[slot] = my_b64[r];
     }
   }
   __syncthreads();
-  /* digit-contiguous global writes */
   for (uint32_t j = threadIdx.x; j < count; j += blockDim.x) {
-    uint32_t d = ls_dig[j];
+    KeyT k = ls_key[j];
+    uint32_t d = (uint32_t)(k >> (8 * byte_idx)) & 0xFF;
     uint32_t pos = offsets[blockIdx.x * RADIX + d] + bases[d] + (j - tileoff[d]);
-    keys_out[pos] = ls_key[j];
+    keys_out[pos] = k;
     a0_out[pos] = ls_a0[j];
     if (HAS_A1) a1_out[pos] = ls_a1[j];
     if (HAS_B64) b64_out[pos] = ls_b64[j];
@@ -570,20 +576,28 @@ __global__ __launch_bounds__(BLK) void k_onesweep_pass(
     uint32_t n, int byte_idx,
     const uint32_t* bases /* [256] exclusive digit bases */,
     uint32_t* status /* [ntiles*256] */, uint32_t* ticket, uint32_t* error) {
-  constexpr uint32_t OT = (uint32_t)TILE_ROUNDS * BLK;  /* elements per tile */
+  /* barrier-light ranking (see k_radix_scatter) + decoupled lookback:
+   * tiles are ticketed (dispatch order is undefined on CDNA4), per-tile
+   * digit totals publish as AGG/INC status words, the lookback overlaps
+   * the LDS staging stores. */
+  constexpr uint32_t OT = (uint32_t)TILE_ROUNDS * BLK;
   constexpr int OWPB = BLK / WAVE;
-  __shared__ uint32_t tilecnt[RADIX];
-  __shared__ uint32_t wavehist[OWPB][RADIX];
-  __shared__ uint32_t tileoff[RADIX];
-  __shared__ uint32_t excl[RADIX];
-  __shared__ uint32_t s_tile;
-  __shared__ KeyT ls_key[OT];
+  constexpr int RW = OWPB * TILE_ROUNDS;
+  constexpr size_t CNT_BYTES = (size_t)RW * RADIX * 2;
+  constexpr size_t KEY_BYTES = (size_t)OT * sizeof(KeyT);
+  constexpr size_t UNION_BYTES = CNT_BYTES > KEY_BYTES ? CNT_BYTES : KEY_BYTES;
+  __shared__ __attribute__((aligned(16))) uint8_t u_raw[UNION_BYTES];
   __shared__ uint32_t ls_a0[OT];
   __shared__ uint32_t ls_a1[HAS_A1 ? OT : 1];
   __shared__ uint64_t ls_b64[HAS_B64 ? OT : 1];
-  __shared__ uint8_t ls_dig[OT];
+  __shared__ uint32_t tileoff[RADIX];
+  __shared__ uint32_t excl[RADIX];
+  __shared__ uint32_t s_tile;
+  uint16_t* cnt = (uint16_t*)u_raw;
+  KeyT* ls_key = (KeyT*)u_raw;
   if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
-  for (int i = threadIdx.x; i < RADIX; i += blockDim.x) tilecnt[i] = 0;
+  for (uint32_t i = threadIdx.x; i < (uint32_t)RW * RADIX / 2; i += blockDim.x)
+    ((uint32_t*)cnt)[i] = 0;
   __syncthreads();
   const uint32_t tile = s_tile;
   const uint32_t start = tile * OT;
@@ -592,22 +606,17 @@ __global__ __launch_bounds__(BLK) void k_onesweep_pass(
   const int lane = threadIdx.x & (WAVE - 1);
   const int wv = threadIdx.x / WAVE;
   const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
-  uint32_t my_seq[TILE_ROUNDS];
-  uint32_t my_dig[TILE_ROUNDS];
   KeyT my_key[TILE_ROUNDS];
   uint32_t my_a0[TILE_ROUNDS];
-  uint32_t my_a1v[TILE_ROUNDS];
+  uint32_t my_a1v[HAS_A1 ? TILE_ROUNDS : 1];
   uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
-  int round = 0;
-  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
-    uint32_t i = r0 + threadIdx.x;
+  uint16_t my_rank[TILE_ROUNDS];
+  uint16_t my_dig[TILE_ROUNDS];
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
     bool active = i < end;
     KeyT key = active ? keys_in[i] : (KeyT)0;
-    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0xFFFFFFFFu;
-    __syncthreads();
-    for (int j = threadIdx.x; j < OWPB * RADIX; j += blockDim.x)
-      ((uint32_t*)wavehist)[j] = 0;
-    __syncthreads();
+    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0u;
     uint64_t m = ~0ull;
     for (int b = 0; b < 8; b++) {
       uint64_t bb = __ballot((d >> b) & 1);
@@ -616,71 +625,69 @@ __global__ __launch_bounds__(BLK) void k_onesweep_pass(
     uint64_t act = __ballot(active);
     m &= act;
     uint32_t lane_rank = (uint32_t)__popcll(m & lt_mask);
-    if (active && lane_rank == 0) wavehist[wv][d] = (uint32_t)__popcll(m);
-    __syncthreads();
-    my_key[round] = key;
-    my_dig[round] = d;
-    my_a0[round] = active ? a0_in[i] : 0;
-    if (HAS_A1) my_a1v[round] = active ? a1_in[i] : 0;
-    if (HAS_B64) my_b64[round] = active ? b64_in[i] : 0;
-    if (active) {
-      uint32_t prior = 0;
-      for (int w = 0; w < wv; w++) prior += wavehist[w][d];
-      my_seq[round] = tilecnt[d] + prior + lane_rank;
-    } else {
-      my_seq[round] = 0;
-    }
-    __syncthreads();
-    for (int j = threadIdx.x; j < RADIX; j += blockDim.x) {
-      uint32_t sum = 0;
-      for (int w = 0; w < OWPB; w++) sum += wavehist[w][j];
-      tilecnt[j] += sum;
-    }
+    if (active && lane_rank == 0)
+      cnt[(r * OWPB + wv) * RADIX + d] = (uint16_t)__popcll(m);
+    my_key[r] = key;
+    my_a0[r] = active ? a0_in[i] : 0;
+    if (HAS_A1) my_a1v[r] = active ? a1_in[i] : 0;
+    if (HAS_B64) my_b64[r] = active ? b64_in[i] : 0;
+    my_rank[r] = (uint16_t)lane_rank;
+    my_dig[r] = (uint16_t)d;
   }
   __syncthreads();
-  /* publish aggregate (or inclusive for tile 0) — single-word relaxed
-     agent-scope stores (sc1), data-is-the-flag */
+  /* per-digit exclusive prefix over (round,wave) slots + publish totals */
   if (threadIdx.x < RADIX) {
     uint32_t d = threadIdx.x;
-    uint32_t v = tilecnt[d] & OS_CNT_MASK;
+    uint32_t running = 0;
+    for (int rw = 0; rw < RW; rw++) {
+      uint32_t t = cnt[rw * RADIX + d];
+      cnt[rw * RADIX + d] = (uint16_t)running;
+      running += t;
+    }
+    tileoff[d] = running;
     if (tile == 0)
-      __hip_atomic_store((os_gu32*)&status[d], v | OS_INC,
+      __hip_atomic_store((os_gu32*)&status[d], (running & OS_CNT_MASK) | OS_INC,
                          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     else
-      __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d], v | OS_AGG,
+      __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d],
+                         (running & OS_CNT_MASK) | OS_AGG,
                          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
   }
-  /* tile-local digit offsets */
-  {
-    int t = threadIdx.x;
-    uint32_t v = (t < RADIX) ? tilecnt[t] : 0;
-    uint32_t* sc = (uint32_t*)wavehist;
-    if (t < RADIX) sc[t] = v;
-    __syncthreads();
-    for (int st = 1; st < RADIX; st <<= 1) {
-      uint32_t add = (t >= st && t < RADIX) ? sc[t - st] : 0;
-      __syncthreads();
-      if (t < RADIX) sc[t] += add;
-      __syncthreads();
+  __syncthreads();
+  if (wv == 0) {
+    uint32_t v0 = tileoff[4 * lane], v1 = tileoff[4 * lane + 1],
+             v2 = tileoff[4 * lane + 2], v3 = tileoff[4 * lane + 3];
+    uint32_t sum = v0 + v1 + v2 + v3;
+    uint32_t inc = sum;
+    for (int s2 = 1; s2 < WAVE; s2 <<= 1) {
+      uint32_t t = __shfl_up(inc, s2);
+      if (lane >= s2) inc += t;
     }
-    if (t < RADIX) tileoff[t] = sc[t] - v;
-    __syncthreads();
+    uint32_t base0 = inc - sum;
+    tileoff[4 * lane] = base0;
+    tileoff[4 * lane + 1] = base0 + v0;
+    tileoff[4 * lane + 2] = base0 + v0 + v1;
+    tileoff[4 * lane + 3] = base0 + v0 + v1 + v2;
   }
-  /* place into LDS grouped by digit, then digit-contiguous global writes */
-  round = 0;
-  for (uint32_t r0 = start; r0 < start + OT; r0 += blockDim.x, round++) {
-    uint32_t i = r0 + threadIdx.x;
+  __syncthreads();
+  uint16_t my_slot[TILE_ROUNDS];
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t d = my_dig[r];
+    my_slot[r] = (uint16_t)(tileoff[d] + cnt[(r * OWPB + wv) * RADIX + d] +
+                            my_rank[r]);
+  }
+  __syncthreads(); /* cnt consumed; region becomes the key staging */
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
     if (i < end) {
-      uint32_t slot = tileoff[my_dig[round]] + my_seq[round];
-      ls_key[slot] = my_key[round];
-      ls_a0[slot] = my_a0[round];
-      if (HAS_A1) ls_a1[slot] = my_a1v[round];
-      if (HAS_B64) ls_b64[slot] = my_b64[round];
-      ls_dig[slot] = (uint8_t)my_dig[round];
+      uint32_t slot = my_slot[r];
+      ls_key[slot] = my_key[r];
+      ls_a0[slot] = my_a0[r];
+      if (HAS_A1) ls_a1[slot] = my_a1v[r];
+      if (HAS_B64) ls_b64[slot] = my_b64[r];
     }
   }
-  /* lookback overlapped after placement: thread d resolves digit d's
-     exclusive prefix while the other waves finished their LDS stores */
+  /* lookback overlapped after placement */
   if (threadIdx.x < RADIX) {
     uint32_t d = threadIdx.x;
     uint32_t e = 0;
@@ -695,17 +702,19 @@ __global__ __launch_bounds__(BLK) void k_onesweep_pass(
         if (++spins > 100000000u) { atomicAdd(error, 1u); break; }
         __builtin_amdgcn_s_sleep(4);
       }
+      uint32_t tot = ((d + 1 < RADIX) ? tileoff[d + 1] : count) - tileoff[d];
       __hip_atomic_store((os_gu32*)&status[(uint64_t)tile * RADIX + d],
-                         ((e + tilecnt[d]) & OS_CNT_MASK) | OS_INC,
+                         ((e + tot) & OS_CNT_MASK) | OS_INC,
                          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
     excl[d] = e;
   }
   __syncthreads();
   for (uint32_t j = threadIdx.x; j < count; j += blockDim.x) {
-    uint32_t d = ls_dig[j];
+    KeyT k = ls_key[j];
+    uint32_t d = (uint32_t)(k >> (8 * byte_idx)) & 0xFF;
     uint32_t pos = bases[d] + excl[d] + (j - tileoff[d]);
-    keys_out[pos] = ls_key[j];
+    keys_out[pos] = k;
     a0_out[pos] = ls_a0[j];
     if (HAS_A1) a1_out[pos] = ls_a1[j];
     if (HAS_B64) b64_out[pos] = ls_b64[j];
