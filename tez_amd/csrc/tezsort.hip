@@ -1457,6 +1457,101 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
   }
 }
 
+/* Staged-gather emit v2: each wave takes 64 records; every lane issues ALL
+ * of its own record's dword loads at once (64 records x ~23 loads of MLP —
+ * the rotate pipeline kept only 2 records in flight and sat 70% parked,
+ * PMC r2), staging the raw source words into LDS; the drain then runs the
+ * proven byte-store path against LDS instead of global.  Records longer
+ * than EMIT2_CAP bytes fall back to the per-record global loop. */
+#define EMIT2_CAP 128
+#define EMIT2_W (EMIT2_CAP / 4 + 2)
+__global__ __launch_bounds__(BLOCK) void k_emit_records_v2(
+    const RecDesc* desc, const uint8_t* same, const uint64_t* scan,
+    const uint32_t* parts, const uint64_t* seg_payload_start,
+    const uint64_t* part_scan_base, uint8_t* out, uint32_t n) {
+  __shared__ uint32_t stage[WPB][WAVE][EMIT2_W];
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  uint32_t wv = threadIdx.x / WAVE;
+  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (uint64_t base = (uint64_t)wave * WAVE; base < n;
+       base += (uint64_t)nwaves * WAVE) {
+    uint32_t i = (uint32_t)base + lane;
+    uint64_t my_src = 0, my_dst = 0, my_h0 = 0, my_h1 = 0;
+    uint32_t my_len = 0, my_hdr = 0, my_ph = 0;
+    if (i < n) {
+      RecDesc v = desc[i];
+      uint32_t p = parts[i];
+      my_dst = seg_payload_start[p] + (scan[i] - part_scan_base[p]);
+      uint8_t prev_same = (i > 0 && parts[i] == parts[i - 1]) ? same[i - 1] : 0;
+      uint8_t hdrbuf[16] = {0};
+      uint32_t hdr = 0;
+      if (same[i]) {
+        if (!prev_same) hdrbuf[hdr++] = 0xFE;
+        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
+        my_src = v.src + v.klen;
+        my_len = v.vlen;
+      } else {
+        if (prev_same) hdrbuf[hdr++] = 0xFD;
+        hdr += d_vint_write(hdrbuf + hdr, v.klen);
+        hdr += d_vint_write(hdrbuf + hdr, v.vlen);
+        my_src = v.src;
+        my_len = v.klen + v.vlen;
+      }
+      my_hdr = hdr;
+      for (int b = 0; b < 8; b++) my_h0 |= (uint64_t)hdrbuf[b] << (8 * b);
+      for (int b = 8; b < 12; b++) my_h1 |= (uint64_t)hdrbuf[b] << (8 * (b - 8));
+    }
+    uint32_t nvalid = (n - base < WAVE) ? (uint32_t)(n - base) : WAVE;
+    uint32_t maxlen = my_len;
+    for (int sh = 32; sh >= 1; sh >>= 1) {
+      uint32_t v2 = __shfl_xor(maxlen, sh);
+      if (v2 > maxlen) maxlen = v2;
+    }
+    if (maxlen <= EMIT2_CAP) {
+      /* stage: aligned dword loads from src&~3 — ALL records in flight */
+      my_ph = (uint32_t)(my_src & 3);
+      const uint32_t* sw = (const uint32_t*)(my_src & ~3ull);
+      uint32_t nw = (my_ph + my_len + 3) >> 2;
+      uint32_t* row = stage[wv][lane];
+      #pragma unroll 4
+      for (uint32_t k = 0; k < nw; k++) row[k] = sw[k];
+      /* drain: per record, whole wave cooperates; source = LDS */
+      for (uint32_t r = 0; r < nvalid; r++) {
+        uint64_t dsto = __shfl(my_dst, r);
+        uint64_t h0 = __shfl(my_h0, r);
+        uint64_t h1 = __shfl(my_h1, r);
+        uint32_t len = __shfl(my_len, r);
+        uint32_t hdr = __shfl(my_hdr, r);
+        uint32_t ph = __shfl(my_ph, r);
+        uint8_t* w = out + dsto;
+        if (lane < hdr)
+          w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
+                               : (uint8_t)(h1 >> (8 * (lane - 8)));
+        w += hdr;
+        const uint8_t* ls = (const uint8_t*)stage[wv][r] + ph;
+        for (uint32_t b = lane; b < len; b += WAVE) w[b] = ls[b];
+      }
+    } else {
+      for (uint32_t r = 0; r < nvalid; r++) {
+        uint64_t src = __shfl(my_src, r);
+        uint64_t dsto = __shfl(my_dst, r);
+        uint64_t h0 = __shfl(my_h0, r);
+        uint64_t h1 = __shfl(my_h1, r);
+        uint32_t len = __shfl(my_len, r);
+        uint32_t hdr = __shfl(my_hdr, r);
+        uint8_t* w = out + dsto;
+        if (lane < hdr)
+          w[lane] = (lane < 8) ? (uint8_t)(h0 >> (8 * lane))
+                               : (uint8_t)(h1 >> (8 * (lane - 8)));
+        w += hdr;
+        const uint8_t* sp2 = (const uint8_t*)(uintptr_t)src;
+        for (uint32_t b = lane; b < len; b += WAVE) w[b] = sp2[b];
+      }
+    }
+  }
+}
+
 /* Staged-span emit: each WAVE owns 64 consecutive sorted records — within a
  * partition their output bytes are one contiguous span.  Every lane stages
  * its own record (header regs + funneled u32 window loads) into a wave-local
@@ -3559,7 +3654,22 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
       const char* e = getenv("TZS_EMIT_SPAN");
       use_span = (e && e[0] == '1') ? 1 : 0;
     }
-    if (use_span && !force_simple) {
+    static int use_v2 = -1;
+    if (use_v2 < 0) {
+      const char* e = getenv("TZS_EMIT_V2");
+      /* measured SLOWER (emit 10.2 -> 28.7 ms at C2): the 64-record LDS
+         drain serializes on shuffle+bank-conflicted byte reads; kept for
+         experiments only */
+      use_v2 = (e && e[0] == '1') ? 1 : 0;
+    }
+    if (use_v2 && !force_simple && !use_span) {
+      hipLaunchKernelGGL(k_emit_records_v2, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
+                         (const RecDesc*)descbuf.p,
+                         (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
+                         (const uint32_t*)s->parts_sorted.p,
+                         (const uint64_t*)d_paystart.p,
+                         (const uint64_t*)d_scanbase.p, d_out, n);
+    } else if (use_span && !force_simple) {
       hipLaunchKernelGGL(k_emit_span, dim3(grid_waves(n)), dim3(BLOCK), 0, 0,
                          (const RecDesc*)descbuf.p,
                          (const uint8_t*)s->same.p, (const uint64_t*)s->scan.p,
