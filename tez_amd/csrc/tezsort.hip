@@ -289,10 +289,19 @@ __global__ void k_check_uniform(const uint64_t* off, const uint32_t* klen, int64
 
 /* ---- partition + composite ---- */
 __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32_t n) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x) {
-    RecView v = rt_view(rt, i);
-    d_part[i] = (d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P;
+  /* 2 records per iteration: the per-record serial hash chain left waves
+     parked on one gather at a time (85% WAIT_ANY, r2 PMC) */
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += 2 * stride) {
+    uint32_t j = i + stride;
+    RecView v0 = rt_view(rt, i);
+    int32_t h0 = d_hash_bytes(v0.content, (int32_t)v0.clen);
+    if (j < n) {
+      RecView v1 = rt_view(rt, j);
+      int32_t h1 = d_hash_bytes(v1.content, (int32_t)v1.clen);
+      d_part[j] = (h1 & 0x7fffffff) % P;
+    }
+    d_part[i] = (h0 & 0x7fffffff) % P;
   }
 }
 
@@ -1081,6 +1090,19 @@ __device__ __forceinline__ uint32_t d_mp_diag(
   return lo;
 }
 
+/* tile-boundary splits computed up front: one thread per boundary, all
+ * searches in flight at once (the in-kernel 2-thread serial search made the
+ * merge latency-bound: ~25 dependent global loads per block). */
+__global__ void k_mp_partition(const uint64_t* ka, uint32_t na, uint64_t maskA,
+                               const uint64_t* kb, uint32_t nb, uint64_t maskB,
+                               uint32_t nblk, uint32_t tile, uint32_t* splits) {
+  uint64_t total = (uint64_t)na + nb;
+  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t <= nblk;
+       t += gridDim.x * blockDim.x) {
+    uint64_t D = min((uint64_t)t * tile, total);
+    splits[t] = d_mp_diag(ka, maskA, na, kb, maskB, nb, D);
+  }
+}
 __global__ __launch_bounds__(MP_BLOCK) void k_merge_path(
     const uint64_t* ka, const uint32_t* pa, uint32_t na, uint64_t maskA,
     uint32_t addA,
@@ -1203,26 +1225,33 @@ __device__ __forceinline__ uint32_t d_mp2_diag(
   }
   return lo_;
 }
+__global__ void k_mp2_partition(const uint64_t* ka, const uint64_t* la,
+                                uint32_t na, uint64_t maskA,
+                                const uint64_t* kb, const uint64_t* lb,
+                                uint32_t nb, uint64_t maskB,
+                                uint32_t nblk, uint32_t tile, uint32_t* splits) {
+  uint64_t total = (uint64_t)na + nb;
+  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t <= nblk;
+       t += gridDim.x * blockDim.x) {
+    uint64_t D = min((uint64_t)t * tile, total);
+    splits[t] = d_mp2_diag(ka, la, maskA, na, kb, lb, maskB, nb, D);
+  }
+}
 __global__ __launch_bounds__(MP_BLOCK) void k_merge_path2(
     const uint64_t* ka, const uint64_t* la, const uint32_t* pa, uint32_t na,
     uint64_t maskA, uint32_t addA,
     const uint64_t* kb, const uint64_t* lb, const uint32_t* pb, uint32_t nb,
     uint64_t maskB, uint32_t addB,
-    uint64_t* kout, uint64_t* loout, uint32_t* pout) {
+    uint64_t* kout, uint64_t* loout, uint32_t* pout,
+    const uint32_t* splits /* [nblk+1] from k_mp2_partition */) {
   __shared__ uint64_t ls_k[MP2_TILE];
   __shared__ uint64_t ls_l[MP2_TILE];
   __shared__ uint32_t ls_p[MP2_TILE];
-  __shared__ uint32_t s_sp[2];
   uint64_t total = (uint64_t)na + nb;
   uint64_t D0 = (uint64_t)blockIdx.x * MP2_TILE;
   if (D0 >= total) return;
   uint64_t D1 = min(D0 + (uint64_t)MP2_TILE, total);
-  if (threadIdx.x < 2) {
-    uint64_t D = threadIdx.x ? D1 : D0;
-    s_sp[threadIdx.x] = d_mp2_diag(ka, la, maskA, na, kb, lb, maskB, nb, D);
-  }
-  __syncthreads();
-  uint32_t a0 = s_sp[0], a1 = s_sp[1];
+  uint32_t a0 = splits[blockIdx.x], a1 = splits[blockIdx.x + 1];
   uint32_t b0 = (uint32_t)(D0 - a0), b1 = (uint32_t)(D1 - a1);
   uint32_t nA = a1 - a0, nB = b1 - b0;
   for (uint32_t i = threadIdx.x; i < nA; i += blockDim.x) {
@@ -2515,7 +2544,13 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
      seg-sort pays a host histogram sync.  The dispatch below supports every
      shape; the gate keeps onesweep to the base (key + record-id) sort where
      it wins. */
-  if (use_onesweep && !has_a1 && !has_b64 && n >= 20000) {
+  /* payload shapes ride onesweep too since the barrier-light rewrite:
+     u64+two-u32 payloads at 1024 threads (130 KB LDS, 16 waves/CU);
+     u32-key+u64 payload at 256 (41 KB, 12 waves/CU — the 1024 variant
+     would need 162 KB).  Pre-rewrite this measured slower and was gated
+     off. */
+  bool os_payload_ok = !has_b64 || sizeof(KeyT) == 4;
+  if (use_onesweep && os_payload_ok && !(has_a1 && has_b64) && n >= 20000) {
     int npasses = nbytes_key - first_byte;
     /* bigger blocks scale the tile while KEEPING 16 waves/CU: 512 threads =
        4096-elem tile (2 blocks x 8 waves), 1024 = 8192-elem tile (1 block x
@@ -2528,7 +2563,9 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
       os_blk = e ? atoi(e) : 1024;
       if (os_blk != 256 && os_blk != 512 && os_blk != 1024) os_blk = 1024;
     }
-    uint32_t os_tile = (uint32_t)TILE_ROUNDS * (uint32_t)os_blk;
+    /* payload shapes pin their block size (LDS budget — see dispatch) */
+    int eff_blk = has_b64 ? 256 : (has_a1 ? 1024 : os_blk);
+    uint32_t os_tile = (uint32_t)TILE_ROUNDS * (uint32_t)eff_blk;
     uint32_t nb_os = nblocks_for(n, os_tile);
     static thread_local DBuf gh, gbases, st, tick;
     if (gh.alloc(4u * npasses * RADIX)) return -12;
@@ -2561,8 +2598,18 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
       if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                       (void)hipEventRecord(evs[nev]); }
       const uint32_t* pbases = (const uint32_t*)((uint32_t*)gbases.p + p * RADIX);
-      /* base shape only (gate above): dispatch on block size */
-      if (os_blk == 1024)
+      if (has_b64) {
+        if constexpr (sizeof(KeyT) == 4)
+          hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, true, 256>), dim3(nb_os),
+                             dim3(256), 0, 0, kin, kout, a0in, a0out, nullptr,
+                             nullptr, b64in, b64out, n, b, pbases, (uint32_t*)st.p,
+                             (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      } else if (has_a1) {
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, true, false, 1024>), dim3(nb_os),
+                           dim3(1024), 0, 0, kin, kout, a0in, a0out, a1in, a1out,
+                           nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      } else if (os_blk == 1024)
         hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false, 1024>), dim3(nb_os),
                            dim3(1024), 0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
                            nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
@@ -4103,14 +4150,20 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
         }
         std::vector<MStream> next;
         uint64_t cursor = 0;
+        static thread_local DBuf splitbuf;
         for (size_t i = 0; i + 1 < streams.size(); i += 2) {
           MStream &A = streams[i], &B = streams[i + 1];
           uint64_t on = (uint64_t)A.n + B.n;
           uint32_t nblk = (uint32_t)((on + MP2_TILE - 1) / MP2_TILE);
+          if (splitbuf.alloc(4ull * (nblk + 1))) return -12;
+          hipLaunchKernelGGL(k_mp2_partition, dim3(grid1d(nblk + 1)), dim3(BLOCK),
+                             0, 0, A.k, A.l, A.n, A.mask, B.k, B.l, B.n, B.mask,
+                             nblk, (uint32_t)MP2_TILE, (uint32_t*)splitbuf.p);
           hipLaunchKernelGGL(k_merge_path2, dim3(nblk), dim3(MP_BLOCK), 0, 0,
                              A.k, A.l, A.pidx, A.n, A.mask, A.add,
                              B.k, B.l, B.pidx, B.n, B.mask, B.add,
-                             outk + cursor, outl + cursor, outp + cursor);
+                             outk + cursor, outl + cursor, outp + cursor,
+                             (const uint32_t*)splitbuf.p);
           MStream m;
           m.k = outk + cursor; m.l = outl + cursor; m.pidx = outp + cursor;
           m.n = (uint32_t)on;
