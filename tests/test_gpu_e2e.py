@@ -109,13 +109,22 @@ def test_exchange_world1_on_device(engine):
     _, _, _, rr, br = m.sorted_columnar()
     plan = ex.plan_send(rr, br, 1)
     sd, srl, skl = ex.pack_send_tensors(m, plan, device)
-    rd, rrl, rkl = ex.exchange(plan, sd, srl, skl)
-    rparts = ex.exchange_parts(plan, P, device)
+    rd, rrl, rkl, rparts = ex.exchange(plan, sd, srl, skl, nparts=P)
+    # product path: each source chunk ingested as a PRE-SORTED segment
+    # (tzs_sorter_add_sorted_segment; no reduce-side re-sort)
     red = ex.reduce_merge(lambda: engine.Sorter(engine.make_conf(P)), rd, rrl, rkl,
-                          rparts)
+                          rparts, src_rec_splits=plan.recv_rec_splits,
+                          src_byte_splits=plan.recv_byte_splits)
     got_data, got_idx = red.output()
     red.close()
+    # legacy unsorted-batch path must agree too
+    red2 = ex.reduce_merge(lambda: engine.Sorter(engine.make_conf(P)), rd, rrl, rkl,
+                           ex.exchange_parts(plan, P, device))
+    got2_data, got2_idx = red2.output()
+    red2.close()
     m.close()
+    assert got2_idx == got_idx
+    assert got2_data == got_data
     engine.free_device(d, off, kl, part)
     assert got_idx == want_idx
     assert got_data == want_data

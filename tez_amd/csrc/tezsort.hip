@@ -2549,7 +2549,16 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
      u32-key+u64 payload at 256 (41 KB, 12 waves/CU — the 1024 variant
      would need 162 KB).  Pre-rewrite this measured slower and was gated
      off. */
-  bool os_payload_ok = !has_b64 || sizeof(KeyT) == 4;
+  static int os_payload = -1;
+  if (os_payload < 0) {
+    const char* e = getenv("TZS_OS_PAYLOAD");
+    os_payload = (e && e[0] == '1') ? 1 : 0;  /* measured: classic wins for
+      payload shapes even after the barrier-light rewrite (C3 1e9 sort_ns
+      551 vs 580 ms) — the per-sort host histogram syncs dominate at
+      refinement sizes */
+  }
+  bool os_payload_ok = (!has_a1 && !has_b64) ||
+                       (os_payload && (!has_b64 || sizeof(KeyT) == 4));
   if (use_onesweep && os_payload_ok && !(has_a1 && has_b64) && n >= 20000) {
     int npasses = nbytes_key - first_byte;
     /* bigger blocks scale the tile while KEEPING 16 waves/CU: 512 threads =
