@@ -137,6 +137,16 @@ int tzs_sorter_flush(tzs_sorter* s);
 
 int tzs_sorter_num_spills(const tzs_sorter* s);
 
+/* Compressed final output: the same partition segments
+ * re-framed as TIF\1 — one zlib stream per segment (device deflate:
+ * 32 KB chunks, fixed-Huffman LZ77 with stored fallback, sync-flush
+ * stitching), CRC32 over the COMPRESSED payload, per the reference's
+ * DefaultCodec framing (IFile.java:352-368; golden-fixture-pinned reader).
+ * index: rawLength keeps the uncompressed accounting, partLength = 4 +
+ * zlib stream + 4.  Buffer owned by the sorter (valid until close). */
+int tzs_sorter_output_compressed(tzs_sorter* s, const void** d_bytes,
+                                 int64_t* nbytes, tzs_index_record* index);
+
 /* Final merged output: IFile bytes for all partitions concatenated (device
  * pointer, owned by the sorter) + index records (host). Valid until close. */
 int tzs_sorter_output(tzs_sorter* s, const void** d_bytes, int64_t* nbytes,

@@ -100,6 +100,9 @@ def lib():
     L.tzs_sorter_num_spills.argtypes = [c.c_void_p]
     L.tzs_sorter_output.argtypes = [c.c_void_p, c.POINTER(c.c_void_p),
                                     c.POINTER(c.c_int64), c.POINTER(TzsIndexRecord)]
+    L.tzs_sorter_output_compressed.argtypes = [c.c_void_p, c.POINTER(c.c_void_p),
+                                               c.POINTER(c.c_int64),
+                                               c.POINTER(TzsIndexRecord)]
     L.tzs_sorter_spill_output.argtypes = [c.c_void_p, c.c_int32, c.POINTER(c.c_void_p),
                                           c.POINTER(c.c_int64), c.POINTER(TzsIndexRecord)]
     L.tzs_sorter_write_files.argtypes = [c.c_void_p, c.c_char_p, c.c_char_p]
@@ -200,6 +203,21 @@ class Sorter:
             ba = (ctypes.c_char * n.value).from_buffer(buf)
             _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba), p, n.value), "d2h")
         return bytes(buf), [(r.start_offset, r.raw_length, r.part_length) for r in idx]
+
+    def output_compressed(self):
+        """TIF\\1 compressed final stream (device deflate) + index."""
+        p = ctypes.c_void_p()
+        n = ctypes.c_int64()
+        idx = (TzsIndexRecord * self.conf.num_partitions)()
+        _ck(lib().tzs_sorter_output_compressed(self.h, ctypes.byref(p),
+                                               ctypes.byref(n), idx),
+            "output_compressed")
+        buf = bytearray(n.value)
+        if n.value:
+            ba = (ctypes.c_char * n.value).from_buffer(buf)
+            _ck(lib().tzs_memcpy_d2h(ctypes.addressof(ba), p, n.value), "d2h")
+        return bytes(buf), [(r.start_offset, r.raw_length, r.part_length)
+                            for r in idx]
 
     def spill_output(self, spill_id):
         p = ctypes.c_void_p()

@@ -2023,6 +2023,211 @@ __global__ void k_patch_segments(uint8_t* out, const uint64_t* seg_start,
   e[2] = (uint8_t)(crc >> 8); e[3] = (uint8_t)crc;
 }
 
+/* ---- device deflate (TIF\\1 compressed IFile segments) ----------------
+ * SURVEY 8f row 2 / VERDICT r1 #9: the reference compresses each IFile
+ * segment with a Hadoop codec (DefaultCodec = one zlib stream per segment,
+ * CRC32 over the COMPRESSED payload - IFile.java:352-368, pinned by the
+ * golden fixture).  MI355X-native encoder: the payload splits into 32 KB
+ * chunks; each chunk is deflated by one wave (greedy LZ77 over an LDS hash
+ * table + fixed-Huffman emit in lane 0, lane-parallel stored-block copy
+ * when incompressible) into its own slot, each chunk ending byte-aligned
+ * via an empty stored (sync-flush) block so the slots concatenate into ONE
+ * valid zlib stream; per-chunk adler32 halves combine on the host.
+ * Compressed BYTES are encoder-specific (any inflater accepts them); the
+ * reference contract is the framing + CRC + decompressed payload. */
+#define DEF_CHUNK 32768
+#define DEF_SLOT  (DEF_CHUNK + DEF_CHUNK / 8 + 64) /* fixed-huffman worst < 9/8 */
+#define DEF_HASH_BITS 12
+#define DEF_HASH (1 << DEF_HASH_BITS)
+
+struct DefChunk {
+  uint64_t in_off;   /* absolute offset in the input stream */
+  uint32_t in_len;   /* <= DEF_CHUNK */
+  uint32_t last;     /* 1 = final chunk of its segment (BFINAL) */
+};
+
+__device__ __forceinline__ uint32_t d_bitrev(uint32_t v, int n) {
+  uint32_t r = 0;
+  for (int i = 0; i < n; i++) { r = (r << 1) | (v & 1); v >>= 1; }
+  return r;
+}
+
+/* length code: base lengths + extra bits (RFC1951 3.2.5) */
+__constant__ uint16_t c_len_base[29] = {3,4,5,6,7,8,9,10,11,13,15,17,19,23,27,31,
+                                        35,43,51,59,67,83,99,115,131,163,195,227,258};
+__constant__ uint8_t c_len_extra[29] = {0,0,0,0,0,0,0,0,1,1,1,1,2,2,2,2,
+                                        3,3,3,3,4,4,4,4,5,5,5,5,0};
+__constant__ uint16_t c_dist_base[30] = {1,2,3,4,5,7,9,13,17,25,33,49,65,97,129,193,
+                                         257,385,513,769,1025,1537,2049,3073,4097,6145,
+                                         8193,12289,16385,24577};
+__constant__ uint8_t c_dist_extra[30] = {0,0,0,0,1,1,2,2,3,3,4,4,5,5,6,6,
+                                         7,7,8,8,9,9,10,10,11,11,12,12,13,13};
+
+struct DefBitWriter {
+  uint8_t* out;
+  uint64_t acc;
+  uint32_t nbits;
+  uint32_t pos;
+  __device__ void put(uint32_t bits, uint32_t n) {
+    acc |= (uint64_t)bits << nbits;
+    nbits += n;
+    while (nbits >= 8) {
+      out[pos++] = (uint8_t)acc;
+      acc >>= 8;
+      nbits -= 8;
+    }
+  }
+  __device__ void align() {
+    if (nbits) { out[pos++] = (uint8_t)acc; acc = 0; nbits = 0; }
+  }
+};
+
+/* fixed-huffman literal/length code (already bit-reversed for LSB packing) */
+__device__ __forceinline__ void d_fh_lit(DefBitWriter& bw, uint32_t v) {
+  if (v < 144) bw.put(d_bitrev(0x30 + v, 8), 8);
+  else bw.put(d_bitrev(0x190 + (v - 144), 9), 9);
+}
+__device__ __forceinline__ void d_fh_len(DefBitWriter& bw, uint32_t len) {
+  int c = 28;
+  while (c > 0 && c_len_base[c] > len) c--;
+  /* code 257+c: 257..279 -> 7 bits (code-256), 280..285 -> 8 bits */
+  uint32_t sym = 257 + c;
+  if (sym < 280) bw.put(d_bitrev(sym - 256, 7), 7);
+  else bw.put(d_bitrev(0xC0 + (sym - 280), 8), 8);
+  if (c_len_extra[c]) bw.put(len - c_len_base[c], c_len_extra[c]);
+}
+__device__ __forceinline__ void d_fh_dist(DefBitWriter& bw, uint32_t dist) {
+  int c = 29;
+  while (c > 0 && c_dist_base[c] > dist) c--;
+  bw.put(d_bitrev((uint32_t)c, 5), 5);
+  if (c_dist_extra[c]) bw.put(dist - c_dist_base[c], c_dist_extra[c]);
+}
+
+/* one wave per chunk; lane 0 runs the serial parse/emit, all lanes build
+ * the hash table and handle the stored-block fallback copy */
+__global__ __launch_bounds__(WAVE) void k_deflate_chunks(
+    const uint8_t* in, const DefChunk* chunks, uint32_t nchunks,
+    uint8_t* slots /* [nchunks][DEF_SLOT] */, uint32_t* out_len,
+    uint32_t* out_adler /* [nchunks][2]: a, b */) {
+  __shared__ uint16_t htab[DEF_HASH];
+  const uint32_t lane = threadIdx.x;
+  for (uint32_t c = blockIdx.x; c < nchunks; c += gridDim.x) {
+    DefChunk ck = chunks[c];
+    const uint8_t* src = in + ck.in_off;
+    uint8_t* dst = slots + (uint64_t)c * DEF_SLOT;
+    for (uint32_t i = lane; i < DEF_HASH; i += WAVE) htab[i] = 0xFFFF;
+    __syncthreads();  /* one wave per block: cheap */
+    /* adler32 halves: lane-parallel partial sums then lane-0 fold.
+       a = 1 + S bytes; b = len + S (len-i)*byte_i  (mod 65521) */
+    uint64_t sa = 0, sb = 0;
+    for (uint32_t i = lane; i < ck.in_len; i += WAVE) {
+      uint32_t v = src[i];
+      sa += v;
+      sb += (uint64_t)v * (ck.in_len - i);
+    }
+    for (int sh = 32; sh >= 1; sh >>= 1) {
+      sa += __shfl_xor(sa, sh);
+      sb += __shfl_xor(sb, sh);
+    }
+    if (lane == 0) {
+      out_adler[2 * c] = (uint32_t)((1 + sa) % 65521u);
+      out_adler[2 * c + 1] = (uint32_t)((ck.in_len + sb) % 65521u);
+      /* serial greedy LZ77 + fixed-huffman */
+      DefBitWriter bw;
+      bw.out = dst;
+      bw.acc = 0; bw.nbits = 0; bw.pos = 0;
+      bw.put(0, 1);      /* BFINAL=0 (the closing stored block carries it) */
+      bw.put(1, 2);      /* BTYPE=01 fixed huffman */
+      uint32_t n2 = ck.in_len;
+      uint32_t i = 0;
+      uint32_t budget = DEF_CHUNK + DEF_CHUNK / 16; /* abort to stored */
+      while (i < n2) {
+        if (bw.pos >= budget) break;
+        uint32_t mlen = 0, mdist = 0;
+        if (i + 3 <= n2) {
+          uint32_t w3 = (uint32_t)src[i] | ((uint32_t)src[i + 1] << 8) |
+                        ((uint32_t)src[i + 2] << 16);
+          uint32_t h = (w3 * 2654435761u) >> (32 - DEF_HASH_BITS);
+          uint32_t cand = htab[h];
+          htab[h] = (uint16_t)i;
+          if (cand != 0xFFFF && cand < i) {
+            const uint8_t* p = src + cand;
+            const uint8_t* q = src + i;
+            uint32_t maxm = n2 - i;
+            if (maxm > 258) maxm = 258;
+            uint32_t l = 0;
+            while (l < maxm && p[l] == q[l]) l++;
+            if (l >= 3) { mlen = l; mdist = i - cand; }
+          }
+        }
+        if (mlen) {
+          d_fh_len(bw, mlen);
+          d_fh_dist(bw, mdist);
+          /* seed the table inside the match (sparsely: every 2nd byte) */
+          uint32_t e = i + mlen;
+          for (uint32_t j = i + 1; j + 3 <= n2 && j < e; j += 2) {
+            uint32_t w3 = (uint32_t)src[j] | ((uint32_t)src[j + 1] << 8) |
+                          ((uint32_t)src[j + 2] << 16);
+            htab[(w3 * 2654435761u) >> (32 - DEF_HASH_BITS)] = (uint16_t)j;
+          }
+          i = e;
+        } else {
+          d_fh_lit(bw, src[i]);
+          i++;
+        }
+      }
+      if (i >= n2) {
+        bw.put(0, 7);  /* end-of-block (code 256, 7 zero bits) */
+        bw.align();
+      }
+      out_len[c] = (i >= n2) ? bw.pos : 0xFFFFFFFFu; /* sentinel: use stored */
+    }
+    __syncthreads();
+    /* broadcast lane 0's decision; stored fallback copies with all lanes */
+    uint32_t marker = (uint32_t)__shfl(lane == 0 ? (int)out_len[c] : 0, 0);
+    uint32_t clen;
+    if (marker == 0xFFFFFFFFu) {
+      /* stored block: BFINAL=0,BTYPE=00 (1 byte 0x00), LEN, NLEN, raw */
+      if (lane == 0) {
+        dst[0] = 0x00;
+        dst[1] = (uint8_t)ck.in_len;
+        dst[2] = (uint8_t)(ck.in_len >> 8);
+        dst[3] = (uint8_t)~dst[1];
+        dst[4] = (uint8_t)~dst[2];
+      }
+      for (uint32_t i = lane; i < ck.in_len; i += WAVE) dst[5 + i] = src[i];
+      clen = 5 + ck.in_len;
+    } else {
+      clen = marker;
+    }
+    /* closing block: empty stored with this chunk's BFINAL (byte-aligns
+       every chunk so slots concatenate; Z_SYNC_FLUSH shape) */
+    if (lane == 0) {
+      dst[clen] = ck.last ? 0x01 : 0x00;
+      dst[clen + 1] = 0x00;
+      dst[clen + 2] = 0x00;
+      dst[clen + 3] = 0xFF;
+      dst[clen + 4] = 0xFF;
+      out_len[c] = clen + 5;
+    }
+  }
+}
+
+/* compact chunk slots into the final stream at host-computed offsets */
+__global__ void k_deflate_gather(const uint8_t* slots, const uint32_t* lens,
+                                 const uint64_t* dst_off, uint32_t nchunks,
+                                 uint8_t* out) {
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint32_t lane = threadIdx.x & (WAVE - 1);
+  uint32_t nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (uint32_t c = wave; c < nchunks; c += nwaves) {
+    const uint8_t* s = slots + (uint64_t)c * DEF_SLOT;
+    uint8_t* d = out + dst_off[c];
+    uint32_t n = lens[c];
+    for (uint32_t i = lane; i < n; i += WAVE) d[i] = s[i];
+  }
+}
+
 /* ---- synthetic generation (bench/tests) ---- */
 __device__ __forceinline__ uint64_t d_splitmix64(uint64_t x) {
   x += 0x9E3779B97F4A7C15ull;
@@ -2870,6 +3075,9 @@ struct tzs_sorter {
   /* scratch kept across calls */
   DBuf skey, sidx, eq, same, sizes, scan, parts_sorted;
   DBuf skey_lo;                            /* merged lo keys (flush merge) */
+  DBuf comp_ifile;                         /* TIF\1 compressed final stream */
+  int64_t comp_len = 0;
+  std::vector<tzs_index_record> comp_index;
   /* final-sort metadata for the exchange path */
   std::vector<uint64_t> final_rec_ranges;  /* [P+1] record index ranges */
   HostRT final_hrt;  /* multi-segment flush tables: owns the device SegDesc
@@ -4248,6 +4456,209 @@ extern "C" int tzs_sorter_counters(const tzs_sorter* s, tzs_counters* out) {
 }
 extern "C" int tzs_sorter_times(const tzs_sorter* s, tzs_times* out) {
   *out = s->times;
+  return 0;
+}
+
+/* CRC32 of arbitrary device ranges (reuses the chunked CRC kernels) */
+static int crc32_ranges(const uint8_t* d_stream,
+                        const std::vector<uint64_t>& starts,
+                        const std::vector<uint64_t>& lens,
+                        std::vector<uint32_t>& out) {
+  int P = (int)starts.size();
+  out.assign(P, 0);
+  if (P == 0) return 0;
+  std::vector<uint64_t> chunkbase(P + 1, 0), scbase(P + 1, 0), groupbase(P + 1, 0);
+  for (int p = 0; p < P; p++) {
+    uint64_t nchunks = (lens[p] + CRC_CHUNK - 1) / CRC_CHUNK;
+    chunkbase[p + 1] = chunkbase[p] + nchunks;
+    scbase[p + 1] = scbase[p] + (nchunks + CRC_SC_CHUNKS - 1) / CRC_SC_CHUNKS;
+    groupbase[p + 1] = groupbase[p] + (nchunks + CRC_GROUP_CHUNKS - 1) / CRC_GROUP_CHUNKS;
+  }
+  uint64_t total_chunks = chunkbase[P], total_sc = scbase[P], total_groups = groupbase[P];
+  DBuf d_start, d_len, d_cb, d_sb, d_gb, d_cc, d_gc, d_gl, d_pc;
+  auto up2 = [&](DBuf& b, const void* src2, size_t sz) -> int {
+    if (b.alloc(sz)) return -12;
+    HIP_CHECK(hipMemcpyAsync(b.p, src2, sz, hipMemcpyHostToDevice));
+    return 0;
+  };
+  if (up2(d_start, starts.data(), 8 * P)) return -12;
+  if (up2(d_len, lens.data(), 8 * P)) return -12;
+  if (up2(d_cb, chunkbase.data(), 8 * (P + 1))) return -12;
+  if (up2(d_sb, scbase.data(), 8 * (P + 1))) return -12;
+  if (up2(d_gb, groupbase.data(), 8 * (P + 1))) return -12;
+  if (d_cc.alloc(4 * (total_chunks ? total_chunks : 1))) return -12;
+  if (d_gc.alloc(4 * (total_groups ? total_groups : 1))) return -12;
+  if (d_gl.alloc(8 * (total_groups ? total_groups : 1))) return -12;
+  if (d_pc.alloc(4 * P)) return -12;
+  if (total_sc)
+    hipLaunchKernelGGL(k_crc_chunks,
+                       dim3((uint32_t)min(total_sc, (uint64_t)4096)), dim3(BLOCK), 0, 0,
+                       d_stream, (const uint64_t*)d_start.p, (const uint64_t*)d_len.p,
+                       (const uint64_t*)d_cb.p, (const uint64_t*)d_sb.p, P,
+                       (uint32_t)total_sc, (uint32_t*)d_cc.p);
+  if (total_groups)
+    hipLaunchKernelGGL(k_crc_combine_groups,
+                       dim3((uint32_t)min(total_groups, (uint64_t)2048)), dim3(BLOCK), 0, 0,
+                       (const uint64_t*)d_len.p, (const uint64_t*)d_cb.p,
+                       (const uint64_t*)d_gb.p, (const uint32_t*)d_cc.p, P,
+                       (uint32_t)total_groups, (uint32_t*)d_gc.p, (uint64_t*)d_gl.p);
+  hipLaunchKernelGGL(k_crc_combine_final, dim3((P * WAVE + BLOCK - 1) / BLOCK),
+                     dim3(BLOCK), 0, 0, (const uint64_t*)d_gb.p,
+                     (const uint32_t*)d_gc.p, (const uint64_t*)d_gl.p, P,
+                     (uint32_t*)d_pc.p);
+  HIP_CHECK(hipMemcpy(out.data(), d_pc.p, 4 * P, hipMemcpyDeviceToHost));
+  return 0;
+}
+
+/* Compressed final output: the same partition segments re-framed as TIF\1
+ * (zlib stream per segment, CRC32 over the COMPRESSED payload — the
+ * reference's DefaultCodec framing, IFile.java:352-368).  rawLength keeps
+ * the uncompressed accounting; partLength = 4 + zlib-stream + 4. */
+extern "C" int tzs_sorter_output_compressed(tzs_sorter* s, const void** d_bytes,
+                                            int64_t* nbytes,
+                                            tzs_index_record* index) {
+  if (!s->flushed) FAIL(-22, "flush first");
+  if (ensure_device_constants()) return -70;
+  int P = s->conf.num_partitions;
+  const uint8_t* src = (const uint8_t*)s->final_ifile.p;
+  /* per-present-segment uncompressed payload range = [start+4, start+raw) */
+  std::vector<DefChunk> chunks;
+  std::vector<int> seg_first(P + 1, 0);
+  for (int p = 0; p < P; p++) {
+    seg_first[p] = (int)chunks.size();
+    const tzs_index_record& ix = s->final_index[p];
+    if (ix.part_length <= 0) continue;
+    uint64_t off = (uint64_t)ix.start_offset + 4;
+    uint64_t len = (uint64_t)ix.raw_length - 4; /* body + tail, no CRC */
+    uint64_t done = 0;
+    while (done < len) {
+      DefChunk ck;
+      ck.in_off = off + done;
+      ck.in_len = (uint32_t)min(len - done, (uint64_t)DEF_CHUNK);
+      done += ck.in_len;
+      ck.last = (done == len) ? 1 : 0;
+      chunks.push_back(ck);
+    }
+  }
+  seg_first[P] = (int)chunks.size();
+  uint32_t nchunks = (uint32_t)chunks.size();
+  std::vector<uint32_t> h_len(nchunks), h_adler(2 * nchunks);
+  const uint32_t BATCH = 32768; /* 1.2 GB of slot scratch */
+  DBuf d_chunks, d_slots, d_len, d_adl;
+  if (d_chunks.alloc(sizeof(DefChunk) * (nchunks ? nchunks : 1))) return -12;
+  if (nchunks)
+    HIP_CHECK(hipMemcpyAsync(d_chunks.p, chunks.data(),
+                             sizeof(DefChunk) * nchunks, hipMemcpyHostToDevice));
+  if (d_len.alloc(4ull * (nchunks ? nchunks : 1))) return -12;
+  if (d_adl.alloc(8ull * (nchunks ? nchunks : 1))) return -12;
+  uint32_t nbatch = nchunks ? min(nchunks, BATCH) : 1;
+  if (d_slots.alloc((uint64_t)nbatch * DEF_SLOT)) return -12;
+  /* pass 1: compressed sizes + adler halves (single-batch runs keep the
+     slots and skip the second deflate) */
+  bool single_batch = nchunks <= BATCH;
+  for (uint32_t b0 = 0; b0 < nchunks; b0 += BATCH) {
+    uint32_t bc = min(BATCH, nchunks - b0);
+    hipLaunchKernelGGL(k_deflate_chunks, dim3(min(bc, 4096u)), dim3(WAVE), 0, 0,
+                       src, (const DefChunk*)d_chunks.p + b0, bc,
+                       (uint8_t*)d_slots.p, (uint32_t*)d_len.p + b0,
+                       (uint32_t*)d_adl.p + 2ull * b0);
+    HIP_CHECK(hipMemcpy(h_len.data() + b0, (uint32_t*)d_len.p + b0, 4ull * bc,
+                        hipMemcpyDeviceToHost));
+  }
+  if (nchunks)
+    HIP_CHECK(hipMemcpy(h_adler.data(), d_adl.p, 8ull * nchunks,
+                        hipMemcpyDeviceToHost));
+  /* layout: per present segment TIF\1 + 0x78 0x9C + deflate + adler + CRC */
+  std::vector<uint64_t> comp_seg(P, 0), seg_start(P, 0);
+  uint64_t total = 0;
+  for (int p = 0; p < P; p++) {
+    uint64_t t = 0;
+    for (int c = seg_first[p]; c < seg_first[p + 1]; c++) t += h_len[c];
+    comp_seg[p] = t;
+    seg_start[p] = total;
+    if (s->final_index[p].part_length > 0)
+      total += 4 + 2 + t + 4 + 4;
+  }
+  if (s->comp_ifile.alloc(total ? total + 16 : 1)) return -12;
+  uint8_t* d_out2 = (uint8_t*)s->comp_ifile.p;
+  /* pass 2: place chunks at their final offsets */
+  std::vector<uint64_t> h_final_off(nchunks);
+  {
+    for (int p = 0; p < P; p++) {
+      if (s->final_index[p].part_length <= 0) continue;
+      uint64_t pos = seg_start[p] + 4 + 2;
+      for (int c = seg_first[p]; c < seg_first[p + 1]; c++) {
+        h_final_off[c] = pos;
+        pos += h_len[c];
+      }
+    }
+    static thread_local DBuf d_off2;
+    if (d_off2.alloc(8ull * (nchunks ? nchunks : 1))) return -12;
+    for (uint32_t b0 = 0; b0 < nchunks; b0 += BATCH) {
+      uint32_t bc = min(BATCH, nchunks - b0);
+      if (!single_batch)
+        hipLaunchKernelGGL(k_deflate_chunks, dim3(min(bc, 4096u)), dim3(WAVE), 0, 0,
+                           src, (const DefChunk*)d_chunks.p + b0, bc,
+                           (uint8_t*)d_slots.p, (uint32_t*)d_len.p + b0,
+                           (uint32_t*)d_adl.p + 2ull * b0);
+      HIP_CHECK(hipMemcpyAsync(d_off2.p, h_final_off.data() + b0, 8ull * bc,
+                               hipMemcpyHostToDevice));
+      hipLaunchKernelGGL(k_deflate_gather, dim3(grid_waves(bc)), dim3(BLOCK), 0, 0,
+                         (const uint8_t*)d_slots.p, (const uint32_t*)d_len.p + b0,
+                         (const uint64_t*)d_off2.p, bc, d_out2);
+      HIP_CHECK(hipDeviceSynchronize());
+    }
+  }
+  /* headers + adler trailers */
+  s->comp_index.assign(P, tzs_index_record{0, 0, 0});
+  for (int p = 0; p < P; p++) {
+    s->comp_index[p].start_offset = (int64_t)seg_start[p];
+    if (s->final_index[p].part_length <= 0) continue;
+    uint32_t a = 1, bsum = 0;
+    for (int c = seg_first[p]; c < seg_first[p + 1]; c++) {
+      /* zlib adler32 combine: a' = a + a2 - 1; b' = b + b2 + len2*(a-1) */
+      uint32_t a2 = h_adler[2 * c], b2 = h_adler[2 * c + 1];
+      uint64_t rem = chunks[c].in_len % 65521u;
+      uint64_t nb2 = ((uint64_t)bsum + b2 + rem * (((uint64_t)a + 65520u) % 65521u)) % 65521u;
+      a = (uint32_t)(((uint64_t)a + a2 + 65520u) % 65521u);
+      bsum = (uint32_t)nb2;
+    }
+    uint8_t head[6] = {'T', 'I', 'F', 1, 0x78, 0x9C};
+    HIP_CHECK(hipMemcpy(d_out2 + seg_start[p], head, 6, hipMemcpyHostToDevice));
+    uint8_t atr[4] = {(uint8_t)(bsum >> 8), (uint8_t)bsum,
+                      (uint8_t)(a >> 8), (uint8_t)a};
+    HIP_CHECK(hipMemcpy(d_out2 + seg_start[p] + 6 + comp_seg[p], atr, 4,
+                        hipMemcpyHostToDevice));
+    s->comp_index[p].raw_length = s->final_index[p].raw_length;
+    s->comp_index[p].part_length = (int64_t)(4 + 2 + comp_seg[p] + 4 + 4);
+  }
+  /* CRC32 over each compressed stream (zlib hdr .. adler) + trailer patch */
+  {
+    std::vector<uint64_t> cst, cln;
+    std::vector<int> cmap;
+    for (int p = 0; p < P; p++)
+      if (s->final_index[p].part_length > 0) {
+        cst.push_back(seg_start[p] + 4);
+        cln.push_back(2 + comp_seg[p] + 4);
+        cmap.push_back(p);
+      }
+    std::vector<uint32_t> crcs;
+    int rc2 = crc32_ranges(d_out2, cst, cln, crcs);
+    if (rc2) return rc2;
+    for (size_t i = 0; i < cmap.size(); i++) {
+      int p = cmap[i];
+      uint32_t crc = crcs[i];
+      uint8_t tr[4] = {(uint8_t)(crc >> 24), (uint8_t)(crc >> 16),
+                       (uint8_t)(crc >> 8), (uint8_t)crc};
+      HIP_CHECK(hipMemcpy(d_out2 + seg_start[p] + 4 + cln[i], tr, 4,
+                          hipMemcpyHostToDevice));
+    }
+  }
+  s->comp_len = (int64_t)total;
+  if (d_bytes) *d_bytes = s->comp_ifile.p;
+  if (nbytes) *nbytes = s->comp_len;
+  if (index && P)
+    memcpy(index, s->comp_index.data(), sizeof(tzs_index_record) * P);
   return 0;
 }
 
