@@ -2178,37 +2178,44 @@ __global__ __launch_bounds__(WAVE) void k_deflate_chunks(
       }
       if (i >= n2) {
         bw.put(0, 7);  /* end-of-block (code 256, 7 zero bits) */
+        /* closing empty stored block (sync flush) carries this chunk's
+           BFINAL; its 3 header bits continue the BIT stream, THEN the
+           stream pads to a byte boundary, then LEN/NLEN — every chunk
+           ends byte-aligned so the slots concatenate */
+        bw.put(ck.last ? 1u : 0u, 1);
+        bw.put(0, 2); /* BTYPE=00 */
         bw.align();
+        bw.out[bw.pos++] = 0x00;
+        bw.out[bw.pos++] = 0x00;
+        bw.out[bw.pos++] = 0xFF;
+        bw.out[bw.pos++] = 0xFF;
+        out_len[c] = bw.pos;
+      } else {
+        out_len[c] = 0xFFFFFFFFu; /* sentinel: use stored */
       }
-      out_len[c] = (i >= n2) ? bw.pos : 0xFFFFFFFFu; /* sentinel: use stored */
     }
     __syncthreads();
     /* broadcast lane 0's decision; stored fallback copies with all lanes */
     uint32_t marker = (uint32_t)__shfl(lane == 0 ? (int)out_len[c] : 0, 0);
-    uint32_t clen;
     if (marker == 0xFFFFFFFFu) {
-      /* stored block: BFINAL=0,BTYPE=00 (1 byte 0x00), LEN, NLEN, raw */
+      /* stored block: BFINAL=0,BTYPE=00 (1 byte 0x00), LEN, NLEN, raw;
+         data ends byte-aligned, so the closing block header starts a new
+         byte here */
       if (lane == 0) {
         dst[0] = 0x00;
         dst[1] = (uint8_t)ck.in_len;
         dst[2] = (uint8_t)(ck.in_len >> 8);
         dst[3] = (uint8_t)~dst[1];
         dst[4] = (uint8_t)~dst[2];
+        uint32_t clen = 5 + ck.in_len;
+        dst[clen] = ck.last ? 0x01 : 0x00;
+        dst[clen + 1] = 0x00;
+        dst[clen + 2] = 0x00;
+        dst[clen + 3] = 0xFF;
+        dst[clen + 4] = 0xFF;
+        out_len[c] = clen + 5;
       }
       for (uint32_t i = lane; i < ck.in_len; i += WAVE) dst[5 + i] = src[i];
-      clen = 5 + ck.in_len;
-    } else {
-      clen = marker;
-    }
-    /* closing block: empty stored with this chunk's BFINAL (byte-aligns
-       every chunk so slots concatenate; Z_SYNC_FLUSH shape) */
-    if (lane == 0) {
-      dst[clen] = ck.last ? 0x01 : 0x00;
-      dst[clen + 1] = 0x00;
-      dst[clen + 2] = 0x00;
-      dst[clen + 3] = 0xFF;
-      dst[clen + 4] = 0xFF;
-      out_len[c] = clen + 5;
     }
   }
 }
