@@ -37,6 +37,17 @@ class OrderedPartitionedKVOutput:
         self._final_merge = str(self.props.get(
             "tez.runtime.enable.final-merge.in.output", "true")).lower() \
             in ("1", "true", "yes")
+        # tez.runtime.compress + codec (ExternalSorter.java:228): the engine
+        # emits DefaultCodec (zlib) TIF\1 segments on device
+        self._compress = str(self.props.get(
+            "tez.runtime.compress", "false")).lower() in ("1", "true", "yes")
+        codec = self.props.get(
+            "tez.runtime.compress.codec",
+            "org.apache.hadoop.io.compress.DefaultCodec")
+        if self._compress and codec != \
+                "org.apache.hadoop.io.compress.DefaultCodec":
+            raise ValueError(f"unsupported compress.codec {codec} "
+                             "(engine emits DefaultCodec/zlib)")
         key_cls = self.props.get("tez.runtime.key.class",
                                  "org.apache.hadoop.io.BytesWritable")
         if key_cls == "org.apache.hadoop.io.Text":
@@ -100,7 +111,10 @@ class OrderedPartitionedKVOutput:
             self._sorter.close()
             self._sorter = None
             return events
-        data, index = self._sorter.output()
+        if self._compress:
+            data, index = self._sorter.output_compressed()
+        else:
+            data, index = self._sorter.output()
         ctr = self._sorter.counters()
         self._events = ev.events_on_flush(
             index, self.num_partitions, self.host, self.port,
