@@ -3521,8 +3521,12 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
   int max_clen = (int)max_klen; /* upper bound on content length */
   static thread_local DBuf inrun_scan, eqcnt;
   static thread_local DBuf lkey, seg, pos, slotpos;
+  static thread_local DBuf lk0;
   if (eqcnt.alloc(4)) return -12;
 
+  bool will_combine = apply_combine && s->conf.combiner == 1;
+  bool lk0_retain_ready = false; /* dense level-0 keys double as the lo
+                                    retention source (same byte range) */
   int level = 1;
   const int max_levels = (max_clen > c0_eff) ? (max_clen - c0_eff + 7) / 8 : 0;
   /* initial ambiguity count; later levels get it from k_eq_update's
@@ -3559,9 +3563,10 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
          Gated to n <= 3e8: the 8n-byte lk0 buffer at a 1e9-record merge
          pushed the peak working set into pool-drop/hipMalloc churn
          (C3 1e9 53 -> 14 GB/s). */
-      static thread_local DBuf lk0;
       const uint64_t* lk0p = nullptr;
-      if ((uint64_t)m * 2 >= n && n <= 300000000u) {
+      /* !lk0_retain_ready: a later level must not overwrite the retained
+         level-0 content (it gathers through sidx after the loop) */
+      if ((uint64_t)m * 2 >= n && n <= 300000000u && !lk0_retain_ready) {
         if (lk0.alloc(8ull * n)) return -12;
         hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
                            lb0, use_len, ser_mode, (uint64_t*)lk0.p, n);
@@ -3571,7 +3576,10 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                          d_idx, d_eq, (uint64_t*)inrun_scan.p, n,
                          lb0, use_len, ser_mode, lk0p, (uint64_t*)lkey.p,
                          (uint32_t*)seg.p, (uint32_t*)pos.p);
-      lk0.release();
+      if (li == 0 && !use_len && lk0p && retain && !will_combine)
+        lk0_retain_ready = true; /* keep the buffer for retention below */
+      else
+        lk0.release();
     }
     /* slotpos = copy of pos (ascending) before sort */
     HIP_CHECK(hipMemcpyAsync(slotpos.p, pos.p, sizeof(uint32_t) * m,
@@ -3614,23 +3622,25 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
      #1): composites move out (nothing below reads them), sorted ids are
      copied (s->sidx stays live for emit + sorted_columnar).  Combiner
      spills retain the FOLDED set instead (built in the combiner branch). */
-  bool will_combine = apply_combine && s->conf.combiner == 1;
   if (retain && !will_combine) {
     std::swap(retain->skey, s->skey);
     if (retain->sidxb.alloc(4ull * n)) return -12;
     HIP_CHECK(hipMemcpyAsync(retain->sidxb.p, s->sidx.p, 4ull * n,
                              hipMemcpyDeviceToDevice));
-    /* lo keys: comparator-source bytes [c0, c0+8) per sorted record —
-       dense build (coalesced) + one 8B gather into sorted order */
+    /* lo keys: comparator-source bytes [c0, c0+8) per sorted record — the
+       refinement's dense level-0 build is the same array when it ran;
+       otherwise build it now, then one 8B gather into sorted order */
     if (retain->skey2.alloc(8ull * n)) return -12;
     {
-      static thread_local DBuf lo_tmp;
-      if (lo_tmp.alloc(8ull * n)) return -12;
-      hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                         c0, 0, ser_mode, (uint64_t*)lo_tmp.p, n);
+      if (!lk0_retain_ready) {
+        if (lk0.alloc(8ull * n)) return -12;
+        hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                           c0, 0, ser_mode, (uint64_t*)lk0.p, n);
+      }
       hipLaunchKernelGGL(k_gather_merge_hi, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
-                         (const uint64_t*)lo_tmp.p, (const uint32_t*)s->sidx.p,
+                         (const uint64_t*)lk0.p, (const uint32_t*)s->sidx.p,
                          (uint64_t*)retain->skey2.p, 0ull, n);
+      lk0.release();
     }
     retain->lo_c0 = c0;
     retain->sort_sb = SB;
