@@ -3578,7 +3578,7 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                          (uint32_t*)seg.p, (uint32_t*)pos.p);
       if (li == 0 && !use_len && lk0p && retain && !will_combine)
         lk0_retain_ready = true; /* keep the buffer for retention below */
-      else
+      else if (!lk0_retain_ready)
         lk0.release();
     }
     /* slotpos = copy of pos (ascending) before sort */
