@@ -934,6 +934,109 @@ __global__ void k_compact_refine(RecTable rt, const uint32_t* sidx,
     pos[j] = i;
   }
 }
+/* fused run-flags + decoupled-lookback scan + compaction: one kernel
+ * replaces k_run_flags + scan_u64 + k_compact_refine (the separate chain
+ * wrote and re-read an 8n flag array and re-read eq — ~24 B/element of
+ * scratch traffic; C3's refinement was the dominant profile block).
+ * Packed u64 counters: (inrun << 32) | runstart, same protocol as
+ * k_scan_lookback. */
+__global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
+    RecTable rt, const uint32_t* sidx, const uint8_t* eq, uint32_t n,
+    int level_byte0, int use_len_level, int ser_mode, const uint64_t* lk0,
+    uint64_t* lkey, uint32_t* seg, uint32_t* pos,
+    uint64_t* status, uint32_t* ticket, uint32_t* error, uint64_t* total_out) {
+  __shared__ uint64_t lds[BLOCK];
+  __shared__ uint32_t s_tile;
+  __shared__ uint64_t s_excl;
+  if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
+  __syncthreads();
+  const uint32_t tile = s_tile;
+  const uint32_t base = tile * SCAN_TILE;
+  uint8_t my_eq[SCAN_ITEMS];
+  uint8_t my_in[SCAN_ITEMS];
+  uint64_t vals[SCAN_ITEMS];
+  uint64_t sum = 0;
+  #pragma unroll
+  for (int j = 0; j < SCAN_ITEMS; j++) {
+    uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
+    uint8_t e = (i < n) ? eq[i] : 0;
+    uint8_t nx = (i + 1 < n) ? eq[i + 1] : 0;
+    uint8_t in = (i < n) && (e || nx);
+    my_eq[j] = e;
+    my_in[j] = in;
+    uint64_t rs = (in && !e) ? 1 : 0;
+    vals[j] = ((uint64_t)in << 32) | rs;
+    sum += vals[j];
+  }
+  lds[threadIdx.x] = sum;
+  __syncthreads();
+  for (int s2 = 1; s2 < BLOCK; s2 <<= 1) {
+    uint64_t t = (threadIdx.x >= (uint32_t)s2) ? lds[threadIdx.x - s2] : 0;
+    __syncthreads();
+    lds[threadIdx.x] += t;
+    __syncthreads();
+  }
+  uint64_t block_total = lds[BLOCK - 1];
+  if (threadIdx.x == 0) {
+    if (tile == 0) {
+      __hip_atomic_store((os_gu64*)&status[0], OSS_INC | block_total,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      s_excl = 0;
+    } else {
+      __hip_atomic_store((os_gu64*)&status[tile], OSS_AGG | block_total,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      uint64_t e = 0;
+      int64_t p = (int64_t)tile - 1;
+      uint32_t spins = 0;
+      while (p >= 0) {
+        uint64_t v = __hip_atomic_load((os_gu64*)&status[p],
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (v & OSS_INC) { e += v & OSS_VAL; break; }
+        if (v & OSS_AGG) { e += v & OSS_VAL; p--; continue; }
+        if (++spins > 100000000u) { atomicAdd(error, 1u); break; }
+        __builtin_amdgcn_s_sleep(4);
+      }
+      __hip_atomic_store((os_gu64*)&status[tile], OSS_INC | (e + block_total),
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      s_excl = e;
+    }
+  }
+  __syncthreads();
+  uint64_t excl = s_excl + lds[threadIdx.x] - sum;
+  #pragma unroll
+  for (int j = 0; j < SCAN_ITEMS; j++) {
+    uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
+    if (i < n && my_in[j]) {
+      uint32_t cj = (uint32_t)(excl >> 32);
+      uint32_t is_start = my_eq[j] ? 0u : 1u;
+      uint32_t sg = (uint32_t)(excl & 0xFFFFFFFFu) + is_start - 1;
+      uint64_t k;
+      if (lk0) {
+        k = lk0[sidx[i]];
+      } else if (use_len_level) {
+        RecView v = rt_view(rt, sidx[i]);
+        k = ser_mode ? v.klen : v.clen;
+      } else {
+        RecView v = rt_view(rt, sidx[i]);
+        const uint8_t* src = ser_mode ? v.key : v.content;
+        uint32_t slen = ser_mode ? v.klen : v.clen;
+        k = 0;
+        for (int b = 0; b < 8; b++) {
+          uint32_t cb = (uint32_t)(level_byte0 + b);
+          uint8_t byte = (cb < slen) ? src[cb] : 0;
+          k |= (uint64_t)byte << (56 - 8 * b);
+        }
+      }
+      lkey[cj] = k;
+      seg[cj] = sg;
+      pos[cj] = i;
+    }
+    excl += vals[j];
+  }
+  if (total_out && threadIdx.x == 0 && (uint64_t)base + SCAN_TILE >= n)
+    *total_out = s_excl + block_total;
+}
+
 /* scatter refined order back: new_sidx[pos_sorted[j]] stays — we permute the
  * POSITIONS: output[orig_pos_slot_j] where slot order = sorted compact order.
  * We write: for compact rank j (after sort), its element moves to the j-th
@@ -3519,7 +3622,7 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
     }
   }
   int max_clen = (int)max_klen; /* upper bound on content length */
-  static thread_local DBuf inrun_scan, eqcnt;
+  static thread_local DBuf eqcnt;
   static thread_local DBuf lkey, seg, pos, slotpos;
   static thread_local DBuf lk0;
   if (eqcnt.alloc(4)) return -12;
@@ -3541,46 +3644,55 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
   for (int li = 0; li <= max_levels; li++) {
     if (neq == 0) break;
     int use_len = (li == max_levels); /* final tiebreak: content length */
-    if (inrun_scan.alloc(sizeof(uint64_t) * n)) return -12;
-    hipLaunchKernelGGL(k_run_flags, dim3(grid1d(n)), dim3(BLOCK), 0, 0, d_eq,
-                       (uint64_t*)inrun_scan.p, n);
-    uint64_t ptotal = 0;
-    /* in-place scan: halves the refinement's largest scratch (8n bytes) —
-       the C3 1e9 shape was pool-drop bound (DESIGN 7a memory rule) */
-    if (scan_u64((uint64_t*)inrun_scan.p, (uint64_t*)inrun_scan.p, n, &ptotal))
-      return -12;
-    uint64_t nruns = ptotal & 0xFFFFFFFFu;
-    uint32_t m = (uint32_t)(ptotal >> 32);
-    if (m == 0) break;
-    if (lkey.alloc(sizeof(uint64_t) * m)) return -12;
-    if (seg.alloc(sizeof(uint32_t) * m)) return -12;
-    if (pos.alloc(sizeof(uint32_t) * m)) return -12;
-    if (slotpos.alloc(sizeof(uint32_t) * m)) return -12;
+    /* m (in-run elements) <= 2*neq: a run of r equal elements carries r-1
+       eq flags and r in-run members, r <= 2(r-1) for r >= 2 */
+    uint64_t mcap64 = 2ull * neq;
+    uint32_t mcap = (uint32_t)(mcap64 < n ? mcap64 : n);
+    if (lkey.alloc(sizeof(uint64_t) * mcap)) return -12;
+    if (seg.alloc(sizeof(uint32_t) * mcap)) return -12;
+    if (pos.alloc(sizeof(uint32_t) * mcap)) return -12;
     int lb0 = c0_eff + 8 * li;
+    uint64_t ptotal = 0;
     {
       /* dense original-order level-key build when most records are still
-         ambiguous (m >= ~n/2): coalesced reads beat the random gather.
-         Gated to n <= 3e8: the 8n-byte lk0 buffer at a 1e9-record merge
-         pushed the peak working set into pool-drop/hipMalloc churn
-         (C3 1e9 53 -> 14 GB/s). */
+         ambiguous: coalesced reads beat the random gather.  Gated to
+         n <= 3e8: the 8n-byte lk0 buffer at a 1e9-record merge pushed the
+         peak working set into pool-drop/hipMalloc churn. */
       const uint64_t* lk0p = nullptr;
       /* !lk0_retain_ready: a later level must not overwrite the retained
          level-0 content (it gathers through sidx after the loop) */
-      if ((uint64_t)m * 2 >= n && n <= 300000000u && !lk0_retain_ready) {
+      if (3ull * neq >= n && n <= 300000000u && !lk0_retain_ready) {
         if (lk0.alloc(8ull * n)) return -12;
         hipLaunchKernelGGL(k_build_lkeys, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
                            lb0, use_len, ser_mode, (uint64_t*)lk0.p, n);
         lk0p = (const uint64_t*)lk0.p;
       }
-      hipLaunchKernelGGL(k_compact_refine, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
-                         d_idx, d_eq, (uint64_t*)inrun_scan.p, n,
-                         lb0, use_len, ser_mode, lk0p, (uint64_t*)lkey.p,
-                         (uint32_t*)seg.p, (uint32_t*)pos.p);
+      /* fused flags + lookback scan + compaction */
+      static thread_local DBuf rc_st, rc_tick;
+      uint32_t nb_rc = nblocks_for(n, SCAN_TILE);
+      if (rc_st.alloc(8ull * nb_rc)) return -12;
+      if (rc_tick.alloc(32)) return -12;
+      HIP_CHECK(hipMemsetAsync(rc_st.p, 0, 8ull * nb_rc));
+      HIP_CHECK(hipMemsetAsync(rc_tick.p, 0, 32));
+      hipLaunchKernelGGL(k_refine_compact_lb, dim3(nb_rc), dim3(BLOCK), 0, 0, rt,
+                         d_idx, d_eq, n, lb0, use_len, ser_mode, lk0p,
+                         (uint64_t*)lkey.p, (uint32_t*)seg.p, (uint32_t*)pos.p,
+                         (uint64_t*)rc_st.p, (uint32_t*)rc_tick.p,
+                         (uint32_t*)rc_tick.p + 1, (uint64_t*)rc_tick.p + 2);
+      uint64_t hh[2] = {0, 0};
+      HIP_CHECK(hipMemcpy(hh, rc_tick.p, 16, hipMemcpyDeviceToHost));
+      if ((uint32_t)(hh[0] >> 32)) FAIL(-70, "refine compact lookback timeout");
+      HIP_CHECK(hipMemcpy(&ptotal, (uint64_t*)rc_tick.p + 2, 8,
+                          hipMemcpyDeviceToHost));
       if (li == 0 && !use_len && lk0p && retain && !will_combine)
         lk0_retain_ready = true; /* keep the buffer for retention below */
       else if (!lk0_retain_ready)
         lk0.release();
     }
+    uint64_t nruns = ptotal & 0xFFFFFFFFu;
+    uint32_t m = (uint32_t)(ptotal >> 32);
+    if (m == 0) break;
+    if (slotpos.alloc(sizeof(uint32_t) * m)) return -12;
     /* slotpos = copy of pos (ascending) before sort */
     HIP_CHECK(hipMemcpyAsync(slotpos.p, pos.p, sizeof(uint32_t) * m,
                              hipMemcpyDeviceToDevice));
@@ -3614,7 +3726,6 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
   /* large-n headroom: refinement scratch and radix ping-pong temps are dead
      from here; return them to the pool before the output-stream allocation
      (at C3's 1e9 records these hold ~60 GB) */
-  inrun_scan.release();
   lkey.release(); seg.release(); pos.release(); slotpos.release();
   radix_release_temps();
 
