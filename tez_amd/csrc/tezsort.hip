@@ -309,6 +309,57 @@ __global__ void k_hash_partition(RecTable rt, int32_t P, int32_t* d_part, uint32
  * masked to the top sort_bytes bytes: the radix sort only orders those
  * (adaptive pass count — DESIGN.md §4); rarer-than-1% ties are resolved by
  * the refinement levels, whose equality test must see the same mask. */
+__device__ __forceinline__ uint64_t d_composite_one(
+    const RecTable& rt, const int32_t* d_part, int32_t P, int pbits,
+    int ref_pb, int ser_mode, uint64_t mask, uint32_t i) {
+  RecView v = rt_view(rt, i);
+  uint32_t part;
+  if (d_part) part = (uint32_t)d_part[i];
+  else part = (uint32_t)((d_hash_bytes(v.content, (int32_t)v.clen) & 0x7fffffff) % P);
+  uint64_t key;
+  if (ser_mode) {
+    uint32_t proxy = ((v.clen > 0 ? (uint32_t)v.content[0] : 0u) << 16) |
+                     ((v.clen > 1 ? (uint32_t)v.content[1] : 0u) << 8) |
+                     (v.clen > 2 ? (uint32_t)v.content[2] : 0u);
+    int pw = 24 - ref_pb;
+    if (pw < 0) pw = 0;
+    uint32_t proxy_t = pw ? (proxy >> (24 - pw)) : 0;
+    uint64_t ser = 0;
+    uint32_t m = v.klen < 8 ? v.klen : 8;
+    for (uint32_t b = 0; b < m; b++) ser |= (uint64_t)v.key[b] << (56 - 8 * b);
+    key = ((uint64_t)part << (64 - pbits))
+          | ((uint64_t)proxy_t << (64 - pbits - pw))
+          | (ser >> (pbits + pw));
+  } else {
+    uint64_t c = 0;
+    uint32_t m = v.clen < 8 ? v.clen : 8;
+    for (uint32_t b = 0; b < m; b++) c |= (uint64_t)v.content[b] << (56 - 8 * b);
+    key = pbits ? (((uint64_t)part << (64 - pbits)) | (c >> pbits)) : c;
+  }
+  return key & mask;
+}
+
+/* 2 records per iteration: independent hash/gather chains keep two record
+ * reads in flight (the kernel measured 64-67% wave-parked) */
+__global__ void k_build_composite2(RecTable rt, const int32_t* d_part, int32_t P,
+                                   int pbits, int ref_pb, int sort_bytes,
+                                   int ser_mode, uint64_t* d_key, uint32_t* d_idx,
+                                   uint32_t n) {
+  uint64_t mask = (sort_bytes >= 8) ? ~0ull : ~0ull << (8 * (8 - sort_bytes));
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += 2 * stride) {
+    uint32_t j = i + stride;
+    uint64_t k0 = d_composite_one(rt, d_part, P, pbits, ref_pb, ser_mode, mask, i);
+    if (j < n) {
+      uint64_t k1 = d_composite_one(rt, d_part, P, pbits, ref_pb, ser_mode, mask, j);
+      d_key[j] = k1;
+      d_idx[j] = j;
+    }
+    d_key[i] = k0;
+    d_idx[i] = i;
+  }
+}
+
 __global__ void k_build_composite(RecTable rt, const int32_t* d_part, int32_t P,
                                   int pbits, int ref_pb, int sort_bytes, int ser_mode,
                                   uint64_t* d_key, uint32_t* d_idx, uint32_t n) {
@@ -464,11 +515,21 @@ __global__ __launch_bounds__(BLK) void k_radix_scatter(
   uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
   uint16_t my_rank[TILE_ROUNDS];
   uint16_t my_dig[TILE_ROUNDS];
+  /* load phase first: every round's gathers issue back-to-back so the
+     memory latency is paid once per tile, not once per ranking round */
+  #pragma unroll
   for (int r = 0; r < TILE_ROUNDS; r++) {
     uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
     bool active = i < end;
-    KeyT key = active ? keys_in[i] : (KeyT)0;
-    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0u;
+    my_key[r] = active ? keys_in[i] : (KeyT)0;
+    my_a0[r] = active ? a0_in[i] : 0;
+    if (HAS_A1) my_a1v[r] = active ? a1_in[i] : 0;
+    if (HAS_B64) my_b64[r] = active ? b64_in[i] : 0;
+  }
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
+    bool active = i < end;
+    uint32_t d = active ? ((uint32_t)(my_key[r] >> (8 * byte_idx)) & 0xFF) : 0u;
     uint64_t m = ~0ull;
     for (int b = 0; b < 8; b++) {
       uint64_t bb = __ballot((d >> b) & 1);
@@ -479,10 +540,6 @@ __global__ __launch_bounds__(BLK) void k_radix_scatter(
     uint32_t lane_rank = (uint32_t)__popcll(m & lt_mask);
     if (active && lane_rank == 0)
       cnt[(r * OWPB + wv) * RADIX + d] = (uint16_t)__popcll(m);
-    my_key[r] = key;
-    my_a0[r] = active ? a0_in[i] : 0;
-    if (HAS_A1) my_a1v[r] = active ? a1_in[i] : 0;
-    if (HAS_B64) my_b64[r] = active ? b64_in[i] : 0;
     my_rank[r] = (uint16_t)lane_rank;
     my_dig[r] = (uint16_t)d;
   }
@@ -621,11 +678,21 @@ __global__ __launch_bounds__(BLK) void k_onesweep_pass(
   uint64_t my_b64[HAS_B64 ? TILE_ROUNDS : 1];
   uint16_t my_rank[TILE_ROUNDS];
   uint16_t my_dig[TILE_ROUNDS];
+  /* load phase first: every round's gathers issue back-to-back so the
+     memory latency is paid once per tile, not once per ranking round */
+  #pragma unroll
   for (int r = 0; r < TILE_ROUNDS; r++) {
     uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
     bool active = i < end;
-    KeyT key = active ? keys_in[i] : (KeyT)0;
-    uint32_t d = active ? ((uint32_t)(key >> (8 * byte_idx)) & 0xFF) : 0u;
+    my_key[r] = active ? keys_in[i] : (KeyT)0;
+    my_a0[r] = active ? a0_in[i] : 0;
+    if (HAS_A1) my_a1v[r] = active ? a1_in[i] : 0;
+    if (HAS_B64) my_b64[r] = active ? b64_in[i] : 0;
+  }
+  for (int r = 0; r < TILE_ROUNDS; r++) {
+    uint32_t i = start + (uint32_t)r * BLK + threadIdx.x;
+    bool active = i < end;
+    uint32_t d = active ? ((uint32_t)(my_key[r] >> (8 * byte_idx)) & 0xFF) : 0u;
     uint64_t m = ~0ull;
     for (int b = 0; b < 8; b++) {
       uint64_t bb = __ballot((d >> b) & 1);
@@ -636,10 +703,6 @@ __global__ __launch_bounds__(BLK) void k_onesweep_pass(
     uint32_t lane_rank = (uint32_t)__popcll(m & lt_mask);
     if (active && lane_rank == 0)
       cnt[(r * OWPB + wv) * RADIX + d] = (uint16_t)__popcll(m);
-    my_key[r] = key;
-    my_a0[r] = active ? a0_in[i] : 0;
-    if (HAS_A1) my_a1v[r] = active ? a1_in[i] : 0;
-    if (HAS_B64) my_b64[r] = active ? b64_in[i] : 0;
     my_rank[r] = (uint16_t)lane_rank;
     my_dig[r] = (uint16_t)d;
   }
@@ -3524,7 +3587,7 @@ static int sort_and_emit(tzs_sorter* s, HostRT& hrt, uint32_t n,
   if (s->sidx.alloc(sizeof(uint32_t) * n)) return -12;
   uint64_t* d_key = (uint64_t*)s->skey.p;
   uint32_t* d_idx = (uint32_t*)s->sidx.p;
-  hipLaunchKernelGGL(k_build_composite, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+  hipLaunchKernelGGL(k_build_composite2, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
                      d_part_unsorted, P, pbits, prm.ref_pb, SB, ser_mode, d_key,
                      d_idx, n);
   (void)hipEventRecord(ev[1]);
