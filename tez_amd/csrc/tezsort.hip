@@ -1376,7 +1376,9 @@ __global__ void k_apply_leaf2(const uint64_t* k, const uint64_t* lo,
  * sees genuinely long shared prefixes (C3's Zipf words were re-refined over
  * the whole union otherwise — the round-2 profile's dominant block). */
 #define MP2_IPT 8
-#define MP2_TILE (MP2_IPT * MP_BLOCK)
+#define MP2_TILE (MP2_IPT * MP_BLOCK) /* 2048x20B; the 4096 tile measured
+  2x slower (148 vs 77 ms at C3 1e9: occupancy 3 -> 2 blocks/CU and the
+  16-deep serial sub-merge) */
 __device__ __forceinline__ uint32_t d_mp2_diag(
     const uint64_t* ka, const uint64_t* la, uint64_t maskA, uint32_t na,
     const uint64_t* kb, const uint64_t* lb, uint64_t maskB, uint32_t nb,
