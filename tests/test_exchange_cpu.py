@@ -88,6 +88,14 @@ def _worker(rank, world, tmpdir):
     assert rrl2.tolist() == want_lens
     assert rkl2.tolist() == [1] * len(want_lens)
     assert rp2.tolist() == want_parts, f"rank {rank} fused parts mismatch"
+    # per-source chunk splits (reduce_merge feeds one pre-sorted segment
+    # per source rank): counts must tile the received tensors exactly
+    want_splits = []
+    for src in range(world):
+        srecs = {p: src + 1 + p for p in range(P)}
+        want_splits.append(sum(srecs[p] for p in ex.parts_for_dest(P, world, rank)))
+    assert plan.recv_rec_splits == want_splits
+    assert sum(plan.recv_byte_splits) == rd2.numel()
     dist.destroy_process_group()
 
 
