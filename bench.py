@@ -193,12 +193,13 @@ def main():
     ap.add_argument("--force-exchange", action="store_true",
                     help="run the all-to-all-v exchange + reduce-merge path even "
                          "at world_size=1 (bench-code validation)")
-    ap.add_argument("--traffic-bytes", type=float, default=3.25e9,
+    ap.add_argument("--traffic-bytes", type=float, default=2.42e9,
                     help="PMC-measured HBM bytes per dominant-kernel "
                          "(k_onesweep_pass) launch at the default 1e8-record "
                          "workload: 2xFETCH_SIZE + WRITE_SIZE per rocprofv3 "
                          "--pmc with the gfx950 FETCH calibration (profiles/"
-                         "r01_pmc_onesweep.txt). Pass 0 to report null.")
+                         "r02_pmc_final.txt; 1.01x of the 2.4 GB algorithmic "
+                         "bytes). Pass 0 to report null.")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
