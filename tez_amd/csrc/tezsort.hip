@@ -842,13 +842,15 @@ __global__ __launch_bounds__(BLOCK) void k_scan_lookback(
     const uint64_t* in, uint64_t* out, uint32_t n,
     uint64_t* status /* [ntiles] */, uint32_t* ticket, uint32_t* error,
     uint64_t* total_out) {
-  __shared__ uint64_t lds[BLOCK];
+  __shared__ uint64_t wsum[WPB + 1];
   __shared__ uint32_t s_tile;
   __shared__ uint64_t s_excl;
   if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
   __syncthreads();
   const uint32_t tile = s_tile;
   const uint32_t base = tile * SCAN_TILE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wv = threadIdx.x / WAVE;
   uint64_t vals[SCAN_ITEMS];
   uint64_t sum = 0;
   #pragma unroll
@@ -857,15 +859,23 @@ __global__ __launch_bounds__(BLOCK) void k_scan_lookback(
     vals[j] = (i < n) ? in[i] : 0;
     sum += vals[j];
   }
-  lds[threadIdx.x] = sum;
-  __syncthreads();
-  for (int s = 1; s < BLOCK; s <<= 1) {
-    uint64_t t = (threadIdx.x >= (uint32_t)s) ? lds[threadIdx.x - s] : 0;
-    __syncthreads();
-    lds[threadIdx.x] += t;
-    __syncthreads();
+  /* wave-shuffle inclusive scan + tiny cross-wave fixup (the round-1
+     Hillis-Steele block scan cost 16 barriers per tile) */
+  uint64_t inc = sum;
+  for (int s2 = 1; s2 < WAVE; s2 <<= 1) {
+    uint64_t t = __shfl_up((unsigned long long)inc, s2);
+    if (lane >= s2) inc += t;
   }
-  uint64_t block_total = lds[BLOCK - 1];
+  if (lane == WAVE - 1) wsum[wv] = inc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t run = 0;
+    for (int w = 0; w < WPB; w++) { uint64_t t = wsum[w]; wsum[w] = run; run += t; }
+    wsum[WPB] = run;
+  }
+  __syncthreads();
+  uint64_t my_incl = wsum[wv] + inc; /* inclusive prefix incl. my sum */
+  uint64_t block_total = wsum[WPB];
   if (threadIdx.x == 0) {
     if (tile == 0) {
       __hip_atomic_store((os_gu64*)&status[0], OSS_INC | block_total,
@@ -891,7 +901,7 @@ __global__ __launch_bounds__(BLOCK) void k_scan_lookback(
     }
   }
   __syncthreads();
-  uint64_t excl = s_excl + lds[threadIdx.x] - sum;
+  uint64_t excl = s_excl + my_incl - sum;
   #pragma unroll
   for (int j = 0; j < SCAN_ITEMS; j++) {
     uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
@@ -1009,13 +1019,15 @@ __global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
     int level_byte0, int use_len_level, int ser_mode, const uint64_t* lk0,
     uint64_t* lkey, uint32_t* seg, uint32_t* pos,
     uint64_t* status, uint32_t* ticket, uint32_t* error, uint64_t* total_out) {
-  __shared__ uint64_t lds[BLOCK];
+  __shared__ uint64_t wsum[WPB + 1];
   __shared__ uint32_t s_tile;
   __shared__ uint64_t s_excl;
   if (threadIdx.x == 0) s_tile = atomicAdd(ticket, 1u);
   __syncthreads();
   const uint32_t tile = s_tile;
   const uint32_t base = tile * SCAN_TILE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wv = threadIdx.x / WAVE;
   uint8_t my_eq[SCAN_ITEMS];
   uint8_t my_in[SCAN_ITEMS];
   uint64_t vals[SCAN_ITEMS];
@@ -1032,15 +1044,21 @@ __global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
     vals[j] = ((uint64_t)in << 32) | rs;
     sum += vals[j];
   }
-  lds[threadIdx.x] = sum;
-  __syncthreads();
-  for (int s2 = 1; s2 < BLOCK; s2 <<= 1) {
-    uint64_t t = (threadIdx.x >= (uint32_t)s2) ? lds[threadIdx.x - s2] : 0;
-    __syncthreads();
-    lds[threadIdx.x] += t;
-    __syncthreads();
+  uint64_t inc = sum;
+  for (int s2 = 1; s2 < WAVE; s2 <<= 1) {
+    uint64_t t = __shfl_up((unsigned long long)inc, s2);
+    if (lane >= s2) inc += t;
   }
-  uint64_t block_total = lds[BLOCK - 1];
+  if (lane == WAVE - 1) wsum[wv] = inc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t run = 0;
+    for (int w = 0; w < WPB; w++) { uint64_t t = wsum[w]; wsum[w] = run; run += t; }
+    wsum[WPB] = run;
+  }
+  __syncthreads();
+  uint64_t my_incl = wsum[wv] + inc;
+  uint64_t block_total = wsum[WPB];
   if (threadIdx.x == 0) {
     if (tile == 0) {
       __hip_atomic_store((os_gu64*)&status[0], OSS_INC | block_total,
@@ -1066,7 +1084,7 @@ __global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
     }
   }
   __syncthreads();
-  uint64_t excl = s_excl + lds[threadIdx.x] - sum;
+  uint64_t excl = s_excl + my_incl - sum;
   #pragma unroll
   for (int j = 0; j < SCAN_ITEMS; j++) {
     uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
