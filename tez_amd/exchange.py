@@ -52,6 +52,74 @@ def plan_send(rec_ranges, byte_ranges, world: int) -> SendPlan:
     return SendPlan(order, byte_splits, rec_splits, seg_bytes, seg_recs)
 
 
+_P2P_CHUNK = 256 << 20  # bytes per P2P op: far from the >1 GB RCCL
+                        # all_to_all_single truncation (see _move_a2av)
+
+
+def _move_a2av(send, recv, in_splits, out_splits, group):
+    """All-to-all-v as grouped P2P send/recv over xGMI (the north star's
+    `ncclGroupStart; ncclSend/ncclRecv` form) with the self-portion moved by
+    a direct device copy.  Replaces torch's all_to_all_single, which was
+    measured to SILENTLY truncate single-peer portions beyond ~1 GB on this
+    RCCL (a 1.76 GB world=1 exchange delivered exactly the first half;
+    reproduced with a pure-torch loop — gpurun_out/r2_dbg4).  Ops are
+    chunked at 256 MB; both sides chunk identically so P2P matching holds.
+    element_size-aware: splits are in ELEMENTS of the tensors' dtype."""
+    import torch
+    import torch.distributed as dist
+    world = dist.get_world_size(group)
+    me = dist.get_rank(group)
+    esz = send.element_size()
+    soff = [0]
+    for v in in_splits:
+        soff.append(soff[-1] + v)
+    roff = [0]
+    for v in out_splits:
+        roff.append(roff[-1] + v)
+    if in_splits[me] != out_splits[me]:
+        raise RuntimeError("self split mismatch")
+    if in_splits[me]:
+        recv[roff[me]:roff[me + 1]].copy_(send[soff[me]:soff[me + 1]])
+    if world == 1:
+        return
+    chunk_elems = max(_P2P_CHUNK // esz, 1)
+
+    def chunks(t, lo, hi):
+        while lo < hi:
+            c = min(hi - lo, chunk_elems)
+            yield t[lo:lo + c]
+            lo += c
+
+    backend = dist.get_backend(group)
+    if backend == "nccl":
+        ops = []
+        for peer in range(world):
+            if peer == me:
+                continue
+            for c in chunks(send, soff[peer], soff[peer + 1]):
+                ops.append(dist.P2POp(dist.isend, c, peer, group=group))
+            for c in chunks(recv, roff[peer], roff[peer + 1]):
+                ops.append(dist.P2POp(dist.irecv, c, peer, group=group))
+        if ops:
+            for req in dist.batch_isend_irecv(ops):
+                req.wait()
+    else:
+        # gloo (CPU tests): true async send/recv; post receives first
+        reqs = []
+        for peer in range(world):
+            if peer == me:
+                continue
+            for c in chunks(recv, roff[peer], roff[peer + 1]):
+                reqs.append(dist.irecv(c, peer, group=group))
+        for peer in range(world):
+            if peer == me:
+                continue
+            for c in chunks(send, soff[peer], soff[peer + 1]):
+                reqs.append(dist.isend(c, peer, group=group))
+        for r in reqs:
+            r.wait()
+
+
 def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None,
              nparts=None):
     """all-to-all-v of (data bytes, per-record lengths+klens).  send_* are
@@ -99,8 +167,8 @@ def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None,
         pos += 2 * r
         rpos += r
     recv_len = torch.empty(2 * total_out_r, dtype=torch.int32, device=dev)
-    dist.all_to_all_single(recv_len, send_len, [2 * x for x in out_r],
-                           [2 * x for x in in_r], group=group)
+    _move_a2av(send_len, recv_len, [2 * x for x in in_r],
+               [2 * x for x in out_r], group)
     recv_reclen = torch.empty(total_out_r, dtype=torch.int32, device=dev)
     recv_klen = torch.empty(total_out_r, dtype=torch.int32, device=dev)
     pos = rpos = 0
@@ -113,7 +181,7 @@ def exchange(plan: SendPlan, send_data, send_reclen, send_klen, group=None,
         rpos += r
     # collective 2: the record bytes
     recv_data = torch.empty(sum(out_b), dtype=torch.uint8, device=dev)
-    dist.all_to_all_single(recv_data, send_data, out_b, in_b, group=group)
+    _move_a2av(send_data, recv_data, in_b, out_b, group)
     # per-source chunk boundaries (lets reduce_merge ingest each source's
     # chunk as one pre-sorted segment — no re-sort on the reduce side)
     plan.recv_rec_splits = out_r
@@ -230,6 +298,12 @@ def reduce_merge(conf_factory, recv_data, recv_reclen, recv_klen,
         parts_t = None
         if recv_parts is not None and int(recv_parts.numel()) == n:
             parts_t = recv_parts.contiguous()
+        if recv_data.is_cuda:
+            # the collectives/cumsum above run on torch's stream; the engine
+            # reads these tensors on the default stream — without this sync a
+            # large all_to_all is still in flight when the composite build
+            # reads the tail (observed: one zeroed record at ~1 GB received)
+            torch.cuda.synchronize()
         if src_rec_splits is not None:
             r0 = 0
             b0 = 0
