@@ -135,19 +135,26 @@ def test_gpu_randomized_config_sweep():
         text = rng.random() < 0.5
         dup = rng.random() < 0.4
         send_empty = 1 if rng.random() < 0.8 else 0
-        nspill = rng.choice([1, 1, 2, 3])
-        # variable-length TezBytes keys: multi-spill merge order is unpinned
-        # in the reference (segments not comparator-sorted — DESIGN.md §3);
-        # keep those single-spill
-        if not text:
+        nspill = rng.choice([1, 1, 2, 3, 5, 8, 40])
+        # FIXED-length TezBytes keys merge byte-exactly (uniform klen =>
+        # content order == comparator order); VARIABLE-length TezBytes
+        # multi-spill order is unpinned in the reference (segments not
+        # comparator-sorted — DESIGN.md §3) and stays single-spill here
+        # (the documented order is pinned in test_gpu_merge2 instead)
+        fixed_len = (not text) and rng.random() < 0.6
+        if not text and not fixed_len:
             nspill = 1
         combiner = 1 if (rng.random() < 0.3) else 0
+        # explicit partitions across spills (the round-1 rc=-22 refusal)
+        explicit = (not combiner) and rng.random() < 0.25
         n_per = rng.randrange(1, 800)
         key_type = tez_amd.KEY_TEXT if text else tez_amd.KEY_BYTES
         comparator = tez_amd.CMP_TEXT if text else tez_amd.CMP_TEZBYTES
         ser = o.serialize_text if text else o.serialize_bytes_writable
+        klen_fix = rng.randrange(1, 20)
         keypool = [bytes(rng.randrange(256) if not text else rng.randrange(97, 123)
-                         for _ in range(rng.randrange(0 if not text else 1, 20)))
+                         for _ in range(klen_fix if fixed_len else
+                                        rng.randrange(0 if not text else 1, 20)))
                    for _ in range(max(1, n_per // (4 if dup else 1)))]
         conf = tez_amd.make_conf(P, key_type=key_type, comparator=comparator,
                                  combiner=combiner,
@@ -156,19 +163,26 @@ def test_gpu_randomized_config_sweep():
         spills = []
         for sp in range(nspill):
             pairs = []
+            parts = []
             for i in range(n_per):
                 k = ser(keypool[rng.randrange(len(keypool))])
                 v = ((rng.randrange(-5, 100)).to_bytes(4, "big", signed=True)
                      if combiner else
                      o.serialize_bytes_writable(b"v%d.%d" % (sp, i)))
                 pairs.append((k, v))
-            for k, v in pairs:
-                s.write(k, v, -1)
+                # deterministic per-key placement (a partitioner must be a
+                # function of the key for merge parity across spills)
+                parts.append((len(k) * 31 + (k[-1] if len(k) > 4 else 0)) % P
+                             if explicit else -1)
+            for (k, v), pt in zip(pairs, parts):
+                s.write(k, v, pt)
             s.spill()
             d, f, kl = o.build_records(pairs)
             spills.append(o.spill(d, f, kl, P, key_type=key_type,
                                   comparator=comparator, combiner=combiner,
-                                  send_empty=bool(send_empty)))
+                                  send_empty=bool(send_empty),
+                                  partitions=(np.array(parts, dtype=np.int32)
+                                              if explicit else None)))
         s.flush()
         got, gidx = s.output()
         s.close()
@@ -176,6 +190,7 @@ def test_gpu_randomized_config_sweep():
         want = (spills[0] if len(spills) == 1 else
                 o.final_merge(spills, P, comparator=comparator, combiner=gate,
                               send_empty=bool(send_empty)))
-        ctx = f"trial={trial} P={P} text={text} dup={dup} nspill={nspill} comb={combiner} n={n_per}"
+        ctx = (f"trial={trial} P={P} text={text} dup={dup} nspill={nspill} "
+               f"comb={combiner} n={n_per} fixed={fixed_len} expl={explicit}")
         assert gidx == o.index_decode(want["index"], P), ctx
         assert got == want["data"], ctx
