@@ -1275,107 +1275,7 @@ __device__ __forceinline__ uint32_t d_mp_diag(
   return lo;
 }
 
-/* tile-boundary splits computed up front: one thread per boundary, all
- * searches in flight at once (the in-kernel 2-thread serial search made the
- * merge latency-bound: ~25 dependent global loads per block). */
-__global__ void k_mp_partition(const uint64_t* ka, uint32_t na, uint64_t maskA,
-                               const uint64_t* kb, uint32_t nb, uint64_t maskB,
-                               uint32_t nblk, uint32_t tile, uint32_t* splits) {
-  uint64_t total = (uint64_t)na + nb;
-  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t <= nblk;
-       t += gridDim.x * blockDim.x) {
-    uint64_t D = min((uint64_t)t * tile, total);
-    splits[t] = d_mp_diag(ka, maskA, na, kb, maskB, nb, D);
-  }
-}
-__global__ __launch_bounds__(MP_BLOCK) void k_merge_path(
-    const uint64_t* ka, const uint32_t* pa, uint32_t na, uint64_t maskA,
-    uint32_t addA,
-    const uint64_t* kb, const uint32_t* pb, uint32_t nb, uint64_t maskB,
-    uint32_t addB,
-    uint64_t* kout, uint32_t* pout) {
-  __shared__ uint64_t ls_k[MP_TILE];
-  __shared__ uint32_t ls_p[MP_TILE];
-  __shared__ uint32_t s_sp[2];
-  uint64_t total = (uint64_t)na + nb;
-  uint64_t D0 = (uint64_t)blockIdx.x * MP_TILE;
-  if (D0 >= total) return;
-  uint64_t D1 = min(D0 + (uint64_t)MP_TILE, total);
-  if (threadIdx.x < 2) {
-    uint64_t D = threadIdx.x ? D1 : D0;
-    s_sp[threadIdx.x] = d_mp_diag(ka, maskA, na, kb, maskB, nb, D);
-  }
-  __syncthreads();
-  uint32_t a0 = s_sp[0], a1 = s_sp[1];
-  uint32_t b0 = (uint32_t)(D0 - a0), b1 = (uint32_t)(D1 - a1);
-  uint32_t nA = a1 - a0, nB = b1 - b0;
-  for (uint32_t i = threadIdx.x; i < nA; i += blockDim.x) {
-    ls_k[i] = ka[a0 + i] & maskA;
-    ls_p[i] = pa[a0 + i] + addA;
-  }
-  for (uint32_t i = threadIdx.x; i < nB; i += blockDim.x) {
-    ls_k[nA + i] = kb[b0 + i] & maskB;
-    ls_p[nA + i] = pb[b0 + i] + addB;
-  }
-  __syncthreads();
-  uint32_t r = threadIdx.x * MP_IPT;
-  uint32_t tile_n = (uint32_t)(D1 - D0);
-  if (r >= tile_n) return;
-  uint32_t cnt = min(tile_n - r, (uint32_t)MP_IPT);
-  /* local diagonal within the LDS tile (A' first, ties to A') */
-  uint32_t lo = (r > nB) ? r - nB : 0, hi = min(r, nA);
-  while (lo < hi) {
-    uint32_t mid = lo + ((hi - lo) >> 1);
-    if (ls_k[mid] <= ls_k[nA + r - 1 - mid]) lo = mid + 1;
-    else hi = mid;
-  }
-  uint32_t i = lo, j = r - lo;
-  uint64_t ok[MP_IPT];
-  uint32_t op[MP_IPT];
-  #pragma unroll
-  for (uint32_t t = 0; t < MP_IPT; t++) {
-    if (t < cnt) {
-      bool takeA = (j >= nB) || (i < nA && ls_k[i] <= ls_k[nA + j]);
-      uint32_t src = takeA ? i++ : nA + (j++);
-      ok[t] = ls_k[src];
-      op[t] = ls_p[src];
-    }
-  }
-  #pragma unroll
-  for (uint32_t t = 0; t < MP_IPT; t++)
-    if (t < cnt) {
-      kout[D0 + r + t] = ok[t];
-      pout[D0 + r + t] = op[t];
-    }
-}
-
-/* gather src[idx[i]] into dst[i], keeping dst's top (partition) bits:
- * used to rebuild a spill's retained composites in serialized form without
- * recomputing its partition placement (explicit partitioners). */
-__global__ void k_gather_merge_hi(const uint64_t* src, const uint32_t* idx,
-                                  uint64_t* dst, uint64_t himask, uint32_t n) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x)
-    dst[i] = (dst[i] & himask) | (src[idx[i]] & ~himask);
-}
-
-/* nondecreasing check over composites (validates add_sorted_segment input) */
-__global__ void k_check_sorted(const uint64_t* k, uint32_t n, uint32_t* bad) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x)
-    if (i > 0 && k[i] < k[i - 1]) atomicAdd(bad, 1u);
-}
-
 /* leaf/pass-through materialization: apply mask + id rebase */
-__global__ void k_apply_leaf(const uint64_t* k, const uint32_t* p, uint32_t n,
-                             uint64_t mask, uint32_t add,
-                             uint64_t* ko, uint32_t* po) {
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x) {
-    ko[i] = k[i] & mask;
-    po[i] = p[i] + add;
-  }
-}
 __global__ void k_apply_leaf2(const uint64_t* k, const uint64_t* lo,
                               const uint32_t* p, uint32_t n,
                               uint64_t mask, uint32_t add,
