@@ -1275,6 +1275,24 @@ __device__ __forceinline__ uint32_t d_mp_diag(
   return lo;
 }
 
+/* nondecreasing check over composites (validates add_sorted_segment input) */
+__global__ void k_check_sorted(const uint64_t* k, uint32_t n, uint32_t* bad) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x)
+    if (i > 0 && k[i] < k[i - 1]) atomicAdd(bad, 1u);
+}
+
+/* gather src[idx[i]] into dst[i], keeping dst's top (partition) bits:
+ * used to rebuild a spill's retained composites in serialized form without
+ * recomputing its partition placement (explicit partitioners), and — with
+ * himask 0 — as a plain permutation gather (lo-key retention). */
+__global__ void k_gather_merge_hi(const uint64_t* src, const uint32_t* idx,
+                                  uint64_t* dst, uint64_t himask, uint32_t n) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x)
+    dst[i] = (dst[i] & himask) | (src[idx[i]] & ~himask);
+}
+
 /* leaf/pass-through materialization: apply mask + id rebase */
 __global__ void k_apply_leaf2(const uint64_t* k, const uint64_t* lo,
                               const uint32_t* p, uint32_t n,
