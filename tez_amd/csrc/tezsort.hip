@@ -4350,7 +4350,12 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
   if (force_resort < 0) force_resort = getenv("TZS_MERGE_RESORT") ? 1 : 0;
   SpillData finalsp;
   if (!all_sorted || force_resort) {
-    /* fallback: stable radix re-sort of the union (round-1 path) */
+    /* fallback: stable radix re-sort of the union (round-1 path).  It
+       recomputes HashPartitioner placement, so explicit partitions need
+       the merge path's retained composites — fail loudly, never mis-place */
+    if (s->have_explicit_parts && lsp > 1)
+      FAIL(-22, "explicit-partition multi-spill merge requires the merge "
+                "path (retained sorted spills); unset TZS_MERGE_RESORT");
     rc = sort_and_emit(s, hrt, (uint32_t)total_n, nullptr, spill_rle.data(),
                        lsp, &finalsp, combine_at_merge);
     if (rc) return rc;
