@@ -64,13 +64,18 @@ def run_step_c3(tez_amd, gen_batches, free_inputs=False):
     return ctr, tms
 
 
-def run_step_c5(tez_amd, d, off, kl, part, n):
+def run_step_c5(tez_amd, batches):
     """C5 slice: range-partitioned (TotalOrderPartitioner-style) explicit
-    placements through the C-ABI, one spill, flush = rename."""
+    placements through the C-ABI.  One batch = single spill, flush = rename;
+    with --spills > 1 each batch is spilled and flush runs the
+    explicit-partition k-way merge (the round-1 rc=-22 refusal)."""
     t0 = time.perf_counter()
     conf = tez_amd.make_conf(128)
     s = tez_amd.Sorter(conf)
-    s.write_batch_device(d, off, kl, part, n)
+    for d, off, kl, part, n in batches:
+        s.write_batch_device(d, off, kl, part, n)
+        if len(batches) > 1:
+            s.spill()
     t1 = time.perf_counter()
     s.flush()
     t2 = time.perf_counter()
@@ -240,9 +245,14 @@ def main():
         c5conf = tez_amd.make_conf(128)
         if args.records == 100_000_000:
             args.records = 500_000_000  # default C5 slice: 5e8 x 100B = 50 GB
-        d, off, kl, c5_part = tez_amd.generate(seed=SEED + 5, n=args.records,
-                                               kind=2, klen=10, vlen=90,
-                                               conf=c5conf)
+        nsp5 = args.spills if args.spills > 1 and "--spills" in sys.argv else 1
+        c5_batches = []
+        per5 = args.records // nsp5
+        for k in range(nsp5):
+            d, off, kl, c5_part = tez_amd.generate(seed=SEED + 5 + 131 * k,
+                                                   n=per5, kind=2, klen=10,
+                                                   vlen=90, conf=c5conf)
+            c5_batches.append((d, off, kl, c5_part, per5))
     elif args.workload == "c3":
         assert n_gpus == 1, "c3 is the single-GPU merge config"
         c3conf = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
@@ -323,7 +333,7 @@ def main():
 
     def one_step():
         if args.workload == "c5":
-            return run_step_c5(tez_amd, d, off, kl, c5_part, args.records)
+            return run_step_c5(tez_amd, c5_batches)
         if args.workload == "c3":
             return run_step_c3(tez_amd, gen_batches, c3_free_inputs)
         if adopt:
@@ -366,7 +376,8 @@ def main():
             for d, off, kl, _n in gen_batches:
                 tez_amd.free_device(d, off, kl)
     elif args.workload == "c5":
-        tez_amd.free_device(d, off, kl, c5_part)
+        for d, off, kl, c5_part, _n in c5_batches:
+            tez_amd.free_device(d, off, kl, c5_part)
     elif not adopt:
         tez_amd.free_device(d, off, kl)
         if part is not None:
