@@ -1528,18 +1528,24 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       uint32_t v2 = __shfl_xor(maxlen, sh);
       if (v2 > maxlen) maxlen = v2;
     }
-    if (nvalid == WAVE && maxlen <= 32 * HB && !force_simple) {
+    uint64_t hdr_ok = __ballot(my_hdr <= 8);
+    if (nvalid == WAVE && maxlen <= 32 * HB && hdr_ok == ~0ull &&
+        !force_simple) {
       /* half h handles records 2r+h; 2-deep rotate pipeline per half.
        * Byte-granular moves are the measured optimum here: a word-funnel
        * STORE variant ran 17.7 vs 10.3 ms (register pressure), and a
        * one-window-load + shuffle-redistribution GATHER variant was
-       * neutral (10.6 ms) — L1 already serves the byte re-reads. */
+       * neutral (10.6 ms) — L1 already serves the byte re-reads.
+       * (dst | hdr | len) ride ONE packed word and headers <= 8 B ride h0
+       * alone: 3 shuffles per rotate instead of 6. */
+      uint64_t my_meta = (my_dst << 12) | ((uint64_t)my_hdr << 8) | my_len;
       uint32_t r0 = half;           /* first record index for this half */
-      uint64_t src0 = __shfl(my_src, r0), dst0 = __shfl(my_dst, r0);
-      uint64_t h00 = __shfl(my_h0, r0), h10 = __shfl(my_h1, r0);
-      uint32_t len0 = __shfl(my_len, r0), hdr0 = __shfl(my_hdr, r0);
+      uint64_t src0 = __shfl(my_src, r0);
+      uint64_t h00 = __shfl(my_h0, r0);
+      uint64_t meta0 = __shfl(my_meta, r0);
       uint8_t b0[HB];
       {
+        uint32_t len0 = (uint32_t)(meta0 & 0xFF);
         const uint8_t* sp = (const uint8_t*)(uintptr_t)src0;
 #pragma unroll
         for (int k = 0; k < HB; k++)
@@ -1547,27 +1553,27 @@ __global__ void k_emit_records(const RecDesc* desc, const uint8_t* same,
       }
       for (uint32_t r = 0; r < 32; r++) {
         uint32_t rn = 2 * (r + 1) + half;
-        uint64_t src1 = 0, dst1 = 0, h01 = 0, h11 = 0;
-        uint32_t len1 = 0, hdr1 = 0;
+        uint64_t src1 = 0, h01 = 0, meta1 = 0;
         uint8_t b1[HB] = {0};
         if (rn < WAVE) {
-          src1 = __shfl(my_src, rn); dst1 = __shfl(my_dst, rn);
-          h01 = __shfl(my_h0, rn); h11 = __shfl(my_h1, rn);
-          len1 = __shfl(my_len, rn); hdr1 = __shfl(my_hdr, rn);
+          src1 = __shfl(my_src, rn);
+          h01 = __shfl(my_h0, rn);
+          meta1 = __shfl(my_meta, rn);
+          uint32_t len1 = (uint32_t)(meta1 & 0xFF);
           const uint8_t* sp = (const uint8_t*)(uintptr_t)src1;
 #pragma unroll
           for (int k = 0; k < HB; k++)
             b1[k] = (hl + k * 32 < len1) ? sp[hl + k * 32] : 0;
         }
-        uint8_t* w = out + dst0;
-        if (hl < hdr0)
-          w[hl] = (hl < 8) ? (uint8_t)(h00 >> (8 * hl))
-                           : (uint8_t)(h10 >> (8 * (hl - 8)));
+        uint32_t len0 = (uint32_t)(meta0 & 0xFF);
+        uint32_t hdr0 = (uint32_t)((meta0 >> 8) & 0xF);
+        uint8_t* w = out + (meta0 >> 12);
+        if (hl < hdr0) w[hl] = (uint8_t)(h00 >> (8 * hl));
         w += hdr0;
 #pragma unroll
         for (int k = 0; k < HB; k++)
           if (hl + k * 32 < len0) w[hl + k * 32] = b0[k];
-        src0 = src1; dst0 = dst1; h00 = h01; h10 = h11; len0 = len1; hdr0 = hdr1;
+        src0 = src1; h00 = h01; meta0 = meta1;
 #pragma unroll
         for (int k = 0; k < HB; k++) b0[k] = b1[k];
       }
@@ -4548,6 +4554,8 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
 extern "C" int tzs_sorter_output(tzs_sorter* s, const void** d_bytes, int64_t* nbytes,
                                  tzs_index_record* index) {
   if (!s->flushed) FAIL(-22, "flush first");
+  if ((int)s->final_index.size() != s->conf.num_partitions)
+    FAIL(-22, "no final output (final merge disabled: use spill_output)");
   if (d_bytes) *d_bytes = s->final_ifile.p;
   if (nbytes) *nbytes = s->final_len;
   if (index)
@@ -4634,6 +4642,8 @@ extern "C" int tzs_sorter_output_compressed(tzs_sorter* s, const void** d_bytes,
                                             int64_t* nbytes,
                                             tzs_index_record* index) {
   if (!s->flushed) FAIL(-22, "flush first");
+  if ((int)s->final_index.size() != s->conf.num_partitions)
+    FAIL(-22, "no final output (final merge disabled: use spill_output)");
   if (ensure_device_constants()) return -70;
   int P = s->conf.num_partitions;
   const uint8_t* src = (const uint8_t*)s->final_ifile.p;
