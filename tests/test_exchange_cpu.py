@@ -149,3 +149,55 @@ def _worker4(rank, world):
                 want += r
     assert bytes(rd.numpy().tobytes()) == want, f"rank {rank}"
     dist.destroy_process_group()
+
+
+def _worker_chunked(rank, world):
+    """Tiny _P2P_CHUNK forces every peer portion through many chunked
+    isend/irecv ops — the matching order must hold across ranks."""
+    import torch
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ.setdefault("MASTER_PORT", "29519")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    ex._P2P_CHUNK_SAVE = ex._P2P_CHUNK
+    try:
+        ex.__dict__["_P2P_CHUNK"] = 5
+        P = 6
+        recs = {p: [f"ckr{rank}p{p}i{i}".encode() for i in range(3 + p)]
+                for p in range(P)}
+        blob = b""
+        rec_ranges, byte_ranges, reclens, klens = [0], [0], [], []
+        for p in range(P):
+            for r in recs[p]:
+                blob += r
+                reclens.append(len(r))
+                klens.append(1)
+            rec_ranges.append(len(reclens))
+            byte_ranges.append(len(blob))
+        plan = ex.plan_send(rec_ranges, byte_ranges, world)
+        sd, srl, skl = b"", [], []
+        for i, p in enumerate(plan.order):
+            sd += blob[byte_ranges[p]:byte_ranges[p + 1]]
+            srl += reclens[rec_ranges[p]:rec_ranges[p + 1]]
+            skl += klens[rec_ranges[p]:rec_ranges[p + 1]]
+        rd, rrl, rkl = ex.exchange(
+            plan, torch.from_numpy(np.frombuffer(sd, dtype=np.uint8).copy()),
+            torch.tensor(srl, dtype=torch.int32),
+            torch.tensor(skl, dtype=torch.int32))
+        want = b""
+        for src in range(world):
+            srecs = {p: [f"ckr{src}p{p}i{i}".encode() for i in range(3 + p)]
+                     for p in range(P)}
+            for p in ex.parts_for_dest(P, world, rank):
+                for r in srecs[p]:
+                    want += r
+        assert bytes(rd.numpy().tobytes()) == want, f"rank {rank} chunked"
+    finally:
+        ex.__dict__["_P2P_CHUNK"] = ex._P2P_CHUNK_SAVE
+    dist.destroy_process_group()
+
+
+def test_exchange_gloo_chunked_p2p():
+    import torch.multiprocessing as mp
+    os.environ["MASTER_PORT"] = "29519"
+    mp.spawn(_worker_chunked, args=(2,), nprocs=2, join=True)
