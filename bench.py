@@ -189,7 +189,8 @@ def main():
                          "world>1 the skew drives the all-to-all-v. "
                          "c5: TeraSort-shaped 10B key + 90B value, range "
                          "partitions (single-GPU slice of configs[4])")
-    ap.add_argument("--spills", type=int, default=32)
+    ap.add_argument("--spills", type=int, default=None,
+                    help="forced spill count (default: 32 for c3, 1 for c5)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--no-adopt", action="store_true",
                     help="force the copying absorb path (default: the sorter "
@@ -245,7 +246,7 @@ def main():
         c5conf = tez_amd.make_conf(128)
         if args.records == 100_000_000:
             args.records = 500_000_000  # default C5 slice: 5e8 x 100B = 50 GB
-        nsp5 = args.spills if args.spills > 1 and "--spills" in sys.argv else 1
+        nsp5 = args.spills or 1
         c5_batches = []
         per5 = args.records // nsp5
         for k in range(nsp5):
@@ -255,6 +256,7 @@ def main():
             c5_batches.append((d, off, kl, c5_part, per5))
     elif args.workload == "c3":
         assert n_gpus == 1, "c3 is the single-GPU merge config"
+        args.spills = args.spills or 32
         c3conf = tez_amd.make_conf(256, key_type=tez_amd.KEY_TEXT,
                                    comparator=tez_amd.CMP_TEXT)
         per = args.records // args.spills
