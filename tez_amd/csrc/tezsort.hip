@@ -634,6 +634,26 @@ __global__ void k_global_hist_all(const KeyT* keys, uint32_t n, int first_byte,
     if (lh[i]) atomicAdd(&counts[i], lh[i]);
 }
 
+/* bases[p][d] = exclusive prefix of counts[p][*] — one block per pass
+ * (replaces a host D2H + prefix + H2D round trip per base sort) */
+__global__ void k_os_bases(const uint32_t* counts, uint32_t* bases) {
+  __shared__ uint32_t tot[RADIX];
+  int p = blockIdx.x;
+  int d = threadIdx.x;
+  tot[d] = counts[p * RADIX + d];
+  __syncthreads();
+  if (d == 0) {
+    uint32_t run = 0;
+    for (int i = 0; i < RADIX; i++) {
+      uint32_t t = tot[i];
+      tot[i] = run;
+      run += t;
+    }
+  }
+  __syncthreads();
+  bases[p * RADIX + d] = tot[d];
+}
+
 template <typename KeyT, bool HAS_A1, bool HAS_B64 = false, int BLK = BLOCK>
 __global__ __launch_bounds__(BLK) void k_onesweep_pass(
     const KeyT* keys_in, KeyT* keys_out,
@@ -2901,18 +2921,8 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
     hipLaunchKernelGGL((k_global_hist_all<KeyT>), dim3(grid1d(n)), dim3(BLOCK),
                        (uint32_t)(4 * npasses * RADIX), 0, kin, n,
                        first_byte, npasses, (uint32_t*)gh.p);
-    std::vector<uint32_t> h_cnt(npasses * RADIX), h_base(npasses * RADIX);
-    HIP_CHECK(hipMemcpy(h_cnt.data(), gh.p, 4u * npasses * RADIX,
-                        hipMemcpyDeviceToHost));
-    for (int p = 0; p < npasses; p++) {
-      uint32_t run = 0;
-      for (int d = 0; d < RADIX; d++) {
-        h_base[p * RADIX + d] = run;
-        run += h_cnt[p * RADIX + d];
-      }
-    }
-    HIP_CHECK(hipMemcpyAsync(gbases.p, h_base.data(), 4u * npasses * RADIX,
-                             hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_os_bases, dim3(npasses), dim3(RADIX), 0, 0,
+                       (const uint32_t*)gh.p, (uint32_t*)gbases.p);
     for (int b = first_byte; b < nbytes_key; b++) {
       int p = b - first_byte;
       HIP_CHECK(hipMemsetAsync(st.p, 0, 4ull * nb_os * RADIX));
