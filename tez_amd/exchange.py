@@ -3,8 +3,9 @@ reference's HTTP fetch (Shuffle/ShuffleScheduler/FetcherOrderedGrouped —
 SURVEY §3b/§8e).  Partition p is owned by rank p % world_size; the per-pair
 byte counts come from the spill index triples (the ShuffleHeader-equivalent
 metadata, ShuffleHeader.java:82-106), exchanged first as a size matrix
-(all_gather), then torch.distributed.all_to_all_single moves the bytes —
-grouped send/recv over the 7 pairwise xGMI links.
+(all_gather); the bytes then move as grouped P2P send/recv over the 7
+pairwise xGMI links (see _move_a2av — torch's all_to_all_single is avoided:
+it silently truncates >1 GB single-peer portions on this RCCL).
 
 Planning functions are torch-free and covered by CPU (gloo) tests; the tensor
 movement uses whatever backend the process group has (nccl=RCCL on the GPU

@@ -48,6 +48,10 @@ class OrderedPartitionedKVOutput:
                 "org.apache.hadoop.io.compress.DefaultCodec":
             raise ValueError(f"unsupported compress.codec {codec} "
                              "(engine emits DefaultCodec/zlib)")
+        if self._compress and not self._final_merge:
+            raise ValueError("tez.runtime.compress with pipelined shuffle "
+                             "(final merge off) is not supported: per-spill "
+                             "segments are served uncompressed")
         key_cls = self.props.get("tez.runtime.key.class",
                                  "org.apache.hadoop.io.BytesWritable")
         if key_cls == "org.apache.hadoop.io.Text":
