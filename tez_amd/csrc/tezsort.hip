@@ -1105,6 +1105,19 @@ __global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
   }
   __syncthreads();
   uint64_t excl = s_excl + my_incl - sum;
+  /* prefetch the dense-key gathers for all 8 elements up front (the
+     guarded in-loop loads serialized one gather at a time) */
+  uint64_t pre_k[SCAN_ITEMS];
+  if (lk0) {
+    uint32_t pre_ix[SCAN_ITEMS];
+    #pragma unroll
+    for (int j = 0; j < SCAN_ITEMS; j++) {
+      uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
+      pre_ix[j] = sidx[i < n ? i : n - 1];
+    }
+    #pragma unroll
+    for (int j = 0; j < SCAN_ITEMS; j++) pre_k[j] = lk0[pre_ix[j]];
+  }
   #pragma unroll
   for (int j = 0; j < SCAN_ITEMS; j++) {
     uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
@@ -1114,7 +1127,7 @@ __global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
       uint32_t sg = (uint32_t)(excl & 0xFFFFFFFFu) + is_start - 1;
       uint64_t k;
       if (lk0) {
-        k = lk0[sidx[i]];
+        k = pre_k[j];
       } else if (use_len_level) {
         RecView v = rt_view(rt, sidx[i]);
         k = ser_mode ? v.klen : v.clen;
@@ -1468,11 +1481,21 @@ __global__ void k_build_desc(RecTable rt, const uint32_t* sidx /* NULL: identity
 __global__ void k_permute_desc(const RecDesc* src, const uint32_t* sidx,
                                RecDesc* dst, const uint8_t* same,
                                const uint32_t* parts, uint64_t* sizes, uint32_t n) {
+  /* 2 records per iteration: two independent sidx->desc gather chains */
+  uint32_t stride = gridDim.x * blockDim.x;
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += gridDim.x * blockDim.x) {
-    RecDesc d = src[sidx[i]];
-    dst[i] = d;
-    sizes[i] = d_rec_emit_size(d, same, parts, i);
+       i += 2 * stride) {
+    uint32_t j = i + stride;
+    uint32_t si = sidx[i];
+    uint32_t sj = (j < n) ? sidx[j] : si;
+    RecDesc d0 = src[si];
+    RecDesc d1 = src[sj];
+    dst[i] = d0;
+    sizes[i] = d_rec_emit_size(d0, same, parts, i);
+    if (j < n) {
+      dst[j] = d1;
+      sizes[j] = d_rec_emit_size(d1, same, parts, j);
+    }
   }
 }
 /* same-as-prev full-key flags are exactly the final eq[] array.
