@@ -399,3 +399,48 @@ def test_varlen_tezbytes_multispill_documented_order(engine):
     got, gidx = s.output()
     s.close()
     assert _read_rows(got, gidx, P) == _expect_documented(pool, P)
+
+
+def test_merge_rle_gate_same_key_events(engine):
+    """The multi-spill auto-RLE gate counts SAME_KEY machine events (equal
+    keys meeting across segments / RLE'd sources), NOT raw adjacent-equal
+    pairs — TezMerger's emergent trace, restated by the oracle merge.  This
+    shape sits in the divergence window: union adjacency 10.25% (a raw
+    gate would enable RLE) but only 1.25% cross-segment events (the
+    machine gate keeps it off), with every spill below its own gate."""
+    P = 1
+    val = o.serialize_bytes_writable
+
+    def spill_keys(tag, ndup, nuniq):
+        ks = []
+        for i in range(ndup):
+            k = o.serialize_bytes_writable(b"%c%03dDUP" % (tag, i))
+            ks += [k, k]  # key appears twice within this spill
+        for i in range(nuniq):
+            ks.append(o.serialize_bytes_writable(b"%c%04dun" % (tag, i)))
+        return ks
+
+    shared = [o.serialize_bytes_writable(b"S%03dshr" % i) for i in range(50)]
+    ka = spill_keys(ord('A'), 180, 2000 - 2 * 180 - 50) + shared
+    kb = spill_keys(ord('B'), 180, 2000 - 2 * 180 - 50) + shared
+    assert len(ka) == len(kb) == 2000
+    conf = engine.make_conf(P, key_type=engine.KEY_BYTES,
+                            comparator=engine.CMP_TEZBYTES)
+    s = engine.Sorter(conf)
+    spills = []
+    for tag, keys in (("a", ka), ("b", kb)):
+        pairs = [(k, val(b"%s%04d" % (tag.encode(), i)))
+                 for i, k in enumerate(keys)]
+        for k, v in pairs:
+            s.write(k, v, -1)
+        s.spill()
+        d, f, kl = o.build_records(pairs)
+        sp = o.spill(d, f, kl, P)
+        assert sp["rle"] == 0  # each spill below its own gate
+        spills.append(sp)
+    s.flush()
+    got, gidx = s.output()
+    s.close()
+    want = o.final_merge(spills, P)
+    assert gidx == o.index_decode(want["index"], P)
+    assert got == want["data"]

@@ -3785,8 +3785,36 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
      loop — after the final level eq[i] means full-key-equal) */
   uint32_t neq_final = neq;
   int writer_rle;
-  if (s->conf.rle >= 0) writer_rle = s->conf.rle;
-  else writer_rle = ((uint64_t)neq_final * 10 > n) ? 1 : 0;
+  if (s->conf.rle >= 0) {
+    writer_rle = s->conf.rle;
+  } else if (nspills_rle > 1) {
+    /* multi-spill merge: the auto gate counts SAME_KEY machine events —
+       equal keys meeting ACROSS segments or carried by an RLE'd source
+       stream — exactly TezMerger's emergent rle trace (the oracle's
+       merge gate; a raw adjacent-equal count would over-RLE unions whose
+       within-spill duplicates sat below each spill's own gate).  Computed
+       by running the provenance kernel with writer_rle=0 and counting. */
+    if (s->same.alloc(n)) return -12;
+    {
+      static thread_local DBuf d_sprle0, cnt0;
+      if (d_sprle0.alloc(nspills_rle)) return -12;
+      if (cnt0.alloc(4)) return -12;
+      HIP_CHECK(hipMemcpyAsync(d_sprle0.p, h_spill_rle, nspills_rle,
+                               hipMemcpyHostToDevice));
+      hipLaunchKernelGGL(k_writer_same, dim3(grid1d(n)), dim3(BLOCK), 0, 0, rt,
+                         d_idx, d_eq, 0, (const uint8_t*)d_sprle0.p,
+                         (uint8_t*)s->same.p, n);
+      HIP_CHECK(hipMemsetAsync(cnt0.p, 0, 4));
+      hipLaunchKernelGGL(k_count_nonzero_u8, dim3(grid1d(n)), dim3(BLOCK), 0, 0,
+                         (const uint8_t*)s->same.p, n, (uint32_t*)cnt0.p);
+      uint32_t ev = 0;
+      HIP_CHECK(hipMemcpy(&ev, cnt0.p, 4, hipMemcpyDeviceToHost));
+      writer_rle = ((uint64_t)ev * 10 > n) ? 1 : 0;
+    }
+  } else {
+    /* single spill: adjacent-equal pairs in sorted order */
+    writer_rle = ((uint64_t)neq_final * 10 > n) ? 1 : 0;
+  }
 
   /* combiner stage: replace the sorted view with folded records */
   s->combine_applied = false;
