@@ -444,3 +444,37 @@ def test_merge_rle_gate_same_key_events(engine):
     want = o.final_merge(spills, P)
     assert gidx == o.index_decode(want["index"], P)
     assert got == want["data"]
+
+
+def test_add_sorted_segment_combiner(engine):
+    """Reduce-side combine (MergeManager.java:916-921 runCombineProcessor):
+    3 pre-sorted segments with shared keys, SUM_INT combiner at the merge —
+    folded output equals the oracle's combined spill of the union."""
+    import struct
+    P = 2
+    keys = [o.serialize_bytes_writable(b"ck%04d" % i) for i in range(300)]
+    segs = []
+    for sid in range(3):
+        seg = [(k, struct.pack(">i", 1000 * sid + i))
+               for i, k in enumerate(keys)]
+        segs.append(_tezbytes_sorted(seg, P))
+    conf = engine.make_conf(P, key_type=engine.KEY_BYTES,
+                            comparator=engine.CMP_TEZBYTES, combiner=1,
+                            min_spills_for_combine=3)
+    s = engine.Sorter(conf)
+    keep = []
+    for seg in segs:
+        d, f, kl = o.build_records(seg)
+        dd, ofp, kp, _ = engine.upload_records(d.tobytes(), f, kl)
+        keep.append((dd, ofp, kp))
+        s.add_sorted_segment(dd, ofp, kp, None, len(seg))
+    s.flush()
+    got, gidx = s.output()
+    s.close()
+    union = [kv for seg in segs for kv in seg]
+    d, f, kl = o.build_records(union)
+    want = o.spill(d, f, kl, P, combiner=1)
+    assert gidx == o.index_decode(want["index"], P)
+    assert got == want["data"]
+    for bufs in keep:
+        engine.free_device(*bufs)
