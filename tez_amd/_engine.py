@@ -255,6 +255,26 @@ class Sorter:
             "sorted_columnar")
         return d, o, k, list(rr), list(br)
 
+    def write_files_compressed(self, local_dir: str, unique_id: str):
+        """Reference on-disk layout with DefaultCodec segments: the TIF\\1
+        stream from the device deflate as file.out + the matching index
+        (partLength = compressed) — servable by the ShuffleHandler and
+        consumable by stock fetchers (IFile.java:352-368)."""
+        import os
+        import zlib
+        data, index = self.output_compressed()
+        base = os.path.join(local_dir, "output", unique_id)
+        os.makedirs(base, exist_ok=True)
+        with open(os.path.join(base, "file.out"), "wb") as f:
+            f.write(data)
+        ix = bytearray()
+        for st, raw, cl in index:
+            ix += int(st).to_bytes(8, "big") + int(raw).to_bytes(8, "big") \
+                + int(cl).to_bytes(8, "big")
+        ix += zlib.crc32(bytes(ix)).to_bytes(8, "big")
+        with open(os.path.join(base, "file.out.index"), "wb") as f:
+            f.write(bytes(ix))
+
     def write_files(self, local_dir: str, unique_id: str):
         _ck(lib().tzs_sorter_write_files(self.h, local_dir.encode(), unique_id.encode()),
             "write_files")
