@@ -348,16 +348,20 @@ __global__ void k_build_composite2(RecTable rt, const int32_t* d_part, int32_t P
                                    uint32_t n) {
   uint64_t mask = (sort_bytes >= 8) ? ~0ull : ~0ull << (8 * (8 - sort_bytes));
   uint32_t stride = gridDim.x * blockDim.x;
-  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += 2 * stride) {
-    uint32_t j = i + stride;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += 4 * stride) {
+    uint32_t i1 = i + stride, i2 = i + 2 * stride, i3 = i + 3 * stride;
     uint64_t k0 = d_composite_one(rt, d_part, P, pbits, ref_pb, ser_mode, mask, i);
-    if (j < n) {
-      uint64_t k1 = d_composite_one(rt, d_part, P, pbits, ref_pb, ser_mode, mask, j);
-      d_key[j] = k1;
-      d_idx[j] = j;
-    }
+    uint64_t k1 = (i1 < n) ? d_composite_one(rt, d_part, P, pbits, ref_pb,
+                                             ser_mode, mask, i1) : 0;
+    uint64_t k2 = (i2 < n) ? d_composite_one(rt, d_part, P, pbits, ref_pb,
+                                             ser_mode, mask, i2) : 0;
+    uint64_t k3 = (i3 < n) ? d_composite_one(rt, d_part, P, pbits, ref_pb,
+                                             ser_mode, mask, i3) : 0;
     d_key[i] = k0;
     d_idx[i] = i;
+    if (i1 < n) { d_key[i1] = k1; d_idx[i1] = i1; }
+    if (i2 < n) { d_key[i2] = k2; d_idx[i2] = i2; }
+    if (i3 < n) { d_key[i3] = k3; d_idx[i3] = i3; }
   }
 }
 
