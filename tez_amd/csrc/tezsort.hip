@@ -2992,11 +2992,18 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
      u32-key+u64 payload at 256 (41 KB, 12 waves/CU — the 1024 variant
      would need 162 KB).  Pre-rewrite this measured slower and was gated
      off. */
-  /* payload shapes stay on the classic path: onesweep for them measured
-     slower (C3 1e9 sort_ns 551 vs 580 ms — per-sort host histogram syncs
-     dominate at refinement sizes), and its a1 instantiation would not fit
-     LDS at the 12-round tile */
-  if (use_onesweep && !has_a1 && !has_b64 && n >= 20000) {
+  /* payload shapes (the refinement's level-key and run-id sorts) can ride
+     onesweep at 256-thread tiles (the 1024 a1 instantiation exceeds LDS at
+     12 rounds); re-testable now that the histogram prefix is device-side
+     (TZS_OS_PAYLOAD=1) */
+  static int os_payload = -1;
+  if (os_payload < 0) {
+    const char* e = getenv("TZS_OS_PAYLOAD");
+    os_payload = (e && e[0] == '1') ? 1 : 0;
+  }
+  bool os_shape_ok = (!has_a1 && !has_b64) ||
+                     (os_payload && !(has_a1 && has_b64));
+  if (use_onesweep && os_shape_ok && n >= 20000) {
     int npasses = nbytes_key - first_byte;
     /* bigger blocks scale the tile while KEEPING 16 waves/CU: 512 threads =
        4096-elem tile (2 blocks x 8 waves), 1024 = 8192-elem tile (1 block x
@@ -3009,7 +3016,8 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
       os_blk = e ? atoi(e) : 1024;
       if (os_blk != 256 && os_blk != 512 && os_blk != 1024) os_blk = 1024;
     }
-    uint32_t os_tile = (uint32_t)TILE_ROUNDS * (uint32_t)os_blk;
+    int eff_blk = (has_a1 || has_b64) ? 256 : os_blk;
+    uint32_t os_tile = (uint32_t)TILE_ROUNDS * (uint32_t)eff_blk;
     uint32_t nb_os = nblocks_for(n, os_tile);
     static thread_local DBuf gh, gbases, st, tick;
     if (gh.alloc(4u * npasses * RADIX)) return -12;
@@ -3032,7 +3040,17 @@ static int radix_sort(KeyT* d_key, uint32_t* d_a0, uint32_t* d_a1, uint32_t n,
       if (nev < 16) { (void)hipEventCreate(&evs[nev]); (void)hipEventCreate(&eve[nev]);
                       (void)hipEventRecord(evs[nev]); }
       const uint32_t* pbases = (const uint32_t*)((uint32_t*)gbases.p + p * RADIX);
-      if (os_blk == 1024)
+      if (has_b64) {
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, true, 256>), dim3(nb_os),
+                           dim3(256), 0, 0, kin, kout, a0in, a0out, nullptr,
+                           nullptr, b64in, b64out, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      } else if (has_a1) {
+        hipLaunchKernelGGL((k_onesweep_pass<KeyT, true, false, 256>), dim3(nb_os),
+                           dim3(256), 0, 0, kin, kout, a0in, a0out, a1in, a1out,
+                           nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
+                           (uint32_t*)tick.p, (uint32_t*)tick.p + 1);
+      } else if (os_blk == 1024)
         hipLaunchKernelGGL((k_onesweep_pass<KeyT, false, false, 1024>), dim3(nb_os),
                            dim3(1024), 0, 0, kin, kout, a0in, a0out, nullptr, nullptr,
                            nullptr, nullptr, n, b, pbases, (uint32_t*)st.p,
