@@ -4475,7 +4475,8 @@ extern "C" int tzs_sorter_flush(tzs_sorter* s) {
    * MergeQueue semantics, :466-706): every spill retained its refined sorted
    * order (skey = masked composites, sidxb = original ids), so the flush
    * merges the k sorted streams with the stable pairwise merge-path kernel
-   * — ceil(log2 k) passes of 24 B/element — instead of re-sorting the union
+   * over 128-bit (composite, lo) keys — ceil(log2 k) passes of ~40 B/element
+   * — instead of re-sorting the union
    * (~8 radix passes).  Ties go to the lower spill, so equal-composite runs
    * come out in global-id order: bit-identical to the stable union re-sort
    * the refinement + emit stages were parity-proven on.
