@@ -129,7 +129,7 @@ def test_gpu_randomized_config_sweep():
         pytest.skip("no GPU")
     import os
     trials = int(os.environ.get("TZS_SWEEP_TRIALS", "32"))
-    rng = random.Random(0xF1122)
+    rng = random.Random(int(os.environ.get("TZS_SWEEP_SEED", "0xF1122"), 0))
     for trial in range(trials):
         P = rng.choice([1, 2, 3, 7, 16, 63, 200])
         text = rng.random() < 0.5

@@ -1056,17 +1056,32 @@ __global__ __launch_bounds__(BLOCK) void k_refine_compact_lb(
   uint8_t my_in[SCAN_ITEMS];
   uint64_t vals[SCAN_ITEMS];
   uint64_t sum = 0;
-  #pragma unroll
-  for (int j = 0; j < SCAN_ITEMS; j++) {
-    uint32_t i = base + threadIdx.x * SCAN_ITEMS + j;
-    uint8_t e = (i < n) ? eq[i] : 0;
-    uint8_t nx = (i + 1 < n) ? eq[i + 1] : 0;
-    uint8_t in = (i < n) && (e || nx);
-    my_eq[j] = e;
-    my_in[j] = in;
-    uint64_t rs = (in && !e) ? 1 : 0;
-    vals[j] = ((uint64_t)in << 32) | rs;
-    sum += vals[j];
+  {
+    /* eq flags for the thread's 8 elements ride ONE aligned u64 load plus
+       one byte (the i+1 lookahead) — 16 serial byte loads left the kernel
+       93% wave-parked (r02 PMC) */
+    uint32_t i0b = base + threadIdx.x * SCAN_ITEMS;
+    uint64_t w = 0;
+    uint8_t nxt = 0;
+    if (i0b + SCAN_ITEMS <= n) {
+      w = *(const uint64_t*)(eq + i0b);
+      nxt = (i0b + SCAN_ITEMS < n) ? eq[i0b + SCAN_ITEMS] : 0;
+    } else if (i0b < n) {
+      for (uint32_t j = 0; i0b + j < n && j < SCAN_ITEMS; j++)
+        w |= (uint64_t)eq[i0b + j] << (8 * j);
+    }
+    #pragma unroll
+    for (int j = 0; j < SCAN_ITEMS; j++) {
+      uint32_t i = i0b + j;
+      uint8_t e = (uint8_t)(w >> (8 * j));
+      uint8_t nx = (j + 1 < SCAN_ITEMS) ? (uint8_t)(w >> (8 * (j + 1))) : nxt;
+      uint8_t in = (i < n) && (e || ((i + 1 < n) && nx));
+      my_eq[j] = e;
+      my_in[j] = in;
+      uint64_t rs = (in && !e) ? 1 : 0;
+      vals[j] = ((uint64_t)in << 32) | rs;
+      sum += vals[j];
+    }
   }
   uint64_t inc = sum;
   for (int s2 = 1; s2 < WAVE; s2 <<= 1) {
