@@ -2833,14 +2833,11 @@ static int scan_u64(const uint64_t* d_in, uint64_t* d_out, uint32_t n, uint64_t*
         hipLaunchKernelGGL(k_scan_lookback, dim3(nb_l), dim3(BLOCK), 0, 0,
                            d_in, d_out, n, (uint64_t*)st.p, (uint32_t*)tick.p,
                            (uint32_t*)tick.p + 1, (uint64_t*)tick.p + 2);
-        uint64_t hh[2] = {0, 0}; /* {ticket+error, total} */
-        HIP_CHECK(hipMemcpy(hh, tick.p, 16, hipMemcpyDeviceToHost));
+        uint64_t hh[3] = {0, 0, 0}; /* {ticket|error, pad, total} */
+        HIP_CHECK(hipMemcpy(hh, tick.p, 24, hipMemcpyDeviceToHost));
         uint32_t err = (uint32_t)(hh[0] >> 32);
         if (!err) {
-          if (h_total) {
-            HIP_CHECK(hipMemcpy(h_total, (uint64_t*)tick.p + 2, 8,
-                                hipMemcpyDeviceToHost));
-          }
+          if (h_total) *h_total = hh[2];
           return 0;
         }
         /* lookback timeout: in-place input is already destroyed — that
@@ -3811,11 +3808,10 @@ static int refine_and_emit(tzs_sorter* s, std::vector<SegDesc> hsegs,
                          (uint64_t*)lkey.p, (uint32_t*)seg.p, (uint32_t*)pos.p,
                          (uint64_t*)rc_st.p, (uint32_t*)rc_tick.p,
                          (uint32_t*)rc_tick.p + 1, (uint64_t*)rc_tick.p + 2);
-      uint64_t hh[2] = {0, 0};
-      HIP_CHECK(hipMemcpy(hh, rc_tick.p, 16, hipMemcpyDeviceToHost));
+      uint64_t hh[3] = {0, 0, 0};
+      HIP_CHECK(hipMemcpy(hh, rc_tick.p, 24, hipMemcpyDeviceToHost));
       if ((uint32_t)(hh[0] >> 32)) FAIL(-70, "refine compact lookback timeout");
-      HIP_CHECK(hipMemcpy(&ptotal, (uint64_t*)rc_tick.p + 2, 8,
-                          hipMemcpyDeviceToHost));
+      ptotal = hh[2];
       if (li == 0 && !use_len && lk0p && retain && !will_combine)
         lk0_retain_ready = true; /* keep the buffer for retention below */
       else if (!lk0_retain_ready)
