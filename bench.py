@@ -440,7 +440,8 @@ def main():
                 "partitions": (199 if args.workload == "c4" else
                                128 if args.workload == "c5" else
                                256 if args.workload == "c3" else PARTS),
-                "parallelism": f"shuffle-shard p%%{n_gpus}" if n_gpus > 1 else "single",
+                "parallelism": ("shuffle-shard p%" + str(n_gpus))
+                               if n_gpus > 1 else "single",
             },
             "roofline": roofline,
             "cpu_baseline": cpu,
